@@ -1,0 +1,157 @@
+/* oracle/simplify.c — CPU restatement of the per-label quadric edge-collapse
+ * simplifier the reference runs through zmesh's `Mesher.get(id,
+ * reduction_factor, max_error, voxel_centered=True)`
+ * (/root/reference/igneous/tasks/mesh/mesh.py:376-381). zmesh sources are
+ * not vendored (parity unpinned — see mc_oracle.c header); this restates the
+ * published algorithm (Garland-Heckbert error quadrics, edge collapse,
+ * reduction target = ntris/reduction_factor, max_error = max allowed
+ * vertex displacement-error in physical nm) with a DETERMINISTIC
+ * matched-pair independent-set schedule that the HIP kernel mirrors
+ * exactly (see DESIGN.md "simplifier contract").
+ *
+ * Contract (canonical; shared bit-exact with igneous_amd/csrc):
+ *  - vertex quadric Q_v = sum over incident faces (in face-index order) of
+ *    the face's UNIT plane quadric (w=1; f32 arithmetic, plain summation in
+ *    ascending face index order) — unit weight keeps the cost in nm^2 so it
+ *    compares directly against max_error^2;
+ *  - candidate per edge (u,v), u<v: placement = midpoint (0.5*(pu+pv)),
+ *    cost = (Qu+Qv)(m) in nm^2;
+ *  - a round: every vertex picks its cheapest incident edge (ties -> the
+ *    smaller peer index); an edge collapses iff both endpoints picked it
+ *    AND cost <= max_error^2;
+ *  - collapse moves both endpoints to the midpoint, remaps v->u (u<v),
+ *    drops degenerate faces; rounds repeat until ntris <= target or no
+ *    edge collapses in a round.
+ */
+
+#include <stdint.h>
+#include <stdlib.h>
+#include <string.h>
+#include <math.h>
+
+typedef struct { float q[10]; } quad10; /* symmetric 4x4: a2,ab,ac,ad,b2,bc,bd,c2,cd,d2 */
+
+static void quad_add_plane(quad10 *Q, float a, float b, float c, float d, float w) {
+  Q->q[0] += w * a * a; Q->q[1] += w * a * b; Q->q[2] += w * a * c; Q->q[3] += w * a * d;
+  Q->q[4] += w * b * b; Q->q[5] += w * b * c; Q->q[6] += w * b * d;
+  Q->q[7] += w * c * c; Q->q[8] += w * c * d;
+  Q->q[9] += w * d * d;
+}
+
+static float quad_eval(const quad10 *Q, float x, float y, float z) {
+  return Q->q[0]*x*x + 2.0f*Q->q[1]*x*y + 2.0f*Q->q[2]*x*z + 2.0f*Q->q[3]*x
+       + Q->q[4]*y*y + 2.0f*Q->q[5]*y*z + 2.0f*Q->q[6]*y
+       + Q->q[7]*z*z + 2.0f*Q->q[8]*z
+       + Q->q[9];
+}
+
+void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
+                       uint32_t *faces, uint32_t *ntris_io,
+                       uint32_t reduction_factor, float max_error) {
+  uint32_t nv = *nverts_io, nt = *ntris_io;
+  if (reduction_factor <= 1 || nt == 0) return;
+  uint32_t target = nt / reduction_factor;
+  if (target < 1) target = 1;
+  const float max_cost = max_error * max_error;
+
+  uint32_t *remap = (uint32_t*)malloc(nv * sizeof(uint32_t));
+  quad10 *Q = (quad10*)malloc(nv * sizeof(quad10));
+  /* pick[v]: encoded best edge for vertex v */
+  uint64_t *pick = (uint64_t*)malloc(nv * sizeof(uint64_t));
+
+  int progress = 1;
+  while (nt > target && progress) {
+    progress = 0;
+    /* 1. vertex quadrics, ascending face order */
+    memset(Q, 0, nv * sizeof(quad10));
+    for (uint32_t t = 0; t < nt; t++) {
+      uint32_t i0 = faces[3*t], i1 = faces[3*t+1], i2 = faces[3*t+2];
+      float *p0 = verts + 3*i0, *p1 = verts + 3*i1, *p2 = verts + 3*i2;
+      float ux = p1[0]-p0[0], uy = p1[1]-p0[1], uz = p1[2]-p0[2];
+      float vx = p2[0]-p0[0], vy = p2[1]-p0[1], vz = p2[2]-p0[2];
+      float nx = uy*vz - uz*vy, ny = uz*vx - ux*vz, nz = ux*vy - uy*vx;
+      float len = sqrtf(nx*nx + ny*ny + nz*nz);
+      if (len <= 0.0f) continue;
+      float inv = 1.0f / len;
+      nx *= inv; ny *= inv; nz *= inv;
+      float d = -(nx*p0[0] + ny*p0[1] + nz*p0[2]);
+      quad_add_plane(&Q[i0], nx, ny, nz, d, 1.0f);
+      quad_add_plane(&Q[i1], nx, ny, nz, d, 1.0f);
+      quad_add_plane(&Q[i2], nx, ny, nz, d, 1.0f);
+    }
+    /* 2. per-vertex best incident edge: encode (costbits<<32 | peer) and
+     * take min. Cost as raw f32 bits (all costs >= 0 so bit order == value
+     * order); tie-break by smaller peer index. Deterministic: min over
+     * edges is order-independent. */
+    for (uint32_t v = 0; v < nv; v++) pick[v] = UINT64_MAX;
+    for (uint32_t t = 0; t < nt; t++) {
+      for (int e = 0; e < 3; e++) {
+        uint32_t a = faces[3*t + e], b = faces[3*t + (e+1)%3];
+        if (a == b) continue;
+        uint32_t u = a < b ? a : b, w = a < b ? b : a;
+        float mx = 0.5f*(verts[3*u]+verts[3*w]);
+        float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+        float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+        quad10 S;
+        for (int k = 0; k < 10; k++) S.q[k] = Q[u].q[k] + Q[w].q[k];
+        float cost = quad_eval(&S, mx, my, mz);
+        if (cost < 0.0f) cost = 0.0f;
+        if (cost > max_cost) continue;
+        uint32_t cb; memcpy(&cb, &cost, 4);
+        uint64_t enc_u = ((uint64_t)cb << 32) | w;
+        uint64_t enc_w = ((uint64_t)cb << 32) | u;
+        if (enc_u < pick[u]) pick[u] = enc_u;
+        if (enc_w < pick[w]) pick[w] = enc_w;
+      }
+    }
+    /* 3. matched pairs collapse to midpoint (u<v keeps u) */
+    for (uint32_t v = 0; v < nv; v++) remap[v] = v;
+    uint32_t collapses = 0;
+    for (uint32_t u = 0; u < nv; u++) {
+      if (pick[u] == UINT64_MAX) continue;
+      uint32_t w = (uint32_t)pick[u];
+      if (w <= u) continue;               /* handle each pair from its min end */
+      if (pick[w] == UINT64_MAX || (uint32_t)pick[w] != u) continue; /* not matched */
+      if (remap[u] != u || remap[w] != w) continue; /* already touched */
+      verts[3*u]   = 0.5f*(verts[3*u]+verts[3*w]);
+      verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+      verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+      remap[w] = u;
+      collapses++;
+    }
+    if (!collapses) break;
+    /* 4. rewrite faces, drop degenerates */
+    uint32_t out = 0;
+    for (uint32_t t = 0; t < nt; t++) {
+      uint32_t i0 = remap[faces[3*t]], i1 = remap[faces[3*t+1]], i2 = remap[faces[3*t+2]];
+      if (i0 == i1 || i1 == i2 || i0 == i2) continue;
+      faces[3*out] = i0; faces[3*out+1] = i1; faces[3*out+2] = i2;
+      out++;
+    }
+    if (out < nt) progress = 1;
+    nt = out;
+  }
+
+  /* 5. compact vertices to those referenced, preserving index order */
+  uint32_t *newidx = (uint32_t*)malloc(nv * sizeof(uint32_t));
+  memset(newidx, 0xFF, nv * sizeof(uint32_t));
+  uint32_t nnv = 0;
+  for (uint32_t t = 0; t < nt; t++)
+    for (int e = 0; e < 3; e++) {
+      uint32_t v = faces[3*t + e];
+      if (newidx[v] == 0xFFFFFFFFu) newidx[v] = 1;
+    }
+  for (uint32_t v = 0; v < nv; v++) {
+    if (newidx[v] == 1) {
+      newidx[v] = nnv;
+      verts[3*nnv] = verts[3*v]; verts[3*nnv+1] = verts[3*v+1]; verts[3*nnv+2] = verts[3*v+2];
+      nnv++;
+    } else newidx[v] = 0xFFFFFFFFu;
+  }
+  for (uint32_t t = 0; t < nt; t++)
+    for (int e = 0; e < 3; e++) faces[3*t + e] = newidx[faces[3*t + e]];
+
+  free(newidx); free(pick); free(Q); free(remap);
+  *nverts_io = nnv;
+  *ntris_io = nt;
+}
