@@ -1,0 +1,19 @@
+"""igneous_amd — MI355X-native engine for igneous's MeshTask hot path.
+
+Drop-in surface mirroring the reference package root
+(/root/reference/igneous/__init__.py): Mesher (GPU-backed zmesh.Mesher
+equivalent), LocalTaskQueue/RegisteredTask (in-process queue), the mesh
+tasks, and create_meshing_tasks. Compute runs as hand-written HIP/CDNA4
+kernels behind the C ABI in include/meshgine.h.
+"""
+from .mesher import Mesher
+from .meshes import Mesh
+from .queue import LocalTaskQueue, RegisteredTask
+from .tasks import (
+    MeshTask, MeshManifestPrefixTask, MeshManifestFilesystemTask,
+)
+from .task_creation import create_meshing_tasks
+from .volume import PrecomputedVolume
+from .lib import Bbox, Vec
+
+__version__ = "0.1.0"

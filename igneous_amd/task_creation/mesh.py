@@ -1,0 +1,142 @@
+"""create_meshing_tasks — drop-in for the reference generator at
+/root/reference/igneous/task_creation/mesh.py:158-267: same signature and
+defaults, same mesh info JSON (:197-208), same provenance record
+(:237-265), same FinelyDividedTaskIterator fan-out into MeshTask objects.
+The fan-out's consumers are GPU dispatch workers (igneous_amd.dispatch)
+instead of an SQS worker fleet.
+"""
+from __future__ import annotations
+
+from time import strftime
+from typing import Optional
+
+from ..lib import Bbox, Vec
+from ..storage import CloudFiles
+from ..tasks.mesh import MeshTask
+from ..volume import PrecomputedVolume
+from .common import FinelyDividedTaskIterator, operator_contact
+
+__all__ = ["create_meshing_tasks", "create_mesh_manifest_tasks"]
+
+
+def create_meshing_tasks(
+    layer_path: str,
+    mip: int,
+    shape=(448, 448, 448),
+    simplification: bool = True,
+    max_simplification_error: float = 40.0,
+    mesh_dir: Optional[str] = None,
+    cdn_cache: bool = False,
+    dust_threshold: Optional[int] = None,
+    object_ids=None,
+    progress: bool = False,
+    fill_missing: bool = False,
+    encoding: str = 'precomputed',
+    spatial_index: bool = True,
+    frag_path: Optional[str] = None,
+    sharded: bool = False,
+    compress: Optional[str] = 'gzip',
+    closed_dataset_edges: bool = True,
+    dust_global: bool = False,
+    fill_holes: int = 0,
+    dry_run: bool = False,
+    exclude_object_ids=[],
+):
+    shape = Vec(*shape)
+
+    assert 0 <= fill_holes <= 103, "fill_holes must be between 0 to 103 inclusive."
+
+    vol = PrecomputedVolume(layer_path, mip)
+
+    if mesh_dir is None:
+        if 'mesh' in vol.info:
+            mesh_dir = vol.info['mesh']
+        else:
+            mesh_dir = 'mesh_mip_{}_err_{}'.format(mip, max_simplification_error)
+
+    if 'mesh' not in vol.info:
+        vol.info['mesh'] = mesh_dir
+        vol.commit_info()
+
+    cf = CloudFiles(layer_path)
+    info_filename = '{}/info'.format(mesh_dir)
+    mesh_info = cf.get_json(info_filename) or {}
+    mesh_info['@type'] = 'neuroglancer_legacy_mesh'
+    mesh_info['mip'] = int(vol.mip)
+    mesh_info['chunk_size'] = shape.tolist()
+    if spatial_index:
+        mesh_info['spatial_index'] = {
+            'resolution': vol.resolution.tolist(),
+            'chunk_size': (shape * vol.resolution).tolist(),
+        }
+    cf.put_json(info_filename, mesh_info)
+
+    class MeshTaskIterator(FinelyDividedTaskIterator):
+        def task(self, shape, offset):
+            return MeshTask(
+                shape=shape.clone(),
+                offset=offset.clone(),
+                layer_path=layer_path,
+                mip=vol.mip,
+                simplification_factor=(0 if not simplification else 100),
+                max_simplification_error=max_simplification_error,
+                mesh_dir=mesh_dir,
+                cache_control=('' if cdn_cache else 'no-cache'),
+                dust_threshold=dust_threshold,
+                dust_global=bool(dust_global),
+                progress=progress,
+                object_ids=object_ids,
+                exclude_object_ids=exclude_object_ids,
+                fill_missing=fill_missing,
+                encoding=encoding,
+                spatial_index=spatial_index,
+                frag_path=frag_path,
+                sharded=sharded,
+                compress=compress,
+                closed_dataset_edges=closed_dataset_edges,
+                fill_holes=fill_holes,
+                dry_run=dry_run,
+            )
+
+        def on_finish(self):
+            vol.provenance.processing.append({
+                'method': {
+                    'task': 'MeshTask',
+                    'layer_path': layer_path,
+                    'mip': vol.mip,
+                    'shape': shape.tolist(),
+                    'simplification': simplification,
+                    'max_simplification_error': max_simplification_error,
+                    'mesh_dir': mesh_dir,
+                    'fill_missing': fill_missing,
+                    'cdn_cache': cdn_cache,
+                    'dust_threshold': dust_threshold,
+                    'encoding': encoding,
+                    'object_ids': object_ids,
+                    'exclude_object_ids': exclude_object_ids,
+                    'spatial_index': spatial_index,
+                    'frag_path': frag_path,
+                    'sharded': sharded,
+                    'compress': compress,
+                    'closed_dataset_edges': closed_dataset_edges,
+                    'dust_global': bool(dust_global),
+                    'fill_holes': int(fill_holes),
+                    'dry_run': bool(dry_run),
+                },
+                'by': operator_contact(),
+                'date': strftime('%Y-%m-%d %H:%M %Z'),
+            })
+            vol.commit_provenance()
+
+    return MeshTaskIterator(vol.mip_bounds(mip), shape)
+
+
+def create_mesh_manifest_tasks(layer_path: str, magnitude: int = 3,
+                               mesh_dir: Optional[str] = None):
+    """Mirror of the reference's manifest fan-out
+    (task_creation/mesh.py:54-89). file:// protocol -> one filesystem task."""
+    from functools import partial
+    from ..tasks.mesh import MeshManifestFilesystemTask
+    assert int(magnitude) == magnitude and magnitude > 0
+    return [partial(MeshManifestFilesystemTask,
+                    layer_path=layer_path, mesh_dir=mesh_dir)]

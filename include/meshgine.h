@@ -1,0 +1,137 @@
+/* meshgine — C ABI of the MI355X-native per-chunk meshing engine.
+ *
+ * This is the drop-in boundary under igneous's MeshTask hot path: it
+ * replaces the work the reference delegates to the zmesh C++/Cython
+ * extension, entry point for entry point:
+ *
+ *   mg_init            — replaces zmesh.Mesher(resolution) construction,
+ *                        /root/reference/igneous/tasks/mesh/mesh.py:151
+ *                        (device context instead of a CPU mesher object;
+ *                        the anisotropic resolution moves to mg_mesh_chunk
+ *                        because the engine is stateless per chunk)
+ *   mg_mesh_chunk      — replaces the pair
+ *                        Mesher.mesh(data, preserve_order=False)
+ *                        (mesh.py:245: one pass over the whole F-order
+ *                        chunk, label-vs-rest binary surfaces for every
+ *                        non-zero label) and the per-label loop
+ *                        Mesher.ids() / Mesher.get(id, reduction_factor,
+ *                        max_error, voxel_centered=True)
+ *                        (mesh.py:374-381: per-label quadric edge-collapse
+ *                        simplification + welded vertex/face extraction).
+ *                        One call produces every label's mesh.
+ *   mg_meshset_free    — replaces Python GC of zmesh.Mesh objects.
+ *   mg_last_error      — error text for the last failed call on this ctx.
+ *
+ * The caller (igneous_amd.tasks.MeshTask, via ctypes) keeps everything the
+ * reference keeps in Python: download, padding, dust/remap/object-id
+ * masking, the shift of vertices into global nm coordinates
+ * (mesh.py:434-435), precomputed encoding (mesh.py:448) and upload.
+ *
+ * Vertices returned are CHUNK-LOCAL nm coordinates (float32), exactly as
+ * zmesh returns them to MeshTask before the Python-side offset is applied.
+ *
+ * Threading: one mg_ctx per device; calls on distinct ctxs may run
+ * concurrently. Calls on one ctx serialize internally. No global state
+ * besides the HIP runtime. Errors: non-zero int return + mg_last_error.
+ * Ownership: the engine allocates all output storage; the caller frees a
+ * result with mg_meshset_free.
+ */
+#ifndef MESHGINE_H
+#define MESHGINE_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct mg_ctx mg_ctx;
+
+/* dtype codes for the label volume */
+#define MG_U32 0
+#define MG_U64 1
+
+/* flags for mg_mesh_chunk */
+#define MG_FLAG_NONE        0u
+#define MG_FLAG_DEVICE_ONLY 1u  /* run all kernels, skip the D2H extract
+                                   (bench: job complete with outputs in
+                                   HBM; sizes still reported) */
+
+/* One label's mesh. verts = 3*nverts float32 chunk-local nm (x,y,z);
+ * faces = 3*ntris uint32 indices into verts. Pointers alias the meshset's
+ * internal storage; freed by mg_meshset_free. */
+typedef struct {
+  uint64_t  label;
+  uint32_t  nverts;
+  uint32_t  ntris;
+  float    *verts;
+  uint32_t *faces;
+} mg_mesh;
+
+typedef struct {
+  uint32_t  nmeshes;
+  mg_mesh  *meshes;      /* sorted by ascending label id */
+  /* opaque storage follows */
+} mg_meshset;
+
+/* Per-call kernel timing/stats, HIP-event measured on the engine stream.
+ * All times in milliseconds for the LAST mg_mesh_chunk on this ctx. */
+typedef struct {
+  double ms_h2d;        /* label volume upload */
+  double ms_count;      /* pass 1: per-segment triangle count */
+  double ms_scan;       /* segment offset scan (+ total readback) */
+  double ms_emit;       /* pass 2: triangle emit (keys + label ids) */
+  double ms_partition;  /* stable partition of triangles by label */
+  double ms_weld;       /* vertex weld + index build + vertex writeout */
+  double ms_simplify;   /* per-label quadric collapse (0 if disabled) */
+  double ms_d2h;        /* result download (0 under MG_FLAG_DEVICE_ONLY) */
+  double ms_total;      /* whole call, device-side */
+  uint64_t total_tris;  /* before simplification */
+  uint64_t total_verts; /* before simplification */
+  uint64_t n_labels;
+  uint64_t bytes_read_algorithmic; /* sx*sy*sz * sizeof(label) */
+} mg_stats;
+
+/* Create a context on HIP device device_id (0-based). Returns NULL on
+ * failure (no device, no HIP). */
+mg_ctx  *mg_init(int device_id);
+void     mg_destroy(mg_ctx *ctx);
+
+/* Mesh every non-zero label of an F-order label volume.
+ *   labels        host pointer, sx*sy*sz elements, F-order (x fastest)
+ *   sx,sy,sz      dimensions INCLUDING any overlap padding (<= 2047 each)
+ *   dtype         MG_U32 | MG_U64
+ *   rx,ry,rz      resolution in nm per voxel (anisotropy, mesh.py:151)
+ *   reduction_factor  target triangle reduction (0/1 = no simplification;
+ *                     mesh.py:376-381 'reduction_factor')
+ *   max_error     max simplification error in nm (mesh.py 'max_error')
+ *   voxel_centered nonzero: voxel centers at integer coordinates
+ *                  (mesh.py:380 voxel_centered=True)
+ *   flags         MG_FLAG_*
+ *   out           receives the meshset (caller frees)
+ * Returns 0 on success. */
+int mg_mesh_chunk(mg_ctx *ctx, const void *labels,
+                  int sx, int sy, int sz, int dtype,
+                  float rx, float ry, float rz,
+                  uint32_t reduction_factor, float max_error,
+                  int voxel_centered, uint32_t flags,
+                  mg_meshset **out);
+
+void mg_meshset_free(mg_meshset *ms);
+
+/* Stats of the last mg_mesh_chunk on this ctx. Returns 0 on success. */
+int mg_get_stats(mg_ctx *ctx, mg_stats *out);
+
+/* Error text of the last failed call on this ctx (thread-local static
+ * lifetime; valid until the next call on the ctx). */
+const char *mg_last_error(mg_ctx *ctx);
+
+/* Engine/device identification (for logging + the bench line). */
+int mg_device_count(void);
+const char *mg_version(void);
+
+#ifdef __cplusplus
+}
+#endif
+
+#endif /* MESHGINE_H */
