@@ -1,0 +1,222 @@
+#!/usr/bin/env python3
+"""Benchmark of the meshgine hot path (BASELINE.json metric): Mvoxels/s
+meshed on the 512^3 uint64 ~50k-label synthetic segmentation chunk
+(configs[2] — the configuration the metric is quoted on), one process per
+GPU, weak scaling (each rank meshes its own chunk per step).
+
+    python bench.py --gpus N --steps K --warmup W
+
+For N>1 the driver launches this under torch.distributed.run with one rank
+per GPU; ranks synchronize with a barrier + torch.cuda.synchronize around
+the timed region and report the MAX elapsed over ranks. Rank 0 prints ONE
+JSON line.
+
+A "step" = one mg_mesh_chunk over the full chunk: marching-cubes count +
+emit, label partition, vertex weld — outputs complete in HBM
+(MG_FLAG_DEVICE_ONLY; the PCIe-inclusive host-extract rate is reported in
+DESIGN.md, never as `value`). Inputs are resident in HBM when the timed
+region starts (staged in warmup; MG_FLAG_SKIP_H2D).
+
+cpu_baseline: this repo's CPU oracle (oracle/, kind "port" — zmesh is not
+installable offline) timed on the host's cores over a bounded
+sample of the same workload, igneous's own process-per-chunk parallelism
+model (cli.py:915-933).
+"""
+import argparse
+import json
+import multiprocessing as mp
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+WORKLOAD = "512^3 uint64 Voronoi chunk, ~50k labels, seed=303 (BASELINE configs[2])"
+SHAPE = (512, 512, 512)
+K_SEEDS = 50000
+SEED = 303
+RESOLUTION = (16.0, 16.0, 40.0)
+PEAK_HBM = 8.0e12  # B/s, MI355X spec (MI355X_MICROARCH.md)
+
+
+def _oracle_baseline_worker(args):
+    sub, res = args
+    import oracle  # oracle/ is on sys.path (checker/baseline leg only)
+    t0 = time.perf_counter()
+    oracle.mesh_chunk(sub, resolution=res)
+    return time.perf_counter() - t0, sub.size
+
+
+def cpu_baseline(data: np.ndarray, budget_s: float = 20.0) -> dict:
+    """Time the CPU oracle on a bounded sample of the same chunk:
+    process-per-subchunk over all host cores (the reference's --parallel
+    worker model), ~budget_s of wall time."""
+    sys.path.insert(0, os.path.join(REPO, "oracle"))
+    import oracle
+    oracle.build()
+    cores = os.cpu_count() or 1
+    # sample: 128^3 (+1 overlap) subchunks of the real chunk
+    subs = []
+    n = SHAPE[0] // 128
+    for z in range(n):
+        for y in range(n):
+            for x in range(n):
+                sub = np.asfortranarray(
+                    data[x * 128:x * 128 + 129,
+                         y * 128:y * 128 + 129,
+                         z * 128:z * 128 + 129])
+                subs.append(sub)
+    # calibrate with one subchunk, single process
+    t_one, vox_one = _oracle_baseline_worker((subs[0], RESOLUTION))
+    per_chunk = t_one
+    target_chunks = max(cores, min(len(subs),
+                                   int(budget_s / per_chunk * cores)))
+    chosen = subs[:target_chunks]
+    t0 = time.perf_counter()
+    with mp.get_context("fork").Pool(cores) as pool:
+        pool.map(_oracle_baseline_worker,
+                 [(s, RESOLUTION) for s in chosen])
+    elapsed = time.perf_counter() - t0
+    vox = sum(s.size for s in chosen)
+    return {
+        "value": round(vox / elapsed / 1e6, 2),
+        "unit": "Mvox/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": (f"{len(chosen)} x 129^3 subchunks of the same 512^3 "
+                   f"chunk, {cores}-process pool, {elapsed:.1f}s"),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(args.gpus, world)
+
+    import torch
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend=backend)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
+
+    from igneous_amd.engine import Engine
+    from igneous_amd.synth import voronoi_labels
+
+    # per-rank chunk: same stats, different seed per rank (weak scaling)
+    data = voronoi_labels(SHAPE, K_SEEDS, SEED + rank, dtype=np.uint64)
+    eng = Engine.get(local_rank)
+
+    def step(skip_h2d=True):
+        eng.mesh_chunk(data, resolution=RESOLUTION, reduction_factor=0,
+                       device_only=True, skip_h2d=skip_h2d)
+
+    # warmup (first call stages the labels into HBM)
+    step(skip_h2d=False)
+    for _ in range(max(0, args.warmup - 1)):
+        step()
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    stats = eng.stats()
+    nvox = int(np.prod(SHAPE))
+    total_vox = nvox * args.steps * world
+    mvox_s = total_vox / elapsed / 1e6
+
+    if rank != 0:
+        if dist is not None:
+            dist.destroy_process_group()
+        return
+
+    # roofline for the dominant kernel (k_emit: the marching-cubes scan +
+    # triangle emit over the whole volume). Algorithmic bytes per launch =
+    # one read of each label voxel.
+    algo_bytes = float(stats["bytes_read_algorithmic"])
+    ms_emit = stats["ms_emit"]
+    achieved = algo_bytes / (ms_emit * 1e-3) if ms_emit > 0 else 0.0
+    traffic = None
+    tfile = os.path.join(REPO, "profiles", "roofline_traffic.json")
+    if os.path.exists(tfile):
+        try:
+            tj = json.load(open(tfile))
+            traffic = tj.get("k_emit_bytes_per_launch")
+        except Exception:
+            traffic = None
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(achieved / 1e9, 1),
+        "peak": round(PEAK_HBM / 1e9, 1),
+        "unit": "GB/s",
+        "frac": round(achieved / PEAK_HBM, 4),
+        "traffic": traffic,
+    }
+
+    cpu = None
+    if not args.no_cpu_baseline and rank == 0 and world == 1:
+        cpu = cpu_baseline(data)
+
+    line = {
+        "metric": "Mvoxels/s meshed (512^3 uint64 seg chunk)",
+        "value": round(mvox_s, 1),
+        "unit": "Mvox/s",
+        "n_gpus": n_gpus if world == 1 else world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1e3, 2),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "u64",
+        "data": "synthetic",
+        "config": {
+            "workload": WORKLOAD,
+            "chunk": list(SHAPE),
+            "labels": K_SEEDS,
+            "resolution_nm": list(RESOLUTION),
+            "simplification": 0,
+            "n_labels_meshed": int(stats["n_labels"]),
+            "total_tris": int(stats["total_tris"]),
+            "kernel_ms": {k: round(stats[k], 3) for k in (
+                "ms_count", "ms_scan", "ms_emit", "ms_partition",
+                "ms_weld", "ms_total")},
+        },
+        "roofline": roofline,
+        "cpu_baseline": cpu,
+    }
+    print(json.dumps(line), flush=True)
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,974 @@
+// meshgine.hip — MI355X (gfx950/CDNA4) per-chunk meshing engine.
+//
+// Implements the C ABI of include/meshgine.h: the compute the reference
+// delegates to zmesh's CPU C++ (multi-label marching cubes at
+// /root/reference/igneous/tasks/mesh/mesh.py:245, per-label welded extract
+// + quadric simplify at mesh.py:374-381) as hand-written HIP kernels.
+//
+// Canonical contract (bit-exact with oracle/mc_oracle.c — see DESIGN.md):
+//   cells in global F-order; per cell distinct non-zero labels in
+//   first-seen corner order; triangles from mc_table.h in table order;
+//   per-label first-seen vertex welding; vertex position
+//   (0.5f*k + shift)*res in f32.
+//
+// Device pipeline (one mg_mesh_chunk call):
+//   H2D labels
+//   [1] k_count        wave-per-64-cell-segment triangle counts
+//                      (+ label hash build)                    -> segcnt
+//   [2] scan           rocPRIM exclusive scan                  -> segoff, T
+//   [3] k_emit         recompute counts, wave prefix, write
+//                      (label_idx, 3x weld keys) per triangle in canonical
+//                      global order
+//   [4] partition      rocPRIM radix-sort by label_idx (stable) + gather
+//   [5] weld           hash insert w/ atomicMin(first-pos), flag first
+//                      occurrences, scan -> vertex ids, emit vertices,
+//                      rewrite faces as per-label indices
+//   [6] (optional) per-label quadric simplification
+//   [7] D2H slices + meshset assembly (host pinned memory)
+//
+// Determinism: every kernel's output is a pure function of its inputs —
+// atomics are only used for first-position minima (order-free), hash slot
+// claims (value-keyed) and the label counter (affects internal ids only;
+// triangle partition order is by internal id but per-label content and the
+// final label-sorted meshset are invariant).
+
+#include <atomic>
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <mutex>
+#include <string>
+#include <vector>
+#include <algorithm>
+
+#include <hip/hip_runtime.h>
+#include <rocprim/rocprim.hpp>
+
+#define MC_TABLE_QUAL __device__ static const
+#include "mc_table.h"
+
+#include "../../include/meshgine.h"
+
+// ---------------------------------------------------------------------------
+// small utilities
+
+#define WAVE 64
+
+static inline uint64_t next_pow2_u64(uint64_t x) {
+  if (x < 2) return 2;
+  return 1ull << (64 - __builtin_clzll(x - 1));
+}
+
+__device__ __forceinline__ uint64_t mix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+// wave-wide inclusive prefix sum (64 lanes)
+__device__ __forceinline__ uint32_t wave_incl_scan(uint32_t x, int lane) {
+  for (int d = 1; d < WAVE; d <<= 1) {
+    uint32_t y = __shfl_up(x, d, WAVE);
+    if (lane >= d) x += y;
+  }
+  return x;
+}
+
+// ---------------------------------------------------------------------------
+// label hash: raw label value -> dense internal id (build in count pass)
+
+struct LabelHash {
+  uint64_t *keys;   // 0 = empty (label 0 never inserted)
+  uint32_t *vals;
+  uint32_t *counter;   // next id
+  uint32_t *overflow;  // error flag
+  uint64_t nslots;     // power of two
+};
+
+__device__ __forceinline__ void label_insert(LabelHash h, uint64_t label) {
+  uint64_t slot = mix64(label) & (h.nslots - 1);
+  for (uint64_t probe = 0; probe < h.nslots; ++probe) {
+    uint64_t cur = h.keys[slot];
+    if (cur == label) return;
+    if (cur == 0) {
+      uint64_t prev = atomicCAS((unsigned long long *)&h.keys[slot], 0ull,
+                                (unsigned long long)label);
+      if (prev == 0) {
+        h.vals[slot] = atomicAdd(h.counter, 1u);
+        return;
+      }
+      if (prev == label) return;
+    }
+    slot = (slot + 1) & (h.nslots - 1);
+  }
+  atomicExch(h.overflow, 1u);
+}
+
+__device__ __forceinline__ uint32_t label_lookup(LabelHash h, uint64_t label) {
+  uint64_t slot = mix64(label) & (h.nslots - 1);
+  for (;;) {
+    uint64_t cur = h.keys[slot];
+    if (cur == label) {
+      // id may have been published by another wave after the CAS; spin-free
+      // read is safe because k_count completed before any lookup kernel.
+      return h.vals[slot];
+    }
+    slot = (slot + 1) & (h.nslots - 1);
+  }
+}
+
+// invert: label id -> label value (fill after count)
+__global__ void k_label_values(LabelHash h, uint64_t *values, uint64_t nslots) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= nslots) return;
+  uint64_t k = h.keys[i];
+  if (k != 0) values[h.vals[i]] = k;
+}
+
+// ---------------------------------------------------------------------------
+// per-cell triangle enumeration, shared by count and emit
+//
+// Segment = up to 64 consecutive cells of one x-row (aligned to 64 cells).
+// One wave per segment; lane = cell index within the segment.
+
+template <typename T>
+__device__ __forceinline__ void load_corners(const T *__restrict__ labels,
+                                             int64_t sx, int64_t sxy,
+                                             int64_t cx, int64_t cy, int64_t cz,
+                                             T c[8]) {
+  const T *p = labels + cx + cy * sx + cz * sxy;
+  c[0] = p[0];            c[1] = p[1];
+  c[2] = p[sx];           c[3] = p[sx + 1];
+  c[4] = p[sxy];          c[5] = p[sxy + 1];
+  c[6] = p[sxy + sx];     c[7] = p[sxy + sx + 1];
+}
+
+template <typename T>
+__device__ __forceinline__ uint32_t cell_tri_count(const T c[8]) {
+  if (c[0] == c[1] && c[0] == c[2] && c[0] == c[3] && c[0] == c[4] &&
+      c[0] == c[5] && c[0] == c[6] && c[0] == c[7])
+    return 0;
+  uint32_t total = 0;
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    T L = c[i];
+    if (L == 0) continue;
+    bool seen = false;
+    for (int j = 0; j < i; ++j) seen |= (c[j] == L);
+    if (seen) continue;
+    unsigned mask = 0;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) mask |= (c[j] == L) ? (1u << j) : 0u;
+    total += MC_TRI_COUNT[mask];
+  }
+  return total;
+}
+
+struct GridDims {
+  int64_t sx, sy, sz;     // voxel dims
+  int64_t ncx, ncy, ncz;  // cell dims
+  int64_t nsegx;          // segments per row
+  int64_t nseg;           // total segments
+};
+
+// [1] count: one wave per segment; build label hash; write per-segment count
+template <typename T>
+__global__ void k_count(const T *__restrict__ labels, GridDims g,
+                        uint32_t *__restrict__ segcnt, LabelHash lh) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave_in_blk = threadIdx.x / WAVE;
+  const int waves_per_blk = blockDim.x / WAVE;
+  const int64_t sxy = g.sx * g.sy;
+  for (int64_t seg = (int64_t)blockIdx.x * waves_per_blk + wave_in_blk;
+       seg < g.nseg;
+       seg += (int64_t)gridDim.x * waves_per_blk) {
+    const int64_t row = seg / g.nsegx;
+    const int64_t segx = seg - row * g.nsegx;
+    const int64_t cy = row % g.ncy;
+    const int64_t cz = row / g.ncy;
+    const int64_t cx = segx * WAVE + lane;
+    uint32_t cnt = 0;
+    if (cx < g.ncx) {
+      T c[8];
+      load_corners(labels, g.sx, sxy, cx, cy, cz, c);
+      if (!(c[0] == c[1] && c[0] == c[2] && c[0] == c[3] && c[0] == c[4] &&
+            c[0] == c[5] && c[0] == c[6] && c[0] == c[7])) {
+        #pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          T L = c[i];
+          if (L == 0) continue;
+          bool seen = false;
+          for (int j = 0; j < i; ++j) seen |= (c[j] == L);
+          if (seen) continue;
+          unsigned mask = 0;
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) mask |= (c[j] == L) ? (1u << j) : 0u;
+          uint32_t nt = MC_TRI_COUNT[mask];
+          if (nt) {
+            cnt += nt;
+            label_insert(lh, (uint64_t)L);
+          }
+        }
+      }
+    }
+    // wave reduce
+    uint32_t incl = wave_incl_scan(cnt, lane);
+    if (lane == WAVE - 1) segcnt[seg] = incl;
+  }
+}
+
+// [3] emit: recompute, wave prefix, write triangle records in canonical order
+template <typename T>
+__global__ void k_emit(const T *__restrict__ labels, GridDims g,
+                       const uint32_t *__restrict__ segoff, LabelHash lh,
+                       uint32_t *__restrict__ tri_label,
+                       uint64_t *__restrict__ tri_keys) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave_in_blk = threadIdx.x / WAVE;
+  const int waves_per_blk = blockDim.x / WAVE;
+  const int64_t sxy = g.sx * g.sy;
+  for (int64_t seg = (int64_t)blockIdx.x * waves_per_blk + wave_in_blk;
+       seg < g.nseg;
+       seg += (int64_t)gridDim.x * waves_per_blk) {
+    const int64_t row = seg / g.nsegx;
+    const int64_t segx = seg - row * g.nsegx;
+    const int64_t cy = row % g.ncy;
+    const int64_t cz = row / g.ncy;
+    const int64_t cx = segx * WAVE + lane;
+    T c[8];
+    uint32_t cnt = 0;
+    bool active = false;
+    if (cx < g.ncx) {
+      load_corners(labels, g.sx, sxy, cx, cy, cz, c);
+      cnt = cell_tri_count(c);
+      active = cnt > 0;
+    }
+    uint32_t incl = wave_incl_scan(cnt, lane);
+    uint32_t base = segoff[seg] + incl - cnt;
+    if (active) {
+      uint64_t cellkey = ((uint64_t)(2 * cz) << 24) |
+                         ((uint64_t)(2 * cy) << 12) | (uint64_t)(2 * cx);
+      uint32_t pos = base;
+      #pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        T L = c[i];
+        if (L == 0) continue;
+        bool seen = false;
+        for (int j = 0; j < i; ++j) seen |= (c[j] == L);
+        if (seen) continue;
+        unsigned mask = 0;
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) mask |= (c[j] == L) ? (1u << j) : 0u;
+        uint32_t nt = MC_TRI_COUNT[mask];
+        if (!nt) continue;
+        uint32_t lid = label_lookup(lh, (uint64_t)L);
+        uint64_t lid_hi = (uint64_t)lid << 36;
+        const signed char *tt = MC_TRI_TABLE[mask];
+        for (uint32_t t = 0; t < nt; ++t) {
+          tri_label[pos] = lid;
+          #pragma unroll
+          for (int v = 0; v < 3; ++v) {
+            int e = tt[3 * t + v];
+            tri_keys[3 * (uint64_t)pos + v] =
+                lid_hi | (cellkey + MC_EDGE_KEYOFF[e]);
+          }
+          ++pos;
+        }
+      }
+    }
+  }
+}
+
+// [4b] gather triangle keys into label-partitioned order
+__global__ void k_gather_keys(const uint64_t *__restrict__ tri_keys,
+                              const uint32_t *__restrict__ order,
+                              uint64_t *__restrict__ keys_sorted,
+                              uint64_t ntris) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= ntris) return;
+  uint64_t src = 3ull * order[i];
+  uint64_t dst = 3ull * i;
+  keys_sorted[dst + 0] = tri_keys[src + 0];
+  keys_sorted[dst + 1] = tri_keys[src + 1];
+  keys_sorted[dst + 2] = tri_keys[src + 2];
+}
+
+// [4c] per-label triangle ranges (labels sorted, every id present)
+__global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
+                               uint32_t *__restrict__ tri_off,
+                               uint64_t ntris, uint32_t nlabels) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= ntris) return;
+  if (i == 0) {
+    tri_off[lab_sorted[0]] = 0;  // first label's range starts at 0
+    tri_off[nlabels] = (uint32_t)ntris;
+  } else if (lab_sorted[i] != lab_sorted[i - 1]) {
+    tri_off[lab_sorted[i]] = (uint32_t)i;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// welding
+
+struct WeldHash {
+  uint64_t *keys;     // 0 = empty (coord bits never all-zero)
+  uint32_t *minp;     // first corner position
+  uint32_t *vtx;      // assigned vertex id
+  uint32_t *overflow; // error flag (bounded probing: no device-side hang)
+  uint64_t nslots;    // power of two
+};
+
+__device__ __forceinline__ uint64_t weld_slot(WeldHash h, uint64_t key) {
+  uint64_t slot = mix64(key) & (h.nslots - 1);
+  for (uint64_t probe = 0; probe < h.nslots; ++probe) {
+    uint64_t cur = h.keys[slot];
+    if (cur == key) return slot;
+    if (cur == 0) {
+      uint64_t prev = atomicCAS((unsigned long long *)&h.keys[slot], 0ull,
+                                (unsigned long long)key);
+      if (prev == 0 || prev == key) return slot;
+    }
+    slot = (slot + 1) & (h.nslots - 1);
+  }
+  atomicExch(h.overflow, 1u);
+  return 0;
+}
+
+__device__ __forceinline__ uint64_t weld_find(WeldHash h, uint64_t key) {
+  uint64_t slot = mix64(key) & (h.nslots - 1);
+  while (h.keys[slot] != key) slot = (slot + 1) & (h.nslots - 1);
+  return slot;
+}
+
+// [5a] insert all corners; record first (minimum) stream position per key
+__global__ void k_weld_insert(const uint64_t *__restrict__ keys_sorted,
+                              WeldHash h, uint64_t ncorners) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= ncorners) return;
+  uint64_t slot = weld_slot(h, keys_sorted[i]);
+  atomicMin(&h.minp[slot], (uint32_t)i);
+}
+
+// [5b] flag first occurrences
+__global__ void k_weld_flags(const uint64_t *__restrict__ keys_sorted,
+                             const WeldHash h, uint8_t *__restrict__ flags,
+                             uint64_t ncorners) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= ncorners) return;
+  uint64_t slot = weld_find(h, keys_sorted[i]);
+  flags[i] = (h.minp[slot] == (uint32_t)i) ? 1 : 0;
+}
+
+// [5d] first occurrences: record vertex id in hash, write vertex position
+__global__ void k_weld_verts(const uint64_t *__restrict__ keys_sorted,
+                             const uint8_t *__restrict__ flags,
+                             const uint32_t *__restrict__ vtx_scan,
+                             WeldHash h, float *__restrict__ verts,
+                             float rx, float ry, float rz, float shift,
+                             uint64_t ncorners) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= ncorners) return;
+  if (!flags[i]) return;
+  uint64_t key = keys_sorted[i];
+  uint64_t slot = weld_find(h, key);
+  uint32_t v = vtx_scan[i];
+  h.vtx[slot] = v;
+  float dx = (float)(uint32_t)(key & 0xFFF);
+  float dy = (float)(uint32_t)((key >> 12) & 0xFFF);
+  float dz = (float)(uint32_t)((key >> 24) & 0xFFF);
+  verts[3ull * v + 0] = (0.5f * dx + shift) * rx;
+  verts[3ull * v + 1] = (0.5f * dy + shift) * ry;
+  verts[3ull * v + 2] = (0.5f * dz + shift) * rz;
+}
+
+// [5e] per-label vertex bases: vbase[l] = vtx_scan at the label's first corner
+__global__ void k_vbase(const uint32_t *__restrict__ tri_off,
+                        const uint32_t *__restrict__ vtx_scan,
+                        uint32_t *__restrict__ vbase,
+                        uint32_t nlabels, uint64_t total_verts) {
+  uint32_t l = blockIdx.x * blockDim.x + threadIdx.x;
+  if (l > nlabels) return;
+  if (l == nlabels) vbase[l] = (uint32_t)total_verts;
+  else vbase[l] = vtx_scan[3ull * tri_off[l]];
+}
+
+// [5f] faces: per-label local vertex indices
+__global__ void k_faces(const uint64_t *__restrict__ keys_sorted,
+                        const WeldHash h,
+                        const uint32_t *__restrict__ vbase,
+                        uint32_t *__restrict__ faces,
+                        uint64_t ncorners) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= ncorners) return;
+  uint64_t key = keys_sorted[i];
+  uint64_t slot = weld_find(h, key);
+  uint32_t lid = (uint32_t)(key >> 36);
+  faces[i] = h.vtx[slot] - vbase[lid];
+}
+
+__global__ void k_iota(uint32_t *p, uint64_t n) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = (uint32_t)i;
+}
+
+// ---------------------------------------------------------------------------
+// context / host side
+
+struct DevBuf {
+  void *ptr = nullptr;
+  size_t cap = 0;
+};
+
+struct mg_ctx {
+  int device = 0;
+  hipStream_t stream = nullptr;
+  std::mutex lock;
+  std::string err;
+  mg_stats stats = {};
+  // device buffers (grow-only cache)
+  DevBuf labels, segcnt, segoff, scan_tmp,
+      lh_keys, lh_vals, lh_misc,
+      tri_label, tri_label_alt, order, order_alt, tri_keys, keys_sorted,
+      tri_off, sort_tmp,
+      wh_keys, wh_minp, wh_vtx, flags, vtx_scan, verts, faces, vbase,
+      label_values, small;
+  uint64_t lh_slots = 1ull << 20;
+  hipEvent_t ev[16] = {};
+};
+
+static thread_local std::string g_err;  // for ctx==NULL failures
+
+#define SET_ERR(ctx, ...)                                     \
+  do {                                                        \
+    char _buf[512];                                           \
+    snprintf(_buf, sizeof(_buf), __VA_ARGS__);                \
+    if (ctx) (ctx)->err = _buf; else g_err = _buf;            \
+  } while (0)
+
+#define HIP_TRY(ctx, call, retcode)                           \
+  do {                                                        \
+    hipError_t _e = (call);                                   \
+    if (_e != hipSuccess) {                                   \
+      SET_ERR(ctx, "%s failed: %s (%s:%d)", #call,            \
+              hipGetErrorString(_e), __FILE__, __LINE__);     \
+      return retcode;                                         \
+    }                                                         \
+  } while (0)
+
+static int ensure(mg_ctx *c, DevBuf &b, size_t bytes) {
+  if (b.cap >= bytes) return 0;
+  if (b.ptr) (void)hipFree(b.ptr);
+  b.ptr = nullptr;
+  b.cap = 0;
+  size_t want = bytes + bytes / 4;  // 25% headroom to damp re-allocs
+  hipError_t e = hipMalloc(&b.ptr, want);
+  if (e != hipSuccess) {
+    e = hipMalloc(&b.ptr, bytes);  // retry exact
+    if (e != hipSuccess) {
+      SET_ERR(c, "hipMalloc(%zu) failed: %s", bytes, hipGetErrorString(e));
+      return 1;
+    }
+    b.cap = bytes;
+    return 0;
+  }
+  b.cap = want;
+  return 0;
+}
+
+extern "C" {
+
+const char *mg_version(void) { return "meshgine 0.1.0 (gfx950)"; }
+
+int mg_device_count(void) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+mg_ctx *mg_init(int device_id) {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess || device_id >= n || n == 0) {
+    return nullptr;
+  }
+  if (hipSetDevice(device_id) != hipSuccess) return nullptr;
+  mg_ctx *c = new mg_ctx();
+  c->device = device_id;
+  if (hipStreamCreateWithFlags(&c->stream, hipStreamNonBlocking) !=
+      hipSuccess) {
+    delete c;
+    return nullptr;
+  }
+  for (auto &e : c->ev)
+    if (hipEventCreate(&e) != hipSuccess) { delete c; return nullptr; }
+  return c;
+}
+
+void mg_destroy(mg_ctx *c) {
+  if (!c) return;
+  (void)hipSetDevice(c->device);
+  for (auto *b : {&c->labels, &c->segcnt, &c->segoff, &c->scan_tmp,
+                  &c->lh_keys, &c->lh_vals, &c->lh_misc, &c->tri_label,
+                  &c->tri_label_alt, &c->order, &c->order_alt, &c->tri_keys,
+                  &c->keys_sorted, &c->tri_off, &c->sort_tmp, &c->wh_keys,
+                  &c->wh_minp, &c->wh_vtx, &c->flags, &c->vtx_scan,
+                  &c->verts, &c->faces, &c->vbase, &c->label_values,
+                  &c->small}) {
+    if (b->ptr) (void)hipFree(b->ptr);
+  }
+  for (auto &e : c->ev) if (e) (void)hipEventDestroy(e);
+  if (c->stream) (void)hipStreamDestroy(c->stream);
+  delete c;
+}
+
+const char *mg_last_error(mg_ctx *c) {
+  return c ? c->err.c_str() : g_err.c_str();
+}
+
+int mg_get_stats(mg_ctx *c, mg_stats *out) {
+  if (!c || !out) return 1;
+  *out = c->stats;
+  return 0;
+}
+
+void mg_meshset_free(mg_meshset *ms) {
+  if (!ms) return;
+  // layout: [mg_meshset][mg_mesh array][host buffer pointers]
+  void **bufs = (void **)((char *)ms + sizeof(mg_meshset) +
+                          sizeof(mg_mesh) * ms->nmeshes);
+  for (int i = 0; i < 2; ++i)
+    if (bufs[i]) (void)hipHostFree(bufs[i]);
+  free(ms);
+}
+
+// ---------------------------------------------------------------------------
+
+static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
+                           int sx, int sy, int sz, int dtype,
+                           float rx, float ry, float rz,
+                           uint32_t reduction_factor, float max_error,
+                           int voxel_centered, uint32_t flags_,
+                           mg_meshset **out);
+
+int mg_mesh_chunk(mg_ctx *c, const void *labels, int sx, int sy, int sz,
+                  int dtype, float rx, float ry, float rz,
+                  uint32_t reduction_factor, float max_error,
+                  int voxel_centered, uint32_t flags, mg_meshset **out) {
+  if (!c) { SET_ERR(c, "null ctx"); return 1; }
+  std::lock_guard<std::mutex> g(c->lock);
+  c->err.clear();
+  if (!labels || !out) { SET_ERR(c, "null argument"); return 1; }
+  if (sx < 1 || sy < 1 || sz < 1 || sx > 2047 || sy > 2047 || sz > 2047) {
+    SET_ERR(c, "dims out of range (1..2047): %d %d %d", sx, sy, sz);
+    return 2;
+  }
+  if (dtype != MG_U32 && dtype != MG_U64) {
+    SET_ERR(c, "bad dtype %d", dtype);
+    return 3;
+  }
+  HIP_TRY(c, hipSetDevice(c->device), 4);
+  return mesh_chunk_impl(c, labels, sx, sy, sz, dtype, rx, ry, rz,
+                         reduction_factor, max_error, voxel_centered,
+                         flags, out);
+}
+
+}  // extern "C"
+
+template <typename T>
+static int run_count_emit(mg_ctx *c, const T *d_labels, const GridDims &g,
+                          LabelHash lh, uint32_t *d_segcnt,
+                          uint32_t *d_segoff, uint64_t *p_total,
+                          uint32_t *p_nlabels);
+
+static double ev_ms(mg_ctx *c, int a, int b) {
+  float ms = 0.f;
+  if (hipEventElapsedTime(&ms, c->ev[a], c->ev[b]) != hipSuccess) return 0.0;
+  return (double)ms;
+}
+
+static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
+                           int sx, int sy, int sz, int dtype,
+                           float rx, float ry, float rz,
+                           uint32_t reduction_factor, float max_error,
+                           int voxel_centered, uint32_t flags_,
+                           mg_meshset **out) {
+  const size_t esize = (dtype == MG_U64) ? 8 : 4;
+  const uint64_t nvox = (uint64_t)sx * sy * sz;
+  GridDims g;
+  g.sx = sx; g.sy = sy; g.sz = sz;
+  g.ncx = sx > 1 ? sx - 1 : 0;
+  g.ncy = sy > 1 ? sy - 1 : 0;
+  g.ncz = sz > 1 ? sz - 1 : 0;
+  g.nsegx = (g.ncx + WAVE - 1) / WAVE;
+  g.nseg = g.nsegx * g.ncy * g.ncz;
+
+  memset(&c->stats, 0, sizeof(c->stats));
+  c->stats.bytes_read_algorithmic = nvox * esize;
+
+  hipStream_t s = c->stream;
+
+  // trivial empty-cell-grid case
+  if (g.nseg == 0) {
+    mg_meshset *ms = (mg_meshset *)calloc(
+        1, sizeof(mg_meshset) + 2 * sizeof(void *));
+    *out = ms;
+    return 0;
+  }
+
+  HIP_TRY(c, hipEventRecord(c->ev[0], s), 10);
+
+  // H2D (skipped when the caller staged the identical volume already)
+  if ((flags_ & MG_FLAG_SKIP_H2D) && c->labels.cap >= nvox * esize) {
+    // resident input: nothing to upload
+  } else {
+    if (ensure(c, c->labels, nvox * esize)) return 11;
+    HIP_TRY(c, hipMemcpyAsync(c->labels.ptr, labels_host, nvox * esize,
+                              hipMemcpyHostToDevice, s), 11);
+  }
+  HIP_TRY(c, hipEventRecord(c->ev[1], s), 11);
+
+  // label hash (grow-and-retry on overflow)
+  uint64_t total_tris = 0;
+  uint32_t nlabels = 0;
+  for (;;) {
+    if (ensure(c, c->lh_keys, c->lh_slots * 8)) return 12;
+    if (ensure(c, c->lh_vals, c->lh_slots * 4)) return 12;
+    if (ensure(c, c->lh_misc, 256)) return 12;
+    if (ensure(c, c->segcnt, g.nseg * 4)) return 12;
+    if (ensure(c, c->segoff, g.nseg * 4)) return 12;
+    HIP_TRY(c, hipMemsetAsync(c->lh_keys.ptr, 0, c->lh_slots * 8, s), 12);
+    HIP_TRY(c, hipMemsetAsync(c->lh_misc.ptr, 0, 256, s), 12);
+    LabelHash lh;
+    lh.keys = (uint64_t *)c->lh_keys.ptr;
+    lh.vals = (uint32_t *)c->lh_vals.ptr;
+    lh.counter = (uint32_t *)c->lh_misc.ptr;
+    lh.overflow = (uint32_t *)c->lh_misc.ptr + 1;
+    lh.nslots = c->lh_slots;
+
+    int rc;
+    if (dtype == MG_U64)
+      rc = run_count_emit<uint64_t>(c, (const uint64_t *)c->labels.ptr, g, lh,
+                                    (uint32_t *)c->segcnt.ptr,
+                                    (uint32_t *)c->segoff.ptr,
+                                    &total_tris, &nlabels);
+    else
+      rc = run_count_emit<uint32_t>(c, (const uint32_t *)c->labels.ptr, g, lh,
+                                    (uint32_t *)c->segcnt.ptr,
+                                    (uint32_t *)c->segoff.ptr,
+                                    &total_tris, &nlabels);
+    if (rc == -100) {  // hash overflow: grow and retry
+      if (c->lh_slots >= (1ull << 27)) {
+        SET_ERR(c, "label hash overflow at %llu slots",
+                (unsigned long long)c->lh_slots);
+        return 13;
+      }
+      c->lh_slots <<= 2;
+      continue;
+    }
+    if (rc) return rc;
+    break;
+  }
+
+  LabelHash lh;
+  lh.keys = (uint64_t *)c->lh_keys.ptr;
+  lh.vals = (uint32_t *)c->lh_vals.ptr;
+  lh.counter = (uint32_t *)c->lh_misc.ptr;
+  lh.overflow = (uint32_t *)c->lh_misc.ptr + 1;
+  lh.nslots = c->lh_slots;
+
+  c->stats.total_tris = total_tris;
+  c->stats.n_labels = nlabels;
+
+  if (total_tris == 0 || nlabels == 0) {
+    mg_meshset *ms = (mg_meshset *)calloc(
+        1, sizeof(mg_meshset) + 2 * sizeof(void *));
+    *out = ms;
+    HIP_TRY(c, hipStreamSynchronize(s), 14);
+    c->stats.ms_h2d = ev_ms(c, 0, 1);
+    return 0;
+  }
+  if (total_tris > (1ull << 31) / 3 * 2) {  // 3T must fit u32 stream index
+    SET_ERR(c, "chunk produces %llu triangles (> corner-index limit); "
+            "split the task shape", (unsigned long long)total_tris);
+    return 15;
+  }
+  if (reduction_factor > 1) {
+    SET_ERR(c, "GPU simplifier not yet wired (reduction_factor=%u); "
+            "round-1 engine supports reduction_factor<=1",
+            reduction_factor);
+    return 16;
+  }
+
+  const uint64_t T = total_tris;
+  const uint64_t NC = 3 * T;  // corners
+
+  // label id -> value
+  if (ensure(c, c->label_values, (uint64_t)nlabels * 8)) return 17;
+  {
+    int blk = 256;
+    int64_t nb = (c->lh_slots + blk - 1) / blk;
+    hipLaunchKernelGGL(k_label_values, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       lh, (uint64_t *)c->label_values.ptr, c->lh_slots);
+  }
+
+  // [3] emit
+  if (ensure(c, c->tri_label, T * 4)) return 18;
+  if (ensure(c, c->tri_keys, NC * 8)) return 18;
+  {
+    int blk = 256;
+    int waves_per_blk = blk / WAVE;
+    int64_t nb = std::min<int64_t>((g.nseg + waves_per_blk - 1) / waves_per_blk,
+                                   8192);
+    if (dtype == MG_U64)
+      hipLaunchKernelGGL(k_emit<uint64_t>, dim3((uint32_t)nb), dim3(blk), 0, s,
+                         (const uint64_t *)c->labels.ptr, g,
+                         (const uint32_t *)c->segoff.ptr, lh,
+                         (uint32_t *)c->tri_label.ptr,
+                         (uint64_t *)c->tri_keys.ptr);
+    else
+      hipLaunchKernelGGL(k_emit<uint32_t>, dim3((uint32_t)nb), dim3(blk), 0, s,
+                         (const uint32_t *)c->labels.ptr, g,
+                         (const uint32_t *)c->segoff.ptr, lh,
+                         (uint32_t *)c->tri_label.ptr,
+                         (uint64_t *)c->tri_keys.ptr);
+  }
+  HIP_TRY(c, hipGetLastError(), 18);
+  HIP_TRY(c, hipEventRecord(c->ev[4], s), 18);
+
+  // [4] stable partition by label id
+  if (ensure(c, c->order, T * 4)) return 19;
+  if (ensure(c, c->order_alt, T * 4)) return 19;
+  if (ensure(c, c->tri_label_alt, T * 4)) return 19;
+  if (ensure(c, c->keys_sorted, NC * 8)) return 19;
+  {
+    int blk = 256;
+    uint64_t nb = (T + blk - 1) / blk;
+    hipLaunchKernelGGL(k_iota, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       (uint32_t *)c->order.ptr, T);
+    unsigned begin_bit = 0;
+    unsigned end_bit = 1;
+    while ((1u << end_bit) < nlabels) ++end_bit;
+    if (nlabels == 1) end_bit = 1;
+    rocprim::double_buffer<uint32_t> d_keys(
+        (uint32_t *)c->tri_label.ptr, (uint32_t *)c->tri_label_alt.ptr);
+    rocprim::double_buffer<uint32_t> d_vals(
+        (uint32_t *)c->order.ptr, (uint32_t *)c->order_alt.ptr);
+    size_t tmp_bytes = 0;
+    hipError_t e = rocprim::radix_sort_pairs(
+        nullptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
+    if (e != hipSuccess) { SET_ERR(c, "radix_sort size query failed"); return 19; }
+    if (ensure(c, c->sort_tmp, tmp_bytes)) return 19;
+    e = rocprim::radix_sort_pairs(
+        c->sort_tmp.ptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
+    if (e != hipSuccess) { SET_ERR(c, "radix_sort failed"); return 19; }
+    // gather
+    hipLaunchKernelGGL(k_gather_keys, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       (const uint64_t *)c->tri_keys.ptr, d_vals.current(),
+                       (uint64_t *)c->keys_sorted.ptr, T);
+    // label ranges
+    if (ensure(c, c->tri_off, ((uint64_t)nlabels + 1) * 4)) return 19;
+    hipLaunchKernelGGL(k_label_ranges, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       d_keys.current(), (uint32_t *)c->tri_off.ptr, T,
+                       nlabels);
+  }
+  HIP_TRY(c, hipGetLastError(), 19);
+  HIP_TRY(c, hipEventRecord(c->ev[5], s), 19);
+
+  // [5] weld — table sized past the worst case (uniques <= NC corners) so
+  // bounded probing cannot overflow on real inputs; the flag is checked
+  // anyway before any kernel that relies on key presence.
+  uint64_t wh_slots = next_pow2_u64(std::max<uint64_t>(1024, NC + (NC >> 1)));
+  if (ensure(c, c->wh_keys, wh_slots * 8)) return 20;
+  if (ensure(c, c->wh_minp, wh_slots * 4)) return 20;
+  if (ensure(c, c->wh_vtx, wh_slots * 4)) return 20;
+  if (ensure(c, c->flags, NC)) return 20;
+  if (ensure(c, c->vtx_scan, NC * 4)) return 20;
+  HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0, wh_slots * 8, s), 20);
+  HIP_TRY(c, hipMemsetAsync(c->wh_minp.ptr, 0xFF, wh_slots * 4, s), 20);
+  HIP_TRY(c, hipMemsetAsync((uint32_t *)c->lh_misc.ptr + 2, 0, 4, s), 20);
+  WeldHash wh;
+  wh.keys = (uint64_t *)c->wh_keys.ptr;
+  wh.minp = (uint32_t *)c->wh_minp.ptr;
+  wh.vtx = (uint32_t *)c->wh_vtx.ptr;
+  wh.overflow = (uint32_t *)c->lh_misc.ptr + 2;
+  wh.nslots = wh_slots;
+  {
+    int blk = 256;
+    uint64_t nb = (NC + blk - 1) / blk;
+    hipLaunchKernelGGL(k_weld_insert, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       (const uint64_t *)c->keys_sorted.ptr, wh, NC);
+    uint32_t wh_ovf = 0;
+    HIP_TRY(c, hipMemcpyAsync(&wh_ovf, (uint32_t *)c->lh_misc.ptr + 2, 4,
+                              hipMemcpyDeviceToHost, s), 20);
+    HIP_TRY(c, hipStreamSynchronize(s), 20);
+    if (wh_ovf) { SET_ERR(c, "weld hash overflow"); return 20; }
+    hipLaunchKernelGGL(k_weld_flags, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       (const uint64_t *)c->keys_sorted.ptr, wh,
+                       (uint8_t *)c->flags.ptr, NC);
+    // scan flags -> vertex ids
+    auto conv = [] __host__ __device__(uint8_t f) -> uint32_t { return f; };
+    auto it = rocprim::make_transform_iterator((uint8_t *)c->flags.ptr, conv);
+    size_t tmp_bytes = 0;
+    hipError_t e = rocprim::exclusive_scan(
+        nullptr, tmp_bytes, it, (uint32_t *)c->vtx_scan.ptr, 0u, NC,
+        rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "weld scan size query failed"); return 20; }
+    if (ensure(c, c->scan_tmp, tmp_bytes)) return 20;
+    e = rocprim::exclusive_scan(
+        c->scan_tmp.ptr, tmp_bytes, it, (uint32_t *)c->vtx_scan.ptr, 0u, NC,
+        rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "weld scan failed"); return 20; }
+  }
+  // total verts = scan[NC-1] + flags[NC-1]
+  uint64_t total_verts = 0;
+  {
+    uint32_t last_scan = 0;
+    uint8_t last_flag = 0;
+    HIP_TRY(c, hipMemcpyAsync(&last_scan,
+                              (uint32_t *)c->vtx_scan.ptr + (NC - 1), 4,
+                              hipMemcpyDeviceToHost, s), 21);
+    HIP_TRY(c, hipMemcpyAsync(&last_flag, (uint8_t *)c->flags.ptr + (NC - 1),
+                              1, hipMemcpyDeviceToHost, s), 21);
+    HIP_TRY(c, hipStreamSynchronize(s), 21);
+    total_verts = (uint64_t)last_scan + last_flag;
+  }
+  c->stats.total_verts = total_verts;
+
+  if (ensure(c, c->verts, total_verts * 12)) return 22;
+  if (ensure(c, c->faces, NC * 4)) return 22;
+  if (ensure(c, c->vbase, ((uint64_t)nlabels + 1) * 4)) return 22;
+  {
+    int blk = 256;
+    uint64_t nb = (NC + blk - 1) / blk;
+    const float shift = voxel_centered ? 0.0f : 0.5f;
+    hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       (const uint64_t *)c->keys_sorted.ptr,
+                       (const uint8_t *)c->flags.ptr,
+                       (const uint32_t *)c->vtx_scan.ptr, wh,
+                       (float *)c->verts.ptr, rx, ry, rz, shift, NC);
+    uint32_t nbl = (nlabels + 1 + 255) / 256;
+    hipLaunchKernelGGL(k_vbase, dim3(nbl), dim3(256), 0, s,
+                       (const uint32_t *)c->tri_off.ptr,
+                       (const uint32_t *)c->vtx_scan.ptr,
+                       (uint32_t *)c->vbase.ptr, nlabels, total_verts);
+    hipLaunchKernelGGL(k_faces, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       (const uint64_t *)c->keys_sorted.ptr, wh,
+                       (const uint32_t *)c->vbase.ptr,
+                       (uint32_t *)c->faces.ptr, NC);
+  }
+  HIP_TRY(c, hipGetLastError(), 22);
+  HIP_TRY(c, hipEventRecord(c->ev[6], s), 22);
+
+  // [7] extract
+  mg_meshset *ms = nullptr;
+  if (flags_ & MG_FLAG_DEVICE_ONLY) {
+    HIP_TRY(c, hipStreamSynchronize(s), 23);
+    ms = (mg_meshset *)calloc(1, sizeof(mg_meshset) + 2 * sizeof(void *));
+    *out = ms;
+  } else {
+    float *h_verts = nullptr;
+    uint32_t *h_faces = nullptr;
+    HIP_TRY(c, hipHostMalloc((void **)&h_verts, total_verts * 12), 24);
+    if (hipHostMalloc((void **)&h_faces, NC * 4) != hipSuccess) {
+      (void)hipHostFree(h_verts);
+      SET_ERR(c, "hipHostMalloc faces failed");
+      return 24;
+    }
+    std::vector<uint32_t> h_tri_off(nlabels + 1), h_vbase(nlabels + 1);
+    std::vector<uint64_t> h_label_values(nlabels);
+    HIP_TRY(c, hipMemcpyAsync(h_verts, c->verts.ptr, total_verts * 12,
+                              hipMemcpyDeviceToHost, s), 24);
+    HIP_TRY(c, hipMemcpyAsync(h_faces, c->faces.ptr, NC * 4,
+                              hipMemcpyDeviceToHost, s), 24);
+    HIP_TRY(c, hipMemcpyAsync(h_tri_off.data(), c->tri_off.ptr,
+                              (nlabels + 1) * 4, hipMemcpyDeviceToHost, s), 24);
+    HIP_TRY(c, hipMemcpyAsync(h_vbase.data(), c->vbase.ptr, (nlabels + 1) * 4,
+                              hipMemcpyDeviceToHost, s), 24);
+    HIP_TRY(c, hipMemcpyAsync(h_label_values.data(), c->label_values.ptr,
+                              nlabels * 8, hipMemcpyDeviceToHost, s), 24);
+    HIP_TRY(c, hipStreamSynchronize(s), 24);
+
+    ms = (mg_meshset *)calloc(
+        1, sizeof(mg_meshset) + sizeof(mg_mesh) * nlabels + 2 * sizeof(void *));
+    ms->nmeshes = nlabels;
+    ms->meshes = (mg_mesh *)((char *)ms + sizeof(mg_meshset));
+    void **bufs = (void **)((char *)ms->meshes + sizeof(mg_mesh) * nlabels);
+    bufs[0] = h_verts;
+    bufs[1] = h_faces;
+    // order meshes by ascending label value
+    std::vector<uint32_t> idx(nlabels);
+    for (uint32_t i = 0; i < nlabels; ++i) idx[i] = i;
+    std::sort(idx.begin(), idx.end(), [&](uint32_t a, uint32_t b) {
+      return h_label_values[a] < h_label_values[b];
+    });
+    for (uint32_t m = 0; m < nlabels; ++m) {
+      uint32_t l = idx[m];
+      mg_mesh &mm = ms->meshes[m];
+      mm.label = h_label_values[l];
+      mm.nverts = h_vbase[l + 1] - h_vbase[l];
+      mm.ntris = h_tri_off[l + 1] - h_tri_off[l];
+      mm.verts = h_verts + 3ull * h_vbase[l];
+      mm.faces = h_faces + 3ull * h_tri_off[l];
+    }
+    *out = ms;
+  }
+  HIP_TRY(c, hipEventRecord(c->ev[7], s), 25);
+  HIP_TRY(c, hipStreamSynchronize(s), 25);
+
+  c->stats.ms_h2d = ev_ms(c, 0, 1);
+  c->stats.ms_count = ev_ms(c, 1, 2);
+  c->stats.ms_scan = ev_ms(c, 2, 3);
+  c->stats.ms_emit = ev_ms(c, 3, 4);
+  c->stats.ms_partition = ev_ms(c, 4, 5);
+  c->stats.ms_weld = ev_ms(c, 5, 6);
+  c->stats.ms_simplify = 0.0;
+  c->stats.ms_d2h = ev_ms(c, 6, 7);
+  c->stats.ms_total = ev_ms(c, 0, 7);
+  return 0;
+}
+
+template <typename T>
+static int run_count_emit(mg_ctx *c, const T *d_labels, const GridDims &g,
+                          LabelHash lh, uint32_t *d_segcnt,
+                          uint32_t *d_segoff, uint64_t *p_total,
+                          uint32_t *p_nlabels) {
+  hipStream_t s = c->stream;
+  HIP_TRY(c, hipEventRecord(c->ev[1], s), 30);
+  int blk = 256;
+  int waves_per_blk = blk / WAVE;
+  int64_t nb = std::min<int64_t>(
+      (g.nseg + waves_per_blk - 1) / waves_per_blk, 8192);
+  hipLaunchKernelGGL(k_count<T>, dim3((uint32_t)nb), dim3(blk), 0, s,
+                     d_labels, g, d_segcnt, lh);
+  HIP_TRY(c, hipGetLastError(), 30);
+  HIP_TRY(c, hipEventRecord(c->ev[2], s), 30);
+
+  // check overflow + read label count
+  uint32_t misc[2] = {0, 0};
+  HIP_TRY(c, hipMemcpyAsync(misc, c->lh_misc.ptr, 8, hipMemcpyDeviceToHost, s),
+          30);
+
+  // scan segcnt -> segoff
+  size_t tmp_bytes = 0;
+  hipError_t e = rocprim::exclusive_scan(
+      nullptr, tmp_bytes, d_segcnt, d_segoff, 0u, (size_t)g.nseg,
+      rocprim::plus<uint32_t>(), s);
+  if (e != hipSuccess) { SET_ERR(c, "seg scan size query failed"); return 30; }
+  if (ensure(c, c->scan_tmp, tmp_bytes)) return 30;
+  e = rocprim::exclusive_scan(
+      c->scan_tmp.ptr, tmp_bytes, d_segcnt, d_segoff, 0u, (size_t)g.nseg,
+      rocprim::plus<uint32_t>(), s);
+  if (e != hipSuccess) { SET_ERR(c, "seg scan failed"); return 30; }
+
+  uint32_t last_off = 0, last_cnt = 0;
+  HIP_TRY(c, hipMemcpyAsync(&last_off, d_segoff + (g.nseg - 1), 4,
+                            hipMemcpyDeviceToHost, s), 30);
+  HIP_TRY(c, hipMemcpyAsync(&last_cnt, d_segcnt + (g.nseg - 1), 4,
+                            hipMemcpyDeviceToHost, s), 30);
+  HIP_TRY(c, hipEventRecord(c->ev[3], s), 30);
+  HIP_TRY(c, hipStreamSynchronize(s), 30);
+  if (misc[1]) return -100;  // label hash overflow
+  *p_nlabels = misc[0];
+  *p_total = (uint64_t)last_off + last_cnt;
+  return 0;
+}

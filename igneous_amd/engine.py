@@ -22,6 +22,7 @@ SO_PATH = os.path.join(_HERE, "csrc", "libmeshgine.so")
 
 MG_U32, MG_U64 = 0, 1
 MG_FLAG_DEVICE_ONLY = 1
+MG_FLAG_SKIP_H2D = 2
 
 _EXPORTED_SYMBOLS = [
     "mg_init", "mg_destroy", "mg_mesh_chunk", "mg_meshset_free",
@@ -137,7 +138,8 @@ class Engine:
     def mesh_chunk(self, labels: np.ndarray, resolution=(1.0, 1.0, 1.0),
                    reduction_factor: int = 0, max_error: float = 40.0,
                    voxel_centered: bool = True,
-                   device_only: bool = False) -> dict:
+                   device_only: bool = False,
+                   skip_h2d: bool = False) -> dict:
         """GPU counterpart of oracle.mesh_chunk: F-order (sx,sy,sz)
         uint32/uint64 labels -> {label: (verts (V,3) f32 nm, faces (F,3) u32)},
         labels ascending. Under device_only returns {} (stats still filled)."""
@@ -149,7 +151,8 @@ class Engine:
         else:
             raise ValueError(f"unsupported label dtype {labels.dtype}")
         sx, sy, sz = labels.shape
-        flags = MG_FLAG_DEVICE_ONLY if device_only else 0
+        flags = (MG_FLAG_DEVICE_ONLY if device_only else 0) | \
+                (MG_FLAG_SKIP_H2D if skip_h2d else 0)
         out = ctypes.POINTER(_MgMeshSet)()
         with self._lock:
             rc = self.lib.mg_mesh_chunk(

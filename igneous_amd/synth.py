@@ -10,8 +10,21 @@ import numpy as np
 
 
 def voronoi_labels(shape, K: int, seed: int, dtype=np.uint64,
-                   background_frac: float = 0.03) -> np.ndarray:
-    """F-order (sx,sy,sz) label volume."""
+                   background_frac: float = 0.03,
+                   cache: bool = True) -> np.ndarray:
+    """F-order (sx,sy,sz) label volume. Deterministic; large volumes are
+    cached on local disk (generation of 512^3/K=50k takes ~30 s)."""
+    import os
+    cache_path = None
+    if cache and int(np.prod(shape)) >= 64 ** 3:
+        cdir = os.environ.get("MESHGINE_CACHE", "/tmp/meshgine_cache")
+        os.makedirs(cdir, exist_ok=True)
+        tag = "x".join(str(int(s)) for s in shape)
+        cache_path = os.path.join(
+            cdir, f"vor_{tag}_K{K}_s{seed}_{np.dtype(dtype).name}"
+                  f"_b{background_frac}.npy")
+        if os.path.exists(cache_path):
+            return np.asfortranarray(np.load(cache_path))
     rng = np.random.default_rng(seed)
     shape = tuple(int(s) for s in shape)
     seeds = np.stack([
@@ -54,4 +67,8 @@ def voronoi_labels(shape, K: int, seed: int, dtype=np.uint64,
         lab[dist > dist_thresh] = 0
         out[:, :, z0:z1] = lab.reshape(
             (shape[0], shape[1], nz), order="F")
+    if cache_path is not None:
+        tmp = cache_path + ".tmp.npy"
+        np.save(tmp, out)
+        os.replace(tmp, cache_path)
     return out

@@ -56,6 +56,11 @@ typedef struct mg_ctx mg_ctx;
 #define MG_FLAG_DEVICE_ONLY 1u  /* run all kernels, skip the D2H extract
                                    (bench: job complete with outputs in
                                    HBM; sizes still reported) */
+#define MG_FLAG_SKIP_H2D    2u  /* label volume already resident in the
+                                   ctx's device buffer from a previous
+                                   call with identical dims/dtype (bench:
+                                   timed region starts with inputs in
+                                   HBM). Caller's responsibility. */
 
 /* One label's mesh. verts = 3*nverts float32 chunk-local nm (x,y,z);
  * faces = 3*ntris uint32 indices into verts. Pointers alias the meshset's

@@ -162,23 +162,35 @@ def main():
         f.write("// generator's docstring for corner/edge numbering and the\n")
         f.write("// face-ambiguity rule. Shared by oracle/ and HIP kernels.\n")
         f.write("#ifndef MC_TABLE_H\n#define MC_TABLE_H\n\n")
+        f.write("// HIP device code defines MC_TABLE_QUAL as __device__ static\n")
+        f.write("// const before including; C host code gets plain static const.\n")
+        f.write("#ifndef MC_TABLE_QUAL\n#define MC_TABLE_QUAL static const\n#endif\n\n")
         f.write(f"#define MC_MAX_TRIS {maxt}\n\n")
         f.write("// number of triangles for each 8-bit corner mask\n")
-        f.write("static const unsigned char MC_TRI_COUNT[256] = {\n")
+        f.write("MC_TABLE_QUAL unsigned char MC_TRI_COUNT[256] = {\n")
         for row in range(0, 256, 16):
             f.write("  " + ", ".join(str(len(table[m])) for m in range(row, row + 16)) + ",\n")
         f.write("};\n\n")
         f.write("// edge ids, 3 per triangle, MC_MAX_TRIS*3 slots per mask, -1 padded\n")
-        f.write(f"static const signed char MC_TRI_TABLE[256][{maxt * 3}] = {{\n")
+        f.write(f"MC_TABLE_QUAL signed char MC_TRI_TABLE[256][{maxt * 3}] = {{\n")
         for m in range(256):
             flat = [e for t in table[m] for e in t]
             flat += [-1] * (maxt * 3 - len(flat))
             f.write("  {" + ", ".join(f"{v}" for v in flat) + "},\n")
         f.write("};\n\n")
         f.write("// doubled-coordinate offset (x,y,z) of each edge midpoint\n")
-        f.write("static const unsigned char MC_EDGE_DOFF[12][3] = {\n")
+        f.write("MC_TABLE_QUAL unsigned char MC_EDGE_DOFF[12][3] = {\n")
         for e in range(12):
             f.write("  {%d, %d, %d},\n" % EDGE_DOFF[e])
+        f.write("};\n\n")
+        f.write("// packed 12-bit-per-axis key offset of each edge midpoint\n")
+        f.write("// relative to the cell's packed doubled origin:\n")
+        f.write("// key = ((2cz)<<24 | (2cy)<<12 | 2cx) + MC_EDGE_KEYOFF[e]\n")
+        f.write("// (no cross-field carries: doubled coords stay < 4096)\n")
+        f.write("MC_TABLE_QUAL unsigned long long MC_EDGE_KEYOFF[12] = {\n")
+        for e in range(12):
+            ox, oy, oz = EDGE_DOFF[e]
+            f.write(f"  0x{(oz << 24) | (oy << 12) | ox:x}ULL,\n")
         f.write("};\n\n#endif // MC_TABLE_H\n")
     print(f"wrote {out}: max {maxt} tris/cell")
 
