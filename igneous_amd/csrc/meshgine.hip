@@ -107,15 +107,13 @@ __device__ __forceinline__ void label_insert(LabelHash h, uint64_t label) {
 
 __device__ __forceinline__ uint32_t label_lookup(LabelHash h, uint64_t label) {
   uint64_t slot = mix64(label) & (h.nslots - 1);
-  for (;;) {
-    uint64_t cur = h.keys[slot];
-    if (cur == label) {
-      // id may have been published by another wave after the CAS; spin-free
-      // read is safe because k_count completed before any lookup kernel.
-      return h.vals[slot];
-    }
+  for (uint64_t probe = 0; probe < h.nslots; ++probe) {
+    // ids were published by k_count, which completed before any lookup
+    if (h.keys[slot] == label) return h.vals[slot];
     slot = (slot + 1) & (h.nslots - 1);
   }
+  atomicExch(h.overflow, 1u);  // bug surfaces as error, not a GPU hang
+  return 0;
 }
 
 // invert: label id -> label value (fill after count)
@@ -337,8 +335,14 @@ __device__ __forceinline__ uint64_t weld_slot(WeldHash h, uint64_t key) {
 
 __device__ __forceinline__ uint64_t weld_find(WeldHash h, uint64_t key) {
   uint64_t slot = mix64(key) & (h.nslots - 1);
-  while (h.keys[slot] != key) slot = (slot + 1) & (h.nslots - 1);
-  return slot;
+  for (uint64_t probe = 0; probe < h.nslots; ++probe) {
+    if (h.keys[slot] == key) return slot;
+    slot = (slot + 1) & (h.nslots - 1);
+  }
+  // unreachable after a verified insert pass; bounded so a logic bug
+  // surfaces as a wrong result + overflow flag instead of a GPU hang
+  atomicExch(h.overflow, 1u);
+  return 0;
 }
 
 // [5a] insert all corners; record first (minimum) stream position per key
