@@ -309,75 +309,77 @@ __global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
 // ---------------------------------------------------------------------------
 // welding
 
-struct WeldHash {
-  uint64_t *keys;     // 0 = empty (coord bits never all-zero)
-  uint32_t *minp;     // first corner position
-  uint32_t *vtx;      // assigned vertex id
-  uint32_t *overflow; // error flag (bounded probing: no device-side hang)
-  uint64_t nslots;    // power of two
+// Packed 16-byte entry: one cache line touch per probe. minp_enc stores
+// ~(first position) via atomicMax so the whole table zero-memsets
+// (0 = "no position yet", and ~i > 0 for every valid i).
+struct WeldEntry {
+  unsigned long long key;  // 0 = empty (coord bits never all-zero)
+  uint32_t minp_enc;       // ~min(position), 0 = none
+  uint32_t vtx;            // assigned vertex id (filled by k_weld_verts)
 };
 
-__device__ __forceinline__ uint64_t weld_slot(WeldHash h, uint64_t key) {
-  uint64_t slot = mix64(key) & (h.nslots - 1);
-  for (uint64_t probe = 0; probe < h.nslots; ++probe) {
-    uint64_t cur = h.keys[slot];
-    if (cur == key) return slot;
-    if (cur == 0) {
-      uint64_t prev = atomicCAS((unsigned long long *)&h.keys[slot], 0ull,
-                                (unsigned long long)key);
-      if (prev == 0 || prev == key) return slot;
-    }
-    slot = (slot + 1) & (h.nslots - 1);
-  }
-  atomicExch(h.overflow, 1u);
-  return 0;
-}
+struct WeldHash {
+  WeldEntry *e;
+  uint32_t *overflow;  // error flag (bounded probing: no device-side hang)
+  uint64_t nslots;     // power of two
+};
 
-__device__ __forceinline__ uint64_t weld_find(WeldHash h, uint64_t key) {
-  uint64_t slot = mix64(key) & (h.nslots - 1);
-  for (uint64_t probe = 0; probe < h.nslots; ++probe) {
-    if (h.keys[slot] == key) return slot;
-    slot = (slot + 1) & (h.nslots - 1);
-  }
-  // unreachable after a verified insert pass; bounded so a logic bug
-  // surfaces as a wrong result + overflow flag instead of a GPU hang
-  atomicExch(h.overflow, 1u);
-  return 0;
-}
-
-// [5a] insert all corners; record first (minimum) stream position per key
+// [5a] insert all corners; record first (minimum) stream position per key;
+// store the claimed slot per corner so later passes never re-probe.
 __global__ void k_weld_insert(const uint64_t *__restrict__ keys_sorted,
-                              WeldHash h, uint64_t ncorners) {
+                              WeldHash h, uint32_t *__restrict__ slots,
+                              uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  uint64_t slot = weld_slot(h, keys_sorted[i]);
-  atomicMin(&h.minp[slot], (uint32_t)i);
+  uint64_t key = keys_sorted[i];
+  uint64_t slot = mix64(key) & (h.nslots - 1);
+  for (uint64_t probe = 0; probe < h.nslots; ++probe) {
+    uint64_t cur = h.e[slot].key;
+    if (cur != key) {
+      if (cur != 0) { slot = (slot + 1) & (h.nslots - 1); continue; }
+      uint64_t prev = atomicCAS(&h.e[slot].key, 0ull, (unsigned long long)key);
+      if (prev != 0 && prev != key) { slot = (slot + 1) & (h.nslots - 1); continue; }
+    }
+    atomicMax(&h.e[slot].minp_enc, ~(uint32_t)i);
+    slots[i] = (uint32_t)slot;
+    return;
+  }
+  atomicExch(h.overflow, 1u);
+  slots[i] = 0;
 }
 
-// [5b] flag first occurrences
-__global__ void k_weld_flags(const uint64_t *__restrict__ keys_sorted,
-                             const WeldHash h, uint8_t *__restrict__ flags,
-                             uint64_t ncorners) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= ncorners) return;
-  uint64_t slot = weld_find(h, keys_sorted[i]);
-  flags[i] = (h.minp[slot] == (uint32_t)i) ? 1 : 0;
+// scan-input functor: 1 where this corner is its key's first occurrence
+struct FirstOccur {
+  const uint32_t *slots;
+  const WeldEntry *e;
+  __device__ uint32_t operator()(uint32_t i) const {
+    return (e[slots[i]].minp_enc == ~i) ? 1u : 0u;
+  }
+};
+
+// total verts = vtx_scan[NC-1] + first_occur(NC-1)
+__global__ void k_total_verts(const uint32_t *__restrict__ vtx_scan,
+                              const uint32_t *__restrict__ slots,
+                              const WeldEntry *__restrict__ e,
+                              uint64_t ncorners, uint32_t *out) {
+  uint32_t i = (uint32_t)(ncorners - 1);
+  *out = vtx_scan[i] + ((e[slots[i]].minp_enc == ~i) ? 1u : 0u);
 }
 
-// [5d] first occurrences: record vertex id in hash, write vertex position
+// [5d] first occurrences: record vertex id in the entry, write the vertex
 __global__ void k_weld_verts(const uint64_t *__restrict__ keys_sorted,
-                             const uint8_t *__restrict__ flags,
+                             const uint32_t *__restrict__ slots,
                              const uint32_t *__restrict__ vtx_scan,
                              WeldHash h, float *__restrict__ verts,
                              float rx, float ry, float rz, float shift,
                              uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  if (!flags[i]) return;
-  uint64_t key = keys_sorted[i];
-  uint64_t slot = weld_find(h, key);
+  WeldEntry &ent = h.e[slots[i]];
+  if (ent.minp_enc != ~(uint32_t)i) return;
   uint32_t v = vtx_scan[i];
-  h.vtx[slot] = v;
+  ent.vtx = v;
+  uint64_t key = keys_sorted[i];
   float dx = (float)(uint32_t)(key & 0xFFF);
   float dy = (float)(uint32_t)((key >> 12) & 0xFFF);
   float dz = (float)(uint32_t)((key >> 24) & 0xFFF);
@@ -400,15 +402,14 @@ __global__ void k_vbase(const uint32_t *__restrict__ tri_off,
 // [5f] faces: per-label local vertex indices
 __global__ void k_faces(const uint64_t *__restrict__ keys_sorted,
                         const WeldHash h,
+                        const uint32_t *__restrict__ slots,
                         const uint32_t *__restrict__ vbase,
                         uint32_t *__restrict__ faces,
                         uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  uint64_t key = keys_sorted[i];
-  uint64_t slot = weld_find(h, key);
-  uint32_t lid = (uint32_t)(key >> 36);
-  faces[i] = h.vtx[slot] - vbase[lid];
+  uint32_t lid = (uint32_t)(keys_sorted[i] >> 36);
+  faces[i] = h.e[slots[i]].vtx - vbase[lid];
 }
 
 __global__ void k_iota(uint32_t *p, uint64_t n) {
@@ -778,40 +779,38 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   HIP_TRY(c, hipGetLastError(), 19);
   HIP_TRY(c, hipEventRecord(c->ev[5], s), 19);
 
-  // [5] weld — table sized past the worst case (uniques <= NC corners) so
-  // bounded probing cannot overflow on real inputs; the flag is checked
-  // anyway before any kernel that relies on key presence.
-  uint64_t wh_slots = next_pow2_u64(std::max<uint64_t>(1024, NC + (NC >> 1)));
-  if (ensure(c, c->wh_keys, wh_slots * 8)) return 20;
-  if (ensure(c, c->wh_minp, wh_slots * 4)) return 20;
-  if (ensure(c, c->wh_vtx, wh_slots * 4)) return 20;
-  if (ensure(c, c->flags, NC)) return 20;
-  if (ensure(c, c->vtx_scan, NC * 4)) return 20;
-  HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0, wh_slots * 8, s), 20);
-  HIP_TRY(c, hipMemsetAsync(c->wh_minp.ptr, 0xFF, wh_slots * 4, s), 20);
-  HIP_TRY(c, hipMemsetAsync((uint32_t *)c->lh_misc.ptr + 2, 0, 4, s), 20);
+  // [5] weld — grow-on-overflow retry; packed 16B entries, zero-memset
+  uint64_t wh_slots = next_pow2_u64(std::max<uint64_t>(1024, NC / 2));
+  uint64_t total_verts = 0;
   WeldHash wh;
-  wh.keys = (uint64_t *)c->wh_keys.ptr;
-  wh.minp = (uint32_t *)c->wh_minp.ptr;
-  wh.vtx = (uint32_t *)c->wh_vtx.ptr;
-  wh.overflow = (uint32_t *)c->lh_misc.ptr + 2;
-  wh.nslots = wh_slots;
-  {
+  for (;;) {
+    if (ensure(c, c->wh_keys, wh_slots * sizeof(WeldEntry))) return 20;
+    if (ensure(c, c->flags, NC * 4)) return 20;  // slot ids
+    if (ensure(c, c->vtx_scan, NC * 4)) return 20;
+    HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0,
+                              wh_slots * sizeof(WeldEntry), s), 20);
+    HIP_TRY(c, hipMemsetAsync((uint32_t *)c->lh_misc.ptr + 2, 0, 4, s), 20);
+    wh.e = (WeldEntry *)c->wh_keys.ptr;
+    wh.overflow = (uint32_t *)c->lh_misc.ptr + 2;
+    wh.nslots = wh_slots;
     int blk = 256;
     uint64_t nb = (NC + blk - 1) / blk;
     hipLaunchKernelGGL(k_weld_insert, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint64_t *)c->keys_sorted.ptr, wh, NC);
+                       (const uint64_t *)c->keys_sorted.ptr, wh,
+                       (uint32_t *)c->flags.ptr, NC);
     uint32_t wh_ovf = 0;
     HIP_TRY(c, hipMemcpyAsync(&wh_ovf, (uint32_t *)c->lh_misc.ptr + 2, 4,
                               hipMemcpyDeviceToHost, s), 20);
     HIP_TRY(c, hipStreamSynchronize(s), 20);
-    if (wh_ovf) { SET_ERR(c, "weld hash overflow"); return 20; }
-    hipLaunchKernelGGL(k_weld_flags, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint64_t *)c->keys_sorted.ptr, wh,
-                       (uint8_t *)c->flags.ptr, NC);
-    // scan flags -> vertex ids
-    auto conv = [] __host__ __device__(uint8_t f) -> uint32_t { return f; };
-    auto it = rocprim::make_transform_iterator((uint8_t *)c->flags.ptr, conv);
+    if (wh_ovf) {
+      if (wh_slots >= (1ull << 32)) { SET_ERR(c, "weld hash overflow"); return 20; }
+      wh_slots <<= 1;
+      continue;
+    }
+    // scan first-occurrence flags -> vertex ids (flags computed on the fly)
+    FirstOccur fo{(const uint32_t *)c->flags.ptr, (const WeldEntry *)c->wh_keys.ptr};
+    auto it = rocprim::make_transform_iterator(
+        rocprim::counting_iterator<uint32_t>(0), fo);
     size_t tmp_bytes = 0;
     hipError_t e = rocprim::exclusive_scan(
         nullptr, tmp_bytes, it, (uint32_t *)c->vtx_scan.ptr, 0u, NC,
@@ -822,19 +821,17 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
         c->scan_tmp.ptr, tmp_bytes, it, (uint32_t *)c->vtx_scan.ptr, 0u, NC,
         rocprim::plus<uint32_t>(), s);
     if (e != hipSuccess) { SET_ERR(c, "weld scan failed"); return 20; }
-  }
-  // total verts = scan[NC-1] + flags[NC-1]
-  uint64_t total_verts = 0;
-  {
-    uint32_t last_scan = 0;
-    uint8_t last_flag = 0;
-    HIP_TRY(c, hipMemcpyAsync(&last_scan,
-                              (uint32_t *)c->vtx_scan.ptr + (NC - 1), 4,
+    hipLaunchKernelGGL(k_total_verts, dim3(1), dim3(1), 0, s,
+                       (const uint32_t *)c->vtx_scan.ptr,
+                       (const uint32_t *)c->flags.ptr,
+                       (const WeldEntry *)c->wh_keys.ptr, NC,
+                       (uint32_t *)c->lh_misc.ptr + 3);
+    uint32_t tv = 0;
+    HIP_TRY(c, hipMemcpyAsync(&tv, (uint32_t *)c->lh_misc.ptr + 3, 4,
                               hipMemcpyDeviceToHost, s), 21);
-    HIP_TRY(c, hipMemcpyAsync(&last_flag, (uint8_t *)c->flags.ptr + (NC - 1),
-                              1, hipMemcpyDeviceToHost, s), 21);
     HIP_TRY(c, hipStreamSynchronize(s), 21);
-    total_verts = (uint64_t)last_scan + last_flag;
+    total_verts = tv;
+    break;
   }
   c->stats.total_verts = total_verts;
 
@@ -847,7 +844,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     const float shift = voxel_centered ? 0.0f : 0.5f;
     hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nb), dim3(blk), 0, s,
                        (const uint64_t *)c->keys_sorted.ptr,
-                       (const uint8_t *)c->flags.ptr,
+                       (const uint32_t *)c->flags.ptr,
                        (const uint32_t *)c->vtx_scan.ptr, wh,
                        (float *)c->verts.ptr, rx, ry, rz, shift, NC);
     uint32_t nbl = (nlabels + 1 + 255) / 256;
@@ -857,6 +854,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                        (uint32_t *)c->vbase.ptr, nlabels, total_verts);
     hipLaunchKernelGGL(k_faces, dim3((uint32_t)nb), dim3(blk), 0, s,
                        (const uint64_t *)c->keys_sorted.ptr, wh,
+                       (const uint32_t *)c->flags.ptr,
                        (const uint32_t *)c->vbase.ptr,
                        (uint32_t *)c->faces.ptr, NC);
   }
