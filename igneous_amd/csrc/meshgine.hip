@@ -261,15 +261,20 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
         uint32_t nt = MC_TRI_COUNT[mask];
         if (!nt) continue;
         uint32_t lid = label_lookup(lh, (uint64_t)L);
-        uint64_t lid_hi = (uint64_t)lid << 36;
+        uint64_t lid_hi = (uint64_t)lid << 37;
         const signed char *tt = MC_TRI_TABLE[mask];
         for (uint32_t t = 0; t < nt; ++t) {
           tri_label[pos] = lid;
           #pragma unroll
           for (int v = 0; v < 3; ++v) {
             int e = tt[3 * t + v];
+            // side bit: is L the edge's UPPER endpoint label? The weld
+            // table has exactly two slots per edge (an edge midpoint is a
+            // vertex only for its two endpoint labels) — collision-free
+            // direct addressing, no hash.
+            uint64_t side = (c[MC_EDGE_CORNERS[e][1]] == L) ? 1ull : 0ull;
             tri_keys[3 * (uint64_t)pos + v] =
-                lid_hi | (cellkey + MC_EDGE_KEYOFF[e]);
+                lid_hi | (side << 36) | (cellkey + MC_EDGE_KEYOFF[e]);
           }
           ++pos;
         }
@@ -309,76 +314,78 @@ __global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
 // ---------------------------------------------------------------------------
 // welding
 
-// Packed 16-byte entry: one cache line touch per probe. minp_enc stores
-// ~(first position) via atomicMax so the whole table zero-memsets
-// (0 = "no position yet", and ~i > 0 for every valid i).
-struct WeldEntry {
-  unsigned long long key;  // 0 = empty (coord bits never all-zero)
-  uint32_t minp_enc;       // ~min(position), 0 = none
-  uint32_t vtx;            // assigned vertex id (filled by k_weld_verts)
-};
+// Direct-addressed weld table: an edge midpoint is a vertex only for the
+// labels of the edge's two endpoint voxels, so (edge, side) is a
+// collision-free address — no hash, no probing, and label-partitioned
+// corner streams touch CONTIGUOUS table lines (the reference's spatial
+// locality survives). minp_enc stores ~(first position) via atomicMax so
+// the table zero-memsets (0 = "no position yet").
+//
+// slot = axis*nvox + linear_voxel(vx,vy,vz), doubled; axis = the key's
+// odd coordinate. Capacity bound: 6*nvox must fit in u32 (checked on the
+// host; matches the reference's own 32-bit mesher task bound,
+// igneous_cli/cli.py:1049-1052).
 
-struct WeldHash {
-  WeldEntry *e;
-  uint32_t *overflow;  // error flag (bounded probing: no device-side hang)
-  uint64_t nslots;     // power of two
-};
+__device__ __forceinline__ uint32_t weld_slot_of_key(uint64_t key,
+                                                     int64_t sx, int64_t sy,
+                                                     int64_t nvox) {
+  uint32_t kx = (uint32_t)(key & 0xFFF);
+  uint32_t ky = (uint32_t)((key >> 12) & 0xFFF);
+  uint32_t kz = (uint32_t)((key >> 24) & 0xFFF);
+  int axis = (kx & 1) ? 0 : ((ky & 1) ? 1 : 2);
+  int64_t vx = kx >> 1, vy = ky >> 1, vz = kz >> 1;
+  int64_t lin = (vz * sy + vy) * sx + vx;
+  uint64_t side = (key >> 36) & 1;
+  return (uint32_t)((((uint64_t)axis * nvox + lin) << 1) | side);
+}
 
-// [5a] insert all corners; record first (minimum) stream position per key;
-// store the claimed slot per corner so later passes never re-probe.
+// [5a] record first (minimum) stream position per (edge, side); store the
+// slot per corner so later passes never recompute the address chain.
 __global__ void k_weld_insert(const uint64_t *__restrict__ keys_sorted,
-                              WeldHash h, uint32_t *__restrict__ slots,
+                              uint32_t *__restrict__ wminp,
+                              uint32_t *__restrict__ slots,
+                              int64_t sx, int64_t sy, int64_t nvox,
                               uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  uint64_t key = keys_sorted[i];
-  uint64_t slot = mix64(key) & (h.nslots - 1);
-  for (uint64_t probe = 0; probe < h.nslots; ++probe) {
-    uint64_t cur = h.e[slot].key;
-    if (cur != key) {
-      if (cur != 0) { slot = (slot + 1) & (h.nslots - 1); continue; }
-      uint64_t prev = atomicCAS(&h.e[slot].key, 0ull, (unsigned long long)key);
-      if (prev != 0 && prev != key) { slot = (slot + 1) & (h.nslots - 1); continue; }
-    }
-    atomicMax(&h.e[slot].minp_enc, ~(uint32_t)i);
-    slots[i] = (uint32_t)slot;
-    return;
-  }
-  atomicExch(h.overflow, 1u);
-  slots[i] = 0;
+  uint32_t slot = weld_slot_of_key(keys_sorted[i], sx, sy, nvox);
+  slots[i] = slot;
+  atomicMax(&wminp[slot], ~(uint32_t)i);
 }
 
 // scan-input functor: 1 where this corner is its key's first occurrence
 struct FirstOccur {
   const uint32_t *slots;
-  const WeldEntry *e;
+  const uint32_t *wminp;
   __device__ uint32_t operator()(uint32_t i) const {
-    return (e[slots[i]].minp_enc == ~i) ? 1u : 0u;
+    return (wminp[slots[i]] == ~i) ? 1u : 0u;
   }
 };
 
 // total verts = vtx_scan[NC-1] + first_occur(NC-1)
 __global__ void k_total_verts(const uint32_t *__restrict__ vtx_scan,
                               const uint32_t *__restrict__ slots,
-                              const WeldEntry *__restrict__ e,
+                              const uint32_t *__restrict__ wminp,
                               uint64_t ncorners, uint32_t *out) {
   uint32_t i = (uint32_t)(ncorners - 1);
-  *out = vtx_scan[i] + ((e[slots[i]].minp_enc == ~i) ? 1u : 0u);
+  *out = vtx_scan[i] + ((wminp[slots[i]] == ~i) ? 1u : 0u);
 }
 
-// [5d] first occurrences: record vertex id in the entry, write the vertex
+// [5d] first occurrences: record vertex id in the table, write the vertex
 __global__ void k_weld_verts(const uint64_t *__restrict__ keys_sorted,
                              const uint32_t *__restrict__ slots,
                              const uint32_t *__restrict__ vtx_scan,
-                             WeldHash h, float *__restrict__ verts,
+                             const uint32_t *__restrict__ wminp,
+                             uint32_t *__restrict__ wvtx,
+                             float *__restrict__ verts,
                              float rx, float ry, float rz, float shift,
                              uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  WeldEntry &ent = h.e[slots[i]];
-  if (ent.minp_enc != ~(uint32_t)i) return;
+  uint32_t slot = slots[i];
+  if (wminp[slot] != ~(uint32_t)i) return;
   uint32_t v = vtx_scan[i];
-  ent.vtx = v;
+  wvtx[slot] = v;
   uint64_t key = keys_sorted[i];
   float dx = (float)(uint32_t)(key & 0xFFF);
   float dy = (float)(uint32_t)((key >> 12) & 0xFFF);
@@ -401,15 +408,15 @@ __global__ void k_vbase(const uint32_t *__restrict__ tri_off,
 
 // [5f] faces: per-label local vertex indices
 __global__ void k_faces(const uint64_t *__restrict__ keys_sorted,
-                        const WeldHash h,
+                        const uint32_t *__restrict__ wvtx,
                         const uint32_t *__restrict__ slots,
                         const uint32_t *__restrict__ vbase,
                         uint32_t *__restrict__ faces,
                         uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  uint32_t lid = (uint32_t)(keys_sorted[i] >> 36);
-  faces[i] = h.e[slots[i]].vtx - vbase[lid];
+  uint32_t lid = (uint32_t)(keys_sorted[i] >> 37);
+  faces[i] = wvtx[slots[i]] - vbase[lid];
 }
 
 __global__ void k_iota(uint32_t *p, uint64_t n) {
@@ -697,6 +704,10 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
             "split the task shape", (unsigned long long)total_tris);
     return 15;
   }
+  if (nlabels >= (1u << 27)) {  // label id shares the 64-bit weld key
+    SET_ERR(c, "%u labels exceed the 2^27 per-chunk label limit", nlabels);
+    return 15;
+  }
   if (reduction_factor > 1) {
     SET_ERR(c, "GPU simplifier not yet wired (reduction_factor=%u); "
             "round-1 engine supports reduction_factor<=1",
@@ -779,36 +790,31 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   HIP_TRY(c, hipGetLastError(), 19);
   HIP_TRY(c, hipEventRecord(c->ev[5], s), 19);
 
-  // [5] weld — grow-on-overflow retry; packed 16B entries, zero-memset
-  uint64_t wh_slots = next_pow2_u64(std::max<uint64_t>(1024, NC / 2));
+  // [5] weld — direct-addressed (edge, side) table, collision-free
+  const uint64_t wslots = 6 * nvox;  // 3 edges/voxel x 2 sides
+  if (wslots >= (1ull << 32)) {
+    SET_ERR(c, "chunk too large for the 32-bit weld table (%d x %d x %d); "
+            "split the task shape (the reference's own mesher bound is "
+            "1023x1023x511, igneous_cli/cli.py:1049-1052)", sx, sy, sz);
+    return 20;
+  }
   uint64_t total_verts = 0;
-  WeldHash wh;
-  for (;;) {
-    if (ensure(c, c->wh_keys, wh_slots * sizeof(WeldEntry))) return 20;
-    if (ensure(c, c->flags, NC * 4)) return 20;  // slot ids
-    if (ensure(c, c->vtx_scan, NC * 4)) return 20;
-    HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0,
-                              wh_slots * sizeof(WeldEntry), s), 20);
-    HIP_TRY(c, hipMemsetAsync((uint32_t *)c->lh_misc.ptr + 2, 0, 4, s), 20);
-    wh.e = (WeldEntry *)c->wh_keys.ptr;
-    wh.overflow = (uint32_t *)c->lh_misc.ptr + 2;
-    wh.nslots = wh_slots;
+  if (ensure(c, c->wh_keys, wslots * 4)) return 20;   // wminp
+  if (ensure(c, c->wh_vtx, wslots * 4)) return 20;    // wvtx
+  if (ensure(c, c->flags, NC * 4)) return 20;         // slot ids
+  if (ensure(c, c->vtx_scan, NC * 4)) return 20;
+  HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0, wslots * 4, s), 20);
+  uint32_t *wminp = (uint32_t *)c->wh_keys.ptr;
+  uint32_t *wvtx = (uint32_t *)c->wh_vtx.ptr;
+  {
     int blk = 256;
     uint64_t nb = (NC + blk - 1) / blk;
     hipLaunchKernelGGL(k_weld_insert, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint64_t *)c->keys_sorted.ptr, wh,
-                       (uint32_t *)c->flags.ptr, NC);
-    uint32_t wh_ovf = 0;
-    HIP_TRY(c, hipMemcpyAsync(&wh_ovf, (uint32_t *)c->lh_misc.ptr + 2, 4,
-                              hipMemcpyDeviceToHost, s), 20);
-    HIP_TRY(c, hipStreamSynchronize(s), 20);
-    if (wh_ovf) {
-      if (wh_slots >= (1ull << 32)) { SET_ERR(c, "weld hash overflow"); return 20; }
-      wh_slots <<= 1;
-      continue;
-    }
+                       (const uint64_t *)c->keys_sorted.ptr, wminp,
+                       (uint32_t *)c->flags.ptr, g.sx, g.sy, (int64_t)nvox,
+                       NC);
     // scan first-occurrence flags -> vertex ids (flags computed on the fly)
-    FirstOccur fo{(const uint32_t *)c->flags.ptr, (const WeldEntry *)c->wh_keys.ptr};
+    FirstOccur fo{(const uint32_t *)c->flags.ptr, wminp};
     auto it = rocprim::make_transform_iterator(
         rocprim::counting_iterator<uint32_t>(0), fo);
     size_t tmp_bytes = 0;
@@ -823,15 +829,13 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     if (e != hipSuccess) { SET_ERR(c, "weld scan failed"); return 20; }
     hipLaunchKernelGGL(k_total_verts, dim3(1), dim3(1), 0, s,
                        (const uint32_t *)c->vtx_scan.ptr,
-                       (const uint32_t *)c->flags.ptr,
-                       (const WeldEntry *)c->wh_keys.ptr, NC,
+                       (const uint32_t *)c->flags.ptr, wminp, NC,
                        (uint32_t *)c->lh_misc.ptr + 3);
     uint32_t tv = 0;
     HIP_TRY(c, hipMemcpyAsync(&tv, (uint32_t *)c->lh_misc.ptr + 3, 4,
                               hipMemcpyDeviceToHost, s), 21);
     HIP_TRY(c, hipStreamSynchronize(s), 21);
     total_verts = tv;
-    break;
   }
   c->stats.total_verts = total_verts;
 
@@ -845,7 +849,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nb), dim3(blk), 0, s,
                        (const uint64_t *)c->keys_sorted.ptr,
                        (const uint32_t *)c->flags.ptr,
-                       (const uint32_t *)c->vtx_scan.ptr, wh,
+                       (const uint32_t *)c->vtx_scan.ptr, wminp, wvtx,
                        (float *)c->verts.ptr, rx, ry, rz, shift, NC);
     uint32_t nbl = (nlabels + 1 + 255) / 256;
     hipLaunchKernelGGL(k_vbase, dim3(nbl), dim3(256), 0, s,
@@ -853,7 +857,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                        (const uint32_t *)c->vtx_scan.ptr,
                        (uint32_t *)c->vbase.ptr, nlabels, total_verts);
     hipLaunchKernelGGL(k_faces, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint64_t *)c->keys_sorted.ptr, wh,
+                       (const uint64_t *)c->keys_sorted.ptr, wvtx,
                        (const uint32_t *)c->flags.ptr,
                        (const uint32_t *)c->vbase.ptr,
                        (uint32_t *)c->faces.ptr, NC);

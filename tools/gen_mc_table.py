@@ -183,6 +183,11 @@ def main():
         for e in range(12):
             f.write("  {%d, %d, %d},\n" % EDGE_DOFF[e])
         f.write("};\n\n")
+        f.write("// corner ids (a,b) of each edge, a < b (a = lower voxel)\n")
+        f.write("MC_TABLE_QUAL unsigned char MC_EDGE_CORNERS[12][2] = {\n")
+        for (a, b) in EDGES:
+            f.write(f"  {{{a}, {b}}},\n")
+        f.write("};\n\n")
         f.write("// packed 12-bit-per-axis key offset of each edge midpoint\n")
         f.write("// relative to the cell's packed doubled origin:\n")
         f.write("// key = ((2cz)<<24 | (2cy)<<12 | 2cx) + MC_EDGE_KEYOFF[e]\n")
