@@ -216,12 +216,16 @@ __global__ void k_count(const T *__restrict__ labels, GridDims g,
   }
 }
 
-// [3] emit: recompute, wave prefix, write triangle records in canonical order
+// [3] emit: recompute, wave prefix, write triangle records in canonical
+// order. Per corner we emit the 32-bit WELD SLOT (edge axis, voxel, side)
+// — a complete key: an edge midpoint is a vertex only for its two
+// endpoint labels, the label id lives per-triangle in tri_label, and the
+// doubled coordinates are recoverable from the slot.
 template <typename T>
 __global__ void k_emit(const T *__restrict__ labels, GridDims g,
                        const uint32_t *__restrict__ segoff, LabelHash lh,
                        uint32_t *__restrict__ tri_label,
-                       uint64_t *__restrict__ tri_keys) {
+                       uint32_t *__restrict__ tri_slots) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_in_blk = threadIdx.x / WAVE;
   const int waves_per_blk = blockDim.x / WAVE;
@@ -245,8 +249,7 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
     uint32_t incl = wave_incl_scan(cnt, lane);
     uint32_t base = segoff[seg] + incl - cnt;
     if (active) {
-      uint64_t cellkey = ((uint64_t)(2 * cz) << 24) |
-                         ((uint64_t)(2 * cy) << 12) | (uint64_t)(2 * cx);
+      const int64_t nvox = g.sx * g.sy * g.sz;
       uint32_t pos = base;
       #pragma unroll
       for (int i = 0; i < 8; ++i) {
@@ -261,20 +264,25 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
         uint32_t nt = MC_TRI_COUNT[mask];
         if (!nt) continue;
         uint32_t lid = label_lookup(lh, (uint64_t)L);
-        uint64_t lid_hi = (uint64_t)lid << 37;
         const signed char *tt = MC_TRI_TABLE[mask];
         for (uint32_t t = 0; t < nt; ++t) {
           tri_label[pos] = lid;
           #pragma unroll
           for (int v = 0; v < 3; ++v) {
             int e = tt[3 * t + v];
-            // side bit: is L the edge's UPPER endpoint label? The weld
-            // table has exactly two slots per edge (an edge midpoint is a
-            // vertex only for its two endpoint labels) — collision-free
-            // direct addressing, no hash.
-            uint64_t side = (c[MC_EDGE_CORNERS[e][1]] == L) ? 1ull : 0ull;
-            tri_keys[3 * (uint64_t)pos + v] =
-                lid_hi | (side << 36) | (cellkey + MC_EDGE_KEYOFF[e]);
+            // slot = (axis, lower-endpoint voxel, side): side = is L the
+            // edge's UPPER endpoint label? Collision-free: a midpoint is
+            // a vertex only for its two endpoint labels.
+            int dx = MC_EDGE_DOFF[e][0], dy = MC_EDGE_DOFF[e][1],
+                dz = MC_EDGE_DOFF[e][2];
+            int axis = (dx & 1) ? 0 : ((dy & 1) ? 1 : 2);
+            int64_t vx = cx + (dx >> 1), vy = cy + (dy >> 1),
+                    vz = cz + (dz >> 1);
+            uint32_t side = (c[MC_EDGE_CORNERS[e][1]] == L) ? 1u : 0u;
+            uint32_t slot = (uint32_t)(
+                (((uint64_t)axis * nvox + (vz * g.sy + vy) * g.sx + vx)
+                 << 1) | side);
+            tri_slots[3 * (uint64_t)pos + v] = slot;
           }
           ++pos;
         }
@@ -283,18 +291,18 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
   }
 }
 
-// [4b] gather triangle keys into label-partitioned order
-__global__ void k_gather_keys(const uint64_t *__restrict__ tri_keys,
+// [4b] gather per-corner weld slots into label-partitioned order
+__global__ void k_gather_keys(const uint32_t *__restrict__ tri_slots,
                               const uint32_t *__restrict__ order,
-                              uint64_t *__restrict__ keys_sorted,
+                              uint32_t *__restrict__ slots_sorted,
                               uint64_t ntris) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ntris) return;
   uint64_t src = 3ull * order[i];
   uint64_t dst = 3ull * i;
-  keys_sorted[dst + 0] = tri_keys[src + 0];
-  keys_sorted[dst + 1] = tri_keys[src + 1];
-  keys_sorted[dst + 2] = tri_keys[src + 2];
+  slots_sorted[dst + 0] = tri_slots[src + 0];
+  slots_sorted[dst + 1] = tri_slots[src + 1];
+  slots_sorted[dst + 2] = tri_slots[src + 2];
 }
 
 // [4c] per-label triangle ranges (labels sorted, every id present)
@@ -326,31 +334,13 @@ __global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
 // host; matches the reference's own 32-bit mesher task bound,
 // igneous_cli/cli.py:1049-1052).
 
-__device__ __forceinline__ uint32_t weld_slot_of_key(uint64_t key,
-                                                     int64_t sx, int64_t sy,
-                                                     int64_t nvox) {
-  uint32_t kx = (uint32_t)(key & 0xFFF);
-  uint32_t ky = (uint32_t)((key >> 12) & 0xFFF);
-  uint32_t kz = (uint32_t)((key >> 24) & 0xFFF);
-  int axis = (kx & 1) ? 0 : ((ky & 1) ? 1 : 2);
-  int64_t vx = kx >> 1, vy = ky >> 1, vz = kz >> 1;
-  int64_t lin = (vz * sy + vy) * sx + vx;
-  uint64_t side = (key >> 36) & 1;
-  return (uint32_t)((((uint64_t)axis * nvox + lin) << 1) | side);
-}
-
-// [5a] record first (minimum) stream position per (edge, side); store the
-// slot per corner so later passes never recompute the address chain.
-__global__ void k_weld_insert(const uint64_t *__restrict__ keys_sorted,
+// [5a] record first (minimum) stream position per (edge, side)
+__global__ void k_weld_insert(const uint32_t *__restrict__ slots_sorted,
                               uint32_t *__restrict__ wminp,
-                              uint32_t *__restrict__ slots,
-                              int64_t sx, int64_t sy, int64_t nvox,
                               uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  uint32_t slot = weld_slot_of_key(keys_sorted[i], sx, sy, nvox);
-  slots[i] = slot;
-  atomicMax(&wminp[slot], ~(uint32_t)i);
+  atomicMax(&wminp[slots_sorted[i]], ~(uint32_t)i);
 }
 
 // scan-input functor: 1 where this corner is its key's first occurrence
@@ -372,24 +362,32 @@ __global__ void k_total_verts(const uint32_t *__restrict__ vtx_scan,
 }
 
 // [5d] first occurrences: record vertex id in the table, write the vertex
-__global__ void k_weld_verts(const uint64_t *__restrict__ keys_sorted,
-                             const uint32_t *__restrict__ slots,
+// (doubled coordinates decoded from the slot: axis, voxel, +1 on axis)
+__global__ void k_weld_verts(const uint32_t *__restrict__ slots_sorted,
                              const uint32_t *__restrict__ vtx_scan,
                              const uint32_t *__restrict__ wminp,
                              uint32_t *__restrict__ wvtx,
                              float *__restrict__ verts,
+                             int64_t sx, int64_t sy, int64_t nvox,
                              float rx, float ry, float rz, float shift,
                              uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  uint32_t slot = slots[i];
+  uint32_t slot = slots_sorted[i];
   if (wminp[slot] != ~(uint32_t)i) return;
   uint32_t v = vtx_scan[i];
   wvtx[slot] = v;
-  uint64_t key = keys_sorted[i];
-  float dx = (float)(uint32_t)(key & 0xFFF);
-  float dy = (float)(uint32_t)((key >> 12) & 0xFFF);
-  float dz = (float)(uint32_t)((key >> 24) & 0xFFF);
+  int64_t eslot = (int64_t)(slot >> 1);
+  int axis = (int)(eslot / nvox);
+  int64_t lin = eslot - (int64_t)axis * nvox;
+  int64_t vz = lin / (sx * sy);
+  int64_t rem = lin - vz * sx * sy;
+  int64_t vy = rem / sx;
+  int64_t vx = rem - vy * sx;
+  // doubled coords: 2*voxel, +1 on the edge's axis
+  float dx = (float)(2 * vx + (axis == 0));
+  float dy = (float)(2 * vy + (axis == 1));
+  float dz = (float)(2 * vz + (axis == 2));
   verts[3ull * v + 0] = (0.5f * dx + shift) * rx;
   verts[3ull * v + 1] = (0.5f * dy + shift) * ry;
   verts[3ull * v + 2] = (0.5f * dz + shift) * rz;
@@ -406,17 +404,18 @@ __global__ void k_vbase(const uint32_t *__restrict__ tri_off,
   else vbase[l] = vtx_scan[3ull * tri_off[l]];
 }
 
-// [5f] faces: per-label local vertex indices
-__global__ void k_faces(const uint64_t *__restrict__ keys_sorted,
+// [5f] faces: per-label local vertex indices (label id read per triangle
+// from the partition-sorted label array)
+__global__ void k_faces(const uint32_t *__restrict__ slots_sorted,
+                        const uint32_t *__restrict__ lab_sorted,
                         const uint32_t *__restrict__ wvtx,
-                        const uint32_t *__restrict__ slots,
                         const uint32_t *__restrict__ vbase,
                         uint32_t *__restrict__ faces,
                         uint64_t ncorners) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ncorners) return;
-  uint32_t lid = (uint32_t)(keys_sorted[i] >> 37);
-  faces[i] = wvtx[slots[i]] - vbase[lid];
+  uint32_t lid = lab_sorted[i / 3];
+  faces[i] = wvtx[slots_sorted[i]] - vbase[lid];
 }
 
 __global__ void k_iota(uint32_t *p, uint64_t n) {
@@ -729,7 +728,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
 
   // [3] emit
   if (ensure(c, c->tri_label, T * 4)) return 18;
-  if (ensure(c, c->tri_keys, NC * 8)) return 18;
+  if (ensure(c, c->tri_keys, NC * 4)) return 18;
   {
     int blk = 256;
     int waves_per_blk = blk / WAVE;
@@ -740,13 +739,13 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                          (const uint64_t *)c->labels.ptr, g,
                          (const uint32_t *)c->segoff.ptr, lh,
                          (uint32_t *)c->tri_label.ptr,
-                         (uint64_t *)c->tri_keys.ptr);
+                         (uint32_t *)c->tri_keys.ptr);
     else
       hipLaunchKernelGGL(k_emit<uint32_t>, dim3((uint32_t)nb), dim3(blk), 0, s,
                          (const uint32_t *)c->labels.ptr, g,
                          (const uint32_t *)c->segoff.ptr, lh,
                          (uint32_t *)c->tri_label.ptr,
-                         (uint64_t *)c->tri_keys.ptr);
+                         (uint32_t *)c->tri_keys.ptr);
   }
   HIP_TRY(c, hipGetLastError(), 18);
   HIP_TRY(c, hipEventRecord(c->ev[4], s), 18);
@@ -755,7 +754,8 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   if (ensure(c, c->order, T * 4)) return 19;
   if (ensure(c, c->order_alt, T * 4)) return 19;
   if (ensure(c, c->tri_label_alt, T * 4)) return 19;
-  if (ensure(c, c->keys_sorted, NC * 8)) return 19;
+  if (ensure(c, c->keys_sorted, NC * 4)) return 19;
+  uint32_t *lab_sorted = nullptr;
   {
     int blk = 256;
     uint64_t nb = (T + blk - 1) / blk;
@@ -779,13 +779,14 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     if (e != hipSuccess) { SET_ERR(c, "radix_sort failed"); return 19; }
     // gather
     hipLaunchKernelGGL(k_gather_keys, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint64_t *)c->tri_keys.ptr, d_vals.current(),
-                       (uint64_t *)c->keys_sorted.ptr, T);
+                       (const uint32_t *)c->tri_keys.ptr, d_vals.current(),
+                       (uint32_t *)c->keys_sorted.ptr, T);
     // label ranges
     if (ensure(c, c->tri_off, ((uint64_t)nlabels + 1) * 4)) return 19;
     hipLaunchKernelGGL(k_label_ranges, dim3((uint32_t)nb), dim3(blk), 0, s,
                        d_keys.current(), (uint32_t *)c->tri_off.ptr, T,
                        nlabels);
+    lab_sorted = d_keys.current();
   }
   HIP_TRY(c, hipGetLastError(), 19);
   HIP_TRY(c, hipEventRecord(c->ev[5], s), 19);
@@ -801,20 +802,18 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   uint64_t total_verts = 0;
   if (ensure(c, c->wh_keys, wslots * 4)) return 20;   // wminp
   if (ensure(c, c->wh_vtx, wslots * 4)) return 20;    // wvtx
-  if (ensure(c, c->flags, NC * 4)) return 20;         // slot ids
   if (ensure(c, c->vtx_scan, NC * 4)) return 20;
   HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0, wslots * 4, s), 20);
   uint32_t *wminp = (uint32_t *)c->wh_keys.ptr;
   uint32_t *wvtx = (uint32_t *)c->wh_vtx.ptr;
+  uint32_t *slots_sorted = (uint32_t *)c->keys_sorted.ptr;
   {
     int blk = 256;
     uint64_t nb = (NC + blk - 1) / blk;
     hipLaunchKernelGGL(k_weld_insert, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint64_t *)c->keys_sorted.ptr, wminp,
-                       (uint32_t *)c->flags.ptr, g.sx, g.sy, (int64_t)nvox,
-                       NC);
+                       slots_sorted, wminp, NC);
     // scan first-occurrence flags -> vertex ids (flags computed on the fly)
-    FirstOccur fo{(const uint32_t *)c->flags.ptr, wminp};
+    FirstOccur fo{slots_sorted, wminp};
     auto it = rocprim::make_transform_iterator(
         rocprim::counting_iterator<uint32_t>(0), fo);
     size_t tmp_bytes = 0;
@@ -829,7 +828,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     if (e != hipSuccess) { SET_ERR(c, "weld scan failed"); return 20; }
     hipLaunchKernelGGL(k_total_verts, dim3(1), dim3(1), 0, s,
                        (const uint32_t *)c->vtx_scan.ptr,
-                       (const uint32_t *)c->flags.ptr, wminp, NC,
+                       slots_sorted, wminp, NC,
                        (uint32_t *)c->lh_misc.ptr + 3);
     uint32_t tv = 0;
     HIP_TRY(c, hipMemcpyAsync(&tv, (uint32_t *)c->lh_misc.ptr + 3, 4,
@@ -847,18 +846,17 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     uint64_t nb = (NC + blk - 1) / blk;
     const float shift = voxel_centered ? 0.0f : 0.5f;
     hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint64_t *)c->keys_sorted.ptr,
-                       (const uint32_t *)c->flags.ptr,
+                       slots_sorted,
                        (const uint32_t *)c->vtx_scan.ptr, wminp, wvtx,
-                       (float *)c->verts.ptr, rx, ry, rz, shift, NC);
+                       (float *)c->verts.ptr, g.sx, g.sy, (int64_t)nvox,
+                       rx, ry, rz, shift, NC);
     uint32_t nbl = (nlabels + 1 + 255) / 256;
     hipLaunchKernelGGL(k_vbase, dim3(nbl), dim3(256), 0, s,
                        (const uint32_t *)c->tri_off.ptr,
                        (const uint32_t *)c->vtx_scan.ptr,
                        (uint32_t *)c->vbase.ptr, nlabels, total_verts);
     hipLaunchKernelGGL(k_faces, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint64_t *)c->keys_sorted.ptr, wvtx,
-                       (const uint32_t *)c->flags.ptr,
+                       slots_sorted, lab_sorted, wvtx,
                        (const uint32_t *)c->vbase.ptr,
                        (uint32_t *)c->faces.ptr, NC);
   }
