@@ -279,8 +279,10 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
             int64_t vx = cx + (dx >> 1), vy = cy + (dy >> 1),
                     vz = cz + (dz >> 1);
             uint32_t side = (c[MC_EDGE_CORNERS[e][1]] == L) ? 1u : 0u;
+            // voxel-interleaved layout: a voxel's 3 edge slots are
+            // adjacent, so a cell's corners touch few cache lines
             uint32_t slot = (uint32_t)(
-                (((uint64_t)axis * nvox + (vz * g.sy + vy) * g.sx + vx)
+                (((uint64_t)((vz * g.sy + vy) * g.sx + vx) * 3 + axis)
                  << 1) | side);
             tri_slots[3 * (uint64_t)pos + v] = slot;
           }
@@ -378,8 +380,8 @@ __global__ void k_weld_verts(const uint32_t *__restrict__ slots_sorted,
   uint32_t v = vtx_scan[i];
   wvtx[slot] = v;
   int64_t eslot = (int64_t)(slot >> 1);
-  int axis = (int)(eslot / nvox);
-  int64_t lin = eslot - (int64_t)axis * nvox;
+  int axis = (int)(eslot % 3);
+  int64_t lin = eslot / 3;
   int64_t vz = lin / (sx * sy);
   int64_t rem = lin - vz * sx * sy;
   int64_t vy = rem / sx;
