@@ -225,7 +225,7 @@ template <typename T>
 __global__ void k_emit(const T *__restrict__ labels, GridDims g,
                        const uint32_t *__restrict__ segoff, LabelHash lh,
                        uint32_t *__restrict__ tri_label,
-                       uint32_t *__restrict__ tri_slots) {
+                       uint4 *__restrict__ tri_recs) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_in_blk = threadIdx.x / WAVE;
   const int waves_per_blk = blockDim.x / WAVE;
@@ -249,7 +249,6 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
     uint32_t incl = wave_incl_scan(cnt, lane);
     uint32_t base = segoff[seg] + incl - cnt;
     if (active) {
-      const int64_t nvox = g.sx * g.sy * g.sz;
       uint32_t pos = base;
       #pragma unroll
       for (int i = 0; i < 8; ++i) {
@@ -265,27 +264,35 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
         if (!nt) continue;
         uint32_t lid = label_lookup(lh, (uint64_t)L);
         const signed char *tt = MC_TRI_TABLE[mask];
+        // 32-bit voxel linear index of the cell origin (6*nvox < 2^32
+        // is enforced on the host)
+        const uint32_t cell_lin =
+            (uint32_t)((cz * g.sy + cy) * g.sx + cx);
+        const uint32_t usx = (uint32_t)g.sx;
+        const uint32_t usxy = (uint32_t)(g.sx * g.sy);
         for (uint32_t t = 0; t < nt; ++t) {
           tri_label[pos] = lid;
+          uint4 rec;
+          uint32_t *rs = &rec.x;
           #pragma unroll
           for (int v = 0; v < 3; ++v) {
             int e = tt[3 * t + v];
-            // slot = (axis, lower-endpoint voxel, side): side = is L the
-            // edge's UPPER endpoint label? Collision-free: a midpoint is
-            // a vertex only for its two endpoint labels.
+            // slot = (lower-endpoint voxel, axis, side), voxel-
+            // interleaved so a cell's corners share cache lines.
+            // side = is L the edge's UPPER endpoint label? Collision-
+            // free: a midpoint is a vertex only for its two endpoint
+            // labels.
             int dx = MC_EDGE_DOFF[e][0], dy = MC_EDGE_DOFF[e][1],
                 dz = MC_EDGE_DOFF[e][2];
-            int axis = (dx & 1) ? 0 : ((dy & 1) ? 1 : 2);
-            int64_t vx = cx + (dx >> 1), vy = cy + (dy >> 1),
-                    vz = cz + (dz >> 1);
+            uint32_t axis = (dx & 1) ? 0u : ((dy & 1) ? 1u : 2u);
+            uint32_t lin = cell_lin + (uint32_t)(dx >> 1) +
+                           (uint32_t)(dy >> 1) * usx +
+                           (uint32_t)(dz >> 1) * usxy;
             uint32_t side = (c[MC_EDGE_CORNERS[e][1]] == L) ? 1u : 0u;
-            // voxel-interleaved layout: a voxel's 3 edge slots are
-            // adjacent, so a cell's corners touch few cache lines
-            uint32_t slot = (uint32_t)(
-                (((uint64_t)((vz * g.sy + vy) * g.sx + vx) * 3 + axis)
-                 << 1) | side);
-            tri_slots[3 * (uint64_t)pos + v] = slot;
+            rs[v] = ((lin * 3u + axis) << 1) | side;
           }
+          rec.w = lid;
+          tri_recs[pos] = rec;  // one 16-B store per triangle
           ++pos;
         }
       }
@@ -293,18 +300,14 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
   }
 }
 
-// [4b] gather per-corner weld slots into label-partitioned order
-__global__ void k_gather_keys(const uint32_t *__restrict__ tri_slots,
+// [4b] gather triangle records into label-partitioned order (16 B each)
+__global__ void k_gather_recs(const uint4 *__restrict__ tri_recs,
                               const uint32_t *__restrict__ order,
-                              uint32_t *__restrict__ slots_sorted,
+                              uint4 *__restrict__ recs_sorted,
                               uint64_t ntris) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= ntris) return;
-  uint64_t src = 3ull * order[i];
-  uint64_t dst = 3ull * i;
-  slots_sorted[dst + 0] = tri_slots[src + 0];
-  slots_sorted[dst + 1] = tri_slots[src + 1];
-  slots_sorted[dst + 2] = tri_slots[src + 2];
+  recs_sorted[i] = tri_recs[order[i]];
 }
 
 // [4c] per-label triangle ranges (labels sorted, every id present)
@@ -336,63 +339,111 @@ __global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
 // host; matches the reference's own 32-bit mesher task bound,
 // igneous_cli/cli.py:1049-1052).
 
-// [5a] record first (minimum) stream position per (edge, side)
-__global__ void k_weld_insert(const uint32_t *__restrict__ slots_sorted,
-                              uint32_t *__restrict__ wminp,
-                              uint64_t ncorners) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= ncorners) return;
-  atomicMax(&wminp[slots_sorted[i]], ~(uint32_t)i);
+// [5a] record first (minimum) stream position per (edge, side).
+// One thread per TRIANGLE (reads its 16-B record once). A per-workgroup
+// LDS table pre-merges duplicate slots within the block's 768-corner
+// window (the label-partitioned stream is spatially coherent, so ~half
+// the corners repeat a slot seen moments earlier) — cutting the global
+// atomic count roughly in half.
+#define WI_LDS_SLOTS 2048
+__global__ __launch_bounds__(256) void k_weld_insert(
+    const uint4 *__restrict__ recs_sorted, uint32_t *__restrict__ wminp,
+    uint64_t ntris) {
+  __shared__ uint32_t lkey[WI_LDS_SLOTS];
+  __shared__ uint32_t lval[WI_LDS_SLOTS];
+  for (int k = threadIdx.x; k < WI_LDS_SLOTS; k += blockDim.x) {
+    lkey[k] = 0;
+    lval[k] = 0;
+  }
+  __syncthreads();
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t < ntris) {
+    uint4 rec = recs_sorted[t];
+    const uint32_t s[3] = {rec.x, rec.y, rec.z};
+    #pragma unroll
+    for (int v = 0; v < 3; ++v) {
+      uint32_t slot = s[v];
+      uint32_t enc = ~(uint32_t)(3 * t + v);
+      uint32_t key = slot + 1;
+      uint32_t h = (slot * 2654435761u) >> (32 - 11);
+      bool placed = false;
+      for (int probe = 0; probe < 32; ++probe) {
+        uint32_t cur = lkey[h];
+        if (cur == 0) cur = atomicCAS(&lkey[h], 0u, key);
+        if (cur == 0 || cur == key) {
+          atomicMax(&lval[h], enc);
+          placed = true;
+          break;
+        }
+        h = (h + 1) & (WI_LDS_SLOTS - 1);
+      }
+      if (!placed) atomicMax(&wminp[slot], enc);  // rare spill
+    }
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < WI_LDS_SLOTS; k += blockDim.x) {
+    uint32_t key = lkey[k];
+    if (key) atomicMax(&wminp[key - 1], lval[k]);
+  }
 }
 
 // scan-input functor: 1 where this corner is its key's first occurrence
 struct FirstOccur {
-  const uint32_t *slots;
+  const uint4 *recs;
   const uint32_t *wminp;
   __device__ uint32_t operator()(uint32_t i) const {
-    return (wminp[slots[i]] == ~i) ? 1u : 0u;
+    uint4 rec = recs[i / 3];
+    uint32_t slot = (i % 3 == 0) ? rec.x : ((i % 3 == 1) ? rec.y : rec.z);
+    return (wminp[slot] == ~i) ? 1u : 0u;
   }
 };
 
 // total verts = vtx_scan[NC-1] + first_occur(NC-1)
 __global__ void k_total_verts(const uint32_t *__restrict__ vtx_scan,
-                              const uint32_t *__restrict__ slots,
+                              const uint4 *__restrict__ recs,
                               const uint32_t *__restrict__ wminp,
                               uint64_t ncorners, uint32_t *out) {
   uint32_t i = (uint32_t)(ncorners - 1);
-  *out = vtx_scan[i] + ((wminp[slots[i]] == ~i) ? 1u : 0u);
+  uint4 rec = recs[i / 3];
+  uint32_t slot = (i % 3 == 0) ? rec.x : ((i % 3 == 1) ? rec.y : rec.z);
+  *out = vtx_scan[i] + ((wminp[slot] == ~i) ? 1u : 0u);
 }
 
 // [5d] first occurrences: record vertex id in the table, write the vertex
-// (doubled coordinates decoded from the slot: axis, voxel, +1 on axis)
-__global__ void k_weld_verts(const uint32_t *__restrict__ slots_sorted,
+// (doubled coordinates decoded from the slot: voxel, axis, +1 on axis)
+__global__ void k_weld_verts(const uint4 *__restrict__ recs_sorted,
                              const uint32_t *__restrict__ vtx_scan,
                              const uint32_t *__restrict__ wminp,
                              uint32_t *__restrict__ wvtx,
                              float *__restrict__ verts,
-                             int64_t sx, int64_t sy, int64_t nvox,
+                             uint32_t usx, uint32_t usxy,
                              float rx, float ry, float rz, float shift,
-                             uint64_t ncorners) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= ncorners) return;
-  uint32_t slot = slots_sorted[i];
-  if (wminp[slot] != ~(uint32_t)i) return;
-  uint32_t v = vtx_scan[i];
-  wvtx[slot] = v;
-  int64_t eslot = (int64_t)(slot >> 1);
-  int axis = (int)(eslot % 3);
-  int64_t lin = eslot / 3;
-  int64_t vz = lin / (sx * sy);
-  int64_t rem = lin - vz * sx * sy;
-  int64_t vy = rem / sx;
-  int64_t vx = rem - vy * sx;
-  // doubled coords: 2*voxel, +1 on the edge's axis
-  float dx = (float)(2 * vx + (axis == 0));
-  float dy = (float)(2 * vy + (axis == 1));
-  float dz = (float)(2 * vz + (axis == 2));
-  verts[3ull * v + 0] = (0.5f * dx + shift) * rx;
-  verts[3ull * v + 1] = (0.5f * dy + shift) * ry;
-  verts[3ull * v + 2] = (0.5f * dz + shift) * rz;
+                             uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  uint4 rec = recs_sorted[t];
+  const uint32_t s[3] = {rec.x, rec.y, rec.z};
+  #pragma unroll
+  for (int v = 0; v < 3; ++v) {
+    uint32_t i = (uint32_t)(3 * t + v);
+    uint32_t slot = s[v];
+    if (wminp[slot] != ~i) continue;
+    uint32_t vid = vtx_scan[i];
+    wvtx[slot] = vid;
+    uint32_t eslot = slot >> 1;
+    uint32_t axis = eslot % 3u;
+    uint32_t lin = eslot / 3u;
+    uint32_t vz = lin / usxy;
+    uint32_t rem = lin - vz * usxy;
+    uint32_t vy = rem / usx;
+    uint32_t vx = rem - vy * usx;
+    float dx = (float)(2 * vx + (axis == 0));
+    float dy = (float)(2 * vy + (axis == 1));
+    float dz = (float)(2 * vz + (axis == 2));
+    verts[3ull * vid + 0] = (0.5f * dx + shift) * rx;
+    verts[3ull * vid + 1] = (0.5f * dy + shift) * ry;
+    verts[3ull * vid + 2] = (0.5f * dz + shift) * rz;
+  }
 }
 
 // [5e] per-label vertex bases: vbase[l] = vtx_scan at the label's first corner
@@ -406,18 +457,20 @@ __global__ void k_vbase(const uint32_t *__restrict__ tri_off,
   else vbase[l] = vtx_scan[3ull * tri_off[l]];
 }
 
-// [5f] faces: per-label local vertex indices (label id read per triangle
-// from the partition-sorted label array)
-__global__ void k_faces(const uint32_t *__restrict__ slots_sorted,
-                        const uint32_t *__restrict__ lab_sorted,
+// [5f] faces: per-label local vertex indices (the record carries its
+// label id in .w)
+__global__ void k_faces(const uint4 *__restrict__ recs_sorted,
                         const uint32_t *__restrict__ wvtx,
                         const uint32_t *__restrict__ vbase,
                         uint32_t *__restrict__ faces,
-                        uint64_t ncorners) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= ncorners) return;
-  uint32_t lid = lab_sorted[i / 3];
-  faces[i] = wvtx[slots_sorted[i]] - vbase[lid];
+                        uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  uint4 rec = recs_sorted[t];
+  uint32_t base = vbase[rec.w];
+  faces[3 * t + 0] = wvtx[rec.x] - base;
+  faces[3 * t + 1] = wvtx[rec.y] - base;
+  faces[3 * t + 2] = wvtx[rec.z] - base;
 }
 
 __global__ void k_iota(uint32_t *p, uint64_t n) {
@@ -730,7 +783,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
 
   // [3] emit
   if (ensure(c, c->tri_label, T * 4)) return 18;
-  if (ensure(c, c->tri_keys, NC * 4)) return 18;
+  if (ensure(c, c->tri_keys, T * 16)) return 18;
   {
     int blk = 256;
     int waves_per_blk = blk / WAVE;
@@ -741,13 +794,13 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                          (const uint64_t *)c->labels.ptr, g,
                          (const uint32_t *)c->segoff.ptr, lh,
                          (uint32_t *)c->tri_label.ptr,
-                         (uint32_t *)c->tri_keys.ptr);
+                         (uint4 *)c->tri_keys.ptr);
     else
       hipLaunchKernelGGL(k_emit<uint32_t>, dim3((uint32_t)nb), dim3(blk), 0, s,
                          (const uint32_t *)c->labels.ptr, g,
                          (const uint32_t *)c->segoff.ptr, lh,
                          (uint32_t *)c->tri_label.ptr,
-                         (uint32_t *)c->tri_keys.ptr);
+                         (uint4 *)c->tri_keys.ptr);
   }
   HIP_TRY(c, hipGetLastError(), 18);
   HIP_TRY(c, hipEventRecord(c->ev[4], s), 18);
@@ -756,8 +809,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   if (ensure(c, c->order, T * 4)) return 19;
   if (ensure(c, c->order_alt, T * 4)) return 19;
   if (ensure(c, c->tri_label_alt, T * 4)) return 19;
-  if (ensure(c, c->keys_sorted, NC * 4)) return 19;
-  uint32_t *lab_sorted = nullptr;
+  if (ensure(c, c->keys_sorted, T * 16)) return 19;
   {
     int blk = 256;
     uint64_t nb = (T + blk - 1) / blk;
@@ -780,15 +832,14 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
         c->sort_tmp.ptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
     if (e != hipSuccess) { SET_ERR(c, "radix_sort failed"); return 19; }
     // gather
-    hipLaunchKernelGGL(k_gather_keys, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint32_t *)c->tri_keys.ptr, d_vals.current(),
-                       (uint32_t *)c->keys_sorted.ptr, T);
+    hipLaunchKernelGGL(k_gather_recs, dim3((uint32_t)nb), dim3(blk), 0, s,
+                       (const uint4 *)c->tri_keys.ptr, d_vals.current(),
+                       (uint4 *)c->keys_sorted.ptr, T);
     // label ranges
     if (ensure(c, c->tri_off, ((uint64_t)nlabels + 1) * 4)) return 19;
     hipLaunchKernelGGL(k_label_ranges, dim3((uint32_t)nb), dim3(blk), 0, s,
                        d_keys.current(), (uint32_t *)c->tri_off.ptr, T,
                        nlabels);
-    lab_sorted = d_keys.current();
   }
   HIP_TRY(c, hipGetLastError(), 19);
   HIP_TRY(c, hipEventRecord(c->ev[5], s), 19);
@@ -808,14 +859,14 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0, wslots * 4, s), 20);
   uint32_t *wminp = (uint32_t *)c->wh_keys.ptr;
   uint32_t *wvtx = (uint32_t *)c->wh_vtx.ptr;
-  uint32_t *slots_sorted = (uint32_t *)c->keys_sorted.ptr;
+  const uint4 *recs_sorted = (const uint4 *)c->keys_sorted.ptr;
   {
     int blk = 256;
-    uint64_t nb = (NC + blk - 1) / blk;
-    hipLaunchKernelGGL(k_weld_insert, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       slots_sorted, wminp, NC);
+    uint64_t nbt = (T + blk - 1) / blk;
+    hipLaunchKernelGGL(k_weld_insert, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       recs_sorted, wminp, T);
     // scan first-occurrence flags -> vertex ids (flags computed on the fly)
-    FirstOccur fo{slots_sorted, wminp};
+    FirstOccur fo{recs_sorted, wminp};
     auto it = rocprim::make_transform_iterator(
         rocprim::counting_iterator<uint32_t>(0), fo);
     size_t tmp_bytes = 0;
@@ -830,7 +881,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     if (e != hipSuccess) { SET_ERR(c, "weld scan failed"); return 20; }
     hipLaunchKernelGGL(k_total_verts, dim3(1), dim3(1), 0, s,
                        (const uint32_t *)c->vtx_scan.ptr,
-                       slots_sorted, wminp, NC,
+                       recs_sorted, wminp, NC,
                        (uint32_t *)c->lh_misc.ptr + 3);
     uint32_t tv = 0;
     HIP_TRY(c, hipMemcpyAsync(&tv, (uint32_t *)c->lh_misc.ptr + 3, 4,
@@ -845,22 +896,23 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   if (ensure(c, c->vbase, ((uint64_t)nlabels + 1) * 4)) return 22;
   {
     int blk = 256;
-    uint64_t nb = (NC + blk - 1) / blk;
+    uint64_t nbt = (T + blk - 1) / blk;
     const float shift = voxel_centered ? 0.0f : 0.5f;
-    hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       slots_sorted,
+    hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       recs_sorted,
                        (const uint32_t *)c->vtx_scan.ptr, wminp, wvtx,
-                       (float *)c->verts.ptr, g.sx, g.sy, (int64_t)nvox,
-                       rx, ry, rz, shift, NC);
+                       (float *)c->verts.ptr,
+                       (uint32_t)g.sx, (uint32_t)(g.sx * g.sy),
+                       rx, ry, rz, shift, T);
     uint32_t nbl = (nlabels + 1 + 255) / 256;
     hipLaunchKernelGGL(k_vbase, dim3(nbl), dim3(256), 0, s,
                        (const uint32_t *)c->tri_off.ptr,
                        (const uint32_t *)c->vtx_scan.ptr,
                        (uint32_t *)c->vbase.ptr, nlabels, total_verts);
-    hipLaunchKernelGGL(k_faces, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       slots_sorted, lab_sorted, wvtx,
+    hipLaunchKernelGGL(k_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       recs_sorted, wvtx,
                        (const uint32_t *)c->vbase.ptr,
-                       (uint32_t *)c->faces.ptr, NC);
+                       (uint32_t *)c->faces.ptr, T);
   }
   HIP_TRY(c, hipGetLastError(), 22);
   HIP_TRY(c, hipEventRecord(c->ev[6], s), 22);
