@@ -143,7 +143,8 @@ __device__ __forceinline__ void load_corners(const T *__restrict__ labels,
 }
 
 template <typename T>
-__device__ __forceinline__ uint32_t cell_tri_count(const T c[8]) {
+__device__ __forceinline__ uint32_t cell_tri_count(const T c[8],
+                                                   const uint8_t *cnt_tab) {
   if (c[0] == c[1] && c[0] == c[2] && c[0] == c[3] && c[0] == c[4] &&
       c[0] == c[5] && c[0] == c[6] && c[0] == c[7])
     return 0;
@@ -158,7 +159,7 @@ __device__ __forceinline__ uint32_t cell_tri_count(const T c[8]) {
     unsigned mask = 0;
     #pragma unroll
     for (int j = 0; j < 8; ++j) mask |= (c[j] == L) ? (1u << j) : 0u;
-    total += MC_TRI_COUNT[mask];
+    total += cnt_tab[mask];
   }
   return total;
 }
@@ -174,6 +175,10 @@ struct GridDims {
 template <typename T>
 __global__ void k_count(const T *__restrict__ labels, GridDims g,
                         uint32_t *__restrict__ segcnt, LabelHash lh) {
+  __shared__ uint8_t s_cnt[256];
+  for (int k = threadIdx.x; k < 256; k += blockDim.x)
+    s_cnt[k] = MC_TRI_COUNT[k];
+  __syncthreads();
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_in_blk = threadIdx.x / WAVE;
   const int waves_per_blk = blockDim.x / WAVE;
@@ -202,7 +207,7 @@ __global__ void k_count(const T *__restrict__ labels, GridDims g,
           unsigned mask = 0;
           #pragma unroll
           for (int j = 0; j < 8; ++j) mask |= (c[j] == L) ? (1u << j) : 0u;
-          uint32_t nt = MC_TRI_COUNT[mask];
+          uint32_t nt = s_cnt[mask];
           if (nt) {
             cnt += nt;
             label_insert(lh, (uint64_t)L);
@@ -226,6 +231,29 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
                        const uint32_t *__restrict__ segoff, LabelHash lh,
                        uint32_t *__restrict__ tri_label,
                        uint4 *__restrict__ tri_recs) {
+  // all case tables staged in LDS: global-memory byte gathers on these
+  // small tables were the emit kernel's dominant cost (L1 line replays)
+  __shared__ uint8_t s_cnt[256];
+  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
+  __shared__ uint32_t s_comb[12];  // eoff*3 + axis, folded slot math
+  for (int k = threadIdx.x; k < 256; k += blockDim.x) {
+    s_cnt[k] = MC_TRI_COUNT[k];
+    #pragma unroll
+    for (int t = 0; t < MC_MAX_TRIS; ++t)
+      s_pack[k * MC_MAX_TRIS + t] = MC_TRI_PACK[k][t];
+  }
+  if (threadIdx.x < 12) {
+    int e = threadIdx.x;
+    uint32_t eoff = (uint32_t)(MC_EDGE_DOFF[e][0] >> 1) +
+                    (uint32_t)(MC_EDGE_DOFF[e][1] >> 1) * (uint32_t)g.sx +
+                    (uint32_t)(MC_EDGE_DOFF[e][2] >> 1) *
+                        (uint32_t)(g.sx * g.sy);
+    uint32_t axis = (MC_EDGE_DOFF[e][0] & 1)
+                        ? 0u
+                        : ((MC_EDGE_DOFF[e][1] & 1) ? 1u : 2u);
+    s_comb[e] = eoff * 3u + axis;
+  }
+  __syncthreads();
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_in_blk = threadIdx.x / WAVE;
   const int waves_per_blk = blockDim.x / WAVE;
@@ -243,7 +271,7 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
     bool active = false;
     if (cx < g.ncx) {
       load_corners(labels, g.sx, sxy, cx, cy, cz, c);
-      cnt = cell_tri_count(c);
+      cnt = cell_tri_count(c, s_cnt);
       active = cnt > 0;
     }
     uint32_t incl = wave_incl_scan(cnt, lane);
@@ -263,34 +291,21 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
         uint32_t nt = MC_TRI_COUNT[mask];
         if (!nt) continue;
         uint32_t lid = label_lookup(lh, (uint64_t)L);
-        const signed char *tt = MC_TRI_TABLE[mask];
-        // 32-bit voxel linear index of the cell origin (6*nvox < 2^32
-        // is enforced on the host)
-        const uint32_t cell_lin =
-            (uint32_t)((cz * g.sy + cy) * g.sx + cx);
-        const uint32_t usx = (uint32_t)g.sx;
-        const uint32_t usxy = (uint32_t)(g.sx * g.sy);
+        // slot = ((cell_lin + eoff)*3 + axis) << 1 | side
+        //      = ((cell_lin*3 + s_comb[e]) << 1) | side
+        // (voxel-interleaved layout; side = is L the edge's UPPER
+        // endpoint label, baked into MC_TRI_PACK at table-gen time.
+        // Collision-free: a midpoint is a vertex only for its two
+        // endpoint labels. 6*nvox < 2^32 enforced on the host.)
+        const uint32_t cl3 =
+            (uint32_t)((cz * g.sy + cy) * g.sx + cx) * 3u;
         for (uint32_t t = 0; t < nt; ++t) {
+          uint32_t pk = s_pack[mask * MC_MAX_TRIS + t];
           tri_label[pos] = lid;
           uint4 rec;
-          uint32_t *rs = &rec.x;
-          #pragma unroll
-          for (int v = 0; v < 3; ++v) {
-            int e = tt[3 * t + v];
-            // slot = (lower-endpoint voxel, axis, side), voxel-
-            // interleaved so a cell's corners share cache lines.
-            // side = is L the edge's UPPER endpoint label? Collision-
-            // free: a midpoint is a vertex only for its two endpoint
-            // labels.
-            int dx = MC_EDGE_DOFF[e][0], dy = MC_EDGE_DOFF[e][1],
-                dz = MC_EDGE_DOFF[e][2];
-            uint32_t axis = (dx & 1) ? 0u : ((dy & 1) ? 1u : 2u);
-            uint32_t lin = cell_lin + (uint32_t)(dx >> 1) +
-                           (uint32_t)(dy >> 1) * usx +
-                           (uint32_t)(dz >> 1) * usxy;
-            uint32_t side = (c[MC_EDGE_CORNERS[e][1]] == L) ? 1u : 0u;
-            rs[v] = ((lin * 3u + axis) << 1) | side;
-          }
+          rec.x = ((cl3 + s_comb[pk & 15]) << 1) | ((pk >> 4) & 1);
+          rec.y = ((cl3 + s_comb[(pk >> 5) & 15]) << 1) | ((pk >> 9) & 1);
+          rec.z = ((cl3 + s_comb[(pk >> 10) & 15]) << 1) | ((pk >> 14) & 1);
           rec.w = lid;
           tri_recs[pos] = rec;  // one 16-B store per triangle
           ++pos;

@@ -183,6 +183,22 @@ def main():
         for e in range(12):
             f.write("  {%d, %d, %d},\n" % EDGE_DOFF[e])
         f.write("};\n\n")
+        f.write("// packed per-(mask,tri) corner descriptors for the HIP emit\n")
+        f.write("// kernel: 5 bits per corner = edge id (4) | side (1), where\n")
+        f.write("// side = mask bit of the edge's UPPER corner (is the label\n")
+        f.write("// the edge's upper endpoint?). 3 corners -> 15 bits, u16.\n")
+        f.write(f"MC_TABLE_QUAL unsigned short MC_TRI_PACK[256][{maxt}] = {{\n")
+        for m in range(256):
+            packs = []
+            for t in table[m]:
+                p = 0
+                for v, e in enumerate(t):
+                    side = (m >> EDGES[e][1]) & 1
+                    p |= (e | (side << 4)) << (5 * v)
+                packs.append(p)
+            packs += [0] * (maxt - len(packs))
+            f.write("  {" + ", ".join(f"0x{v:04x}" for v in packs) + "},\n")
+        f.write("};\n\n")
         f.write("// corner ids (a,b) of each edge, a < b (a = lower voxel)\n")
         f.write("MC_TABLE_QUAL unsigned char MC_EDGE_CORNERS[12][2] = {\n")
         for (a, b) in EDGES:
