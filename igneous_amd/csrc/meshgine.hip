@@ -493,6 +493,20 @@ __global__ void k_iota(uint32_t *p, uint64_t n) {
   if (i < n) p[i] = (uint32_t)i;
 }
 
+// out = exclusive_scan[n-1] + input[n-1]  (total of a scanned array)
+__global__ void k_last_sum(const uint32_t *scan, const uint32_t *input,
+                           uint64_t n, uint32_t *out) {
+  *out = scan[n - 1] + input[n - 1];
+}
+
+__global__ void k_any_active(const uint8_t *active, uint32_t nlabels,
+                             uint32_t *any) {
+  uint32_t l = blockIdx.x * blockDim.x + threadIdx.x;
+  if (l < nlabels && active[l]) atomicExch(any, 1u);
+}
+
+#include "simplify.hip"
+
 // ---------------------------------------------------------------------------
 // context / host side
 
@@ -513,7 +527,11 @@ struct mg_ctx {
       tri_label, tri_label_alt, order, order_alt, tri_keys, keys_sorted,
       tri_off, sort_tmp,
       wh_keys, wh_minp, wh_vtx, flags, vtx_scan, verts, faces, vbase,
-      label_values, small;
+      label_values, small,
+      simp_fq, simp_valid, simp_pk, simp_pk_alt, simp_pv, simp_pv_alt,
+      simp_Q, simp_pick, simp_remap, simp_flab, simp_flab_alt,
+      simp_faces_alt, simp_verts_alt, simp_vbase_alt, simp_meta, simp_ref,
+      simp_keep, simp_keep_scan;
   uint64_t lh_slots = 1ull << 20;
   hipEvent_t ev[16] = {};
 };
@@ -594,7 +612,13 @@ void mg_destroy(mg_ctx *c) {
                   &c->keys_sorted, &c->tri_off, &c->sort_tmp, &c->wh_keys,
                   &c->wh_minp, &c->wh_vtx, &c->flags, &c->vtx_scan,
                   &c->verts, &c->faces, &c->vbase, &c->label_values,
-                  &c->small}) {
+                  &c->small,
+                  &c->simp_fq, &c->simp_valid, &c->simp_pk, &c->simp_pk_alt,
+                  &c->simp_pv, &c->simp_pv_alt, &c->simp_Q, &c->simp_pick,
+                  &c->simp_remap, &c->simp_flab, &c->simp_flab_alt,
+                  &c->simp_faces_alt, &c->simp_verts_alt, &c->simp_vbase_alt,
+                  &c->simp_meta, &c->simp_ref, &c->simp_keep,
+                  &c->simp_keep_scan}) {
     if (b->ptr) (void)hipFree(b->ptr);
   }
   for (auto &e : c->ev) if (e) (void)hipEventDestroy(e);
@@ -665,6 +689,240 @@ static double ev_ms(mg_ctx *c, int a, int b) {
   float ms = 0.f;
   if (hipEventElapsedTime(&ms, c->ev[a], c->ev[b]) != hipSuccess) return 0.0;
   return (double)ms;
+}
+
+
+// ---------------------------------------------------------------------------
+// GPU quadric simplification driver (kernels in simplify.hip); mirrors
+// oracle/simplify.c round for round. Updates faces/verts/vbase/tri_off
+// in the ctx; p_T/p_V become the post-simplification totals.
+static int run_simplify(mg_ctx *c, uint32_t nlabels,
+                        uint32_t reduction_factor, float max_error,
+                        uint64_t *p_T, uint64_t *p_V) {
+  hipStream_t s = c->stream;
+  uint64_t T = *p_T;
+  const uint64_t V = *p_V;
+  const float max_cost = max_error * max_error;
+  const int blk = 256;
+  if (T == 0 || V == 0) return 0;
+
+  // meta layout: [0,L) nt_cur | [L,2L) target | [2L,3L) nt_new |
+  //              3L..: active u8[L] | any u32 (aligned)
+  const uint64_t L = nlabels;
+  const uint64_t meta_bytes = L * 12 + ((L + 3) & ~3ull) + 4;
+  if (ensure(c, c->simp_meta, meta_bytes)) return 40;
+  uint32_t *nt_cur = (uint32_t *)c->simp_meta.ptr;
+  uint32_t *target = nt_cur + L;
+  uint32_t *nt_new = target + L;
+  uint8_t *active = (uint8_t *)(nt_new + L);
+  uint32_t *d_any = (uint32_t *)((char *)c->simp_meta.ptr +
+                                 L * 12 + ((L + 3) & ~3ull));
+
+  if (ensure(c, c->simp_flab, T * 4)) return 40;
+  if (ensure(c, c->simp_flab_alt, T * 4)) return 40;
+  if (ensure(c, c->simp_faces_alt, T * 12)) return 40;
+  if (ensure(c, c->simp_fq, T * sizeof(SimpPlane))) return 40;
+  if (ensure(c, c->simp_valid, T)) return 40;
+  if (ensure(c, c->simp_pk, 3 * T * 4)) return 40;
+  if (ensure(c, c->simp_pk_alt, 3 * T * 4)) return 40;
+  if (ensure(c, c->simp_pv, 3 * T * 4)) return 40;
+  if (ensure(c, c->simp_pv_alt, 3 * T * 4)) return 40;
+  if (ensure(c, c->simp_Q, V * 40)) return 40;
+  if (ensure(c, c->simp_pick, V * 8)) return 40;
+  if (ensure(c, c->simp_remap, V * 4)) return 40;
+  if (ensure(c, c->simp_keep, T * 4)) return 40;
+  if (ensure(c, c->simp_keep_scan, T * 4)) return 40;
+
+  uint32_t *faces_g = (uint32_t *)c->faces.ptr;
+  float *verts = (float *)c->verts.ptr;
+
+  // faces -> global vertex ids, label per face
+  {
+    uint64_t nbt = (T + blk - 1) / blk;
+    hipLaunchKernelGGL(k_globalize_faces, dim3((uint32_t)nbt), dim3(blk), 0,
+                       s, faces_g, (const uint4 *)c->keys_sorted.ptr,
+                       (const uint32_t *)c->vbase.ptr,
+                       (uint32_t *)c->simp_flab.ptr, T);
+    uint32_t nbl = (uint32_t)((L + 255) / 256);
+    hipLaunchKernelGGL(k_init_simplify, dim3(nbl), dim3(256), 0, s,
+                       (const uint32_t *)c->tri_off.ptr, nt_cur, target,
+                       active, reduction_factor, (uint32_t)L);
+  }
+  HIP_TRY(c, hipGetLastError(), 40);
+
+  for (int round = 0; round < 256; ++round) {
+    // any label still active?
+    HIP_TRY(c, hipMemsetAsync(d_any, 0, 4, s), 41);
+    {
+      uint32_t nbl = (uint32_t)((L + 255) / 256);
+      hipLaunchKernelGGL(k_any_active, dim3(nbl), dim3(256), 0, s,
+                         active, (uint32_t)L, d_any);
+    }
+    uint32_t any = 0;
+    HIP_TRY(c, hipMemcpyAsync(&any, d_any, 4, hipMemcpyDeviceToHost, s), 41);
+    HIP_TRY(c, hipStreamSynchronize(s), 41);
+    if (!any) break;
+
+    uint64_t nbt = (T + blk - 1) / blk;
+    uint64_t NP = 3 * T;
+    uint64_t nbp = (NP + blk - 1) / blk;
+    uint32_t *flab = (uint32_t *)c->simp_flab.ptr;
+
+    hipLaunchKernelGGL(k_face_planes, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       faces_g, verts, active, flab,
+                       (SimpPlane *)c->simp_fq.ptr,
+                       (uint8_t *)c->simp_valid.ptr, T);
+    hipLaunchKernelGGL(k_emit_vf_pairs, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       faces_g, active, flab,
+                       (uint32_t *)c->simp_pk.ptr,
+                       (uint32_t *)c->simp_pv.ptr, T);
+    // stable sort pairs by vertex (face order preserved within vertex)
+    {
+      rocprim::double_buffer<uint32_t> dk((uint32_t *)c->simp_pk.ptr,
+                                          (uint32_t *)c->simp_pk_alt.ptr);
+      rocprim::double_buffer<uint32_t> dv((uint32_t *)c->simp_pv.ptr,
+                                          (uint32_t *)c->simp_pv_alt.ptr);
+      size_t tmp = 0;
+      hipError_t e = rocprim::radix_sort_pairs(nullptr, tmp, dk, dv, NP,
+                                               0u, 32u, s);
+      if (e != hipSuccess) { SET_ERR(c, "simplify sort size query"); return 42; }
+      if (ensure(c, c->sort_tmp, tmp)) return 42;
+      e = rocprim::radix_sort_pairs(c->sort_tmp.ptr, tmp, dk, dv, NP,
+                                    0u, 32u, s);
+      if (e != hipSuccess) { SET_ERR(c, "simplify sort"); return 42; }
+      hipLaunchKernelGGL(k_accum_quadrics, dim3((uint32_t)nbp), dim3(blk), 0,
+                         s, dk.current(), dv.current(),
+                         (const SimpPlane *)c->simp_fq.ptr,
+                         (const uint8_t *)c->simp_valid.ptr,
+                         (float *)c->simp_Q.ptr, NP);
+    }
+    HIP_TRY(c, hipMemsetAsync(c->simp_pick.ptr, 0xFF, V * 8, s), 43);
+    hipLaunchKernelGGL(k_edge_pick, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       faces_g, active, flab, verts,
+                       (const float *)c->simp_Q.ptr,
+                       (unsigned long long *)c->simp_pick.ptr, max_cost, T);
+    {
+      uint64_t nbv = (V + blk - 1) / blk;
+      hipLaunchKernelGGL(k_iota, dim3((uint32_t)nbv), dim3(blk), 0, s,
+                         (uint32_t *)c->simp_remap.ptr, V);
+      hipLaunchKernelGGL(k_collapse, dim3((uint32_t)nbv), dim3(blk), 0, s,
+                         (const unsigned long long *)c->simp_pick.ptr, verts,
+                         (uint32_t *)c->simp_remap.ptr, V);
+    }
+    hipLaunchKernelGGL(k_remap_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       faces_g, (const uint32_t *)c->simp_remap.ptr,
+                       (uint32_t *)c->simp_keep.ptr, T);
+    {
+      size_t tmp = 0;
+      hipError_t e = rocprim::exclusive_scan(
+          nullptr, tmp, (uint32_t *)c->simp_keep.ptr,
+          (uint32_t *)c->simp_keep_scan.ptr, 0u, T,
+          rocprim::plus<uint32_t>(), s);
+      if (e != hipSuccess) { SET_ERR(c, "keep scan size query"); return 44; }
+      if (ensure(c, c->scan_tmp, tmp)) return 44;
+      e = rocprim::exclusive_scan(
+          c->scan_tmp.ptr, tmp, (uint32_t *)c->simp_keep.ptr,
+          (uint32_t *)c->simp_keep_scan.ptr, 0u, T,
+          rocprim::plus<uint32_t>(), s);
+      if (e != hipSuccess) { SET_ERR(c, "keep scan"); return 44; }
+    }
+    hipLaunchKernelGGL(k_last_sum, dim3(1), dim3(1), 0, s,
+                       (const uint32_t *)c->simp_keep_scan.ptr,
+                       (const uint32_t *)c->simp_keep.ptr, T,
+                       (uint32_t *)c->lh_misc.ptr + 4);
+    uint32_t kept = 0;
+    HIP_TRY(c, hipMemcpyAsync(&kept, (uint32_t *)c->lh_misc.ptr + 4, 4,
+                              hipMemcpyDeviceToHost, s), 44);
+    HIP_TRY(c, hipMemsetAsync(nt_new, 0, L * 4, s), 44);
+    hipLaunchKernelGGL(k_compact_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       faces_g, flab, (const uint32_t *)c->simp_keep.ptr,
+                       (const uint32_t *)c->simp_keep_scan.ptr,
+                       (uint32_t *)c->simp_faces_alt.ptr,
+                       (uint32_t *)c->simp_flab_alt.ptr, nt_new, T);
+    {
+      uint32_t nbl = (uint32_t)((L + 255) / 256);
+      hipLaunchKernelGGL(k_update_active, dim3(nbl), dim3(256), 0, s,
+                         nt_new, nt_cur, target, active, d_any, (uint32_t)L);
+    }
+    HIP_TRY(c, hipGetLastError(), 44);
+    HIP_TRY(c, hipStreamSynchronize(s), 44);
+    std::swap(c->faces, c->simp_faces_alt);
+    std::swap(c->simp_flab, c->simp_flab_alt);
+    faces_g = (uint32_t *)c->faces.ptr;
+    T = kept;
+    if (T == 0) break;
+  }
+
+  // final: drop unreferenced vertices (stable), re-localize faces
+  if (ensure(c, c->simp_ref, (V + 1) * 4)) return 45;
+  if (ensure(c, c->simp_verts_alt, V * 12 + 12)) return 45;
+  if (ensure(c, c->simp_vbase_alt, (L + 1) * 4)) return 45;
+  uint32_t *ref = (uint32_t *)c->simp_ref.ptr;
+  uint32_t *newid = (uint32_t *)c->vtx_scan.ptr;  // NC*4 >= V*4, free now
+  HIP_TRY(c, hipMemsetAsync(ref, 0, V * 4, s), 45);
+  uint64_t NCk = 3 * T;
+  if (T > 0) {
+    uint64_t nbc = (NCk + blk - 1) / blk;
+    hipLaunchKernelGGL(k_mark_ref, dim3((uint32_t)nbc), dim3(blk), 0, s,
+                       faces_g, ref, NCk);
+  }
+  {
+    size_t tmp = 0;
+    hipError_t e = rocprim::exclusive_scan(
+        nullptr, tmp, ref, newid, 0u, V, rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "ref scan size query"); return 45; }
+    if (ensure(c, c->scan_tmp, tmp)) return 45;
+    e = rocprim::exclusive_scan(
+        c->scan_tmp.ptr, tmp, ref, newid, 0u, V,
+        rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "ref scan"); return 45; }
+  }
+  hipLaunchKernelGGL(k_last_sum, dim3(1), dim3(1), 0, s, newid, ref, V,
+                     (uint32_t *)c->lh_misc.ptr + 5);
+  uint32_t newV = 0;
+  HIP_TRY(c, hipMemcpyAsync(&newV, (uint32_t *)c->lh_misc.ptr + 5, 4,
+                            hipMemcpyDeviceToHost, s), 45);
+  HIP_TRY(c, hipStreamSynchronize(s), 45);
+  {
+    uint64_t nbv = (V + blk - 1) / blk;
+    hipLaunchKernelGGL(k_scatter_verts, dim3((uint32_t)nbv), dim3(blk), 0, s,
+                       verts, ref, newid, (float *)c->simp_verts_alt.ptr, V);
+    uint32_t nbl = (uint32_t)((L + 2 + 255) / 256);
+    hipLaunchKernelGGL(k_new_vbase, dim3(nbl), dim3(256), 0, s,
+                       (const uint32_t *)c->vbase.ptr, newid,
+                       (uint32_t *)c->simp_vbase_alt.ptr, (uint32_t)L, newV);
+    if (T > 0) {
+      uint64_t nbt = (T + blk - 1) / blk;
+      hipLaunchKernelGGL(k_localize_faces, dim3((uint32_t)nbt), dim3(blk), 0,
+                         s, faces_g, newid,
+                         (const uint32_t *)c->simp_flab.ptr,
+                         (const uint32_t *)c->simp_vbase_alt.ptr,
+                         (uint32_t *)c->simp_faces_alt.ptr, T);
+    }
+  }
+  // tri_off = exclusive scan of per-label counts, [L] = T
+  {
+    size_t tmp = 0;
+    hipError_t e = rocprim::exclusive_scan(
+        nullptr, tmp, nt_cur, (uint32_t *)c->tri_off.ptr, 0u, L,
+        rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "ntoff scan size query"); return 46; }
+    if (ensure(c, c->scan_tmp, tmp)) return 46;
+    e = rocprim::exclusive_scan(
+        c->scan_tmp.ptr, tmp, nt_cur, (uint32_t *)c->tri_off.ptr, 0u, L,
+        rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "ntoff scan"); return 46; }
+    uint32_t Tu = (uint32_t)T;
+    HIP_TRY(c, hipMemcpyAsync((uint32_t *)c->tri_off.ptr + L, &Tu, 4,
+                              hipMemcpyHostToDevice, s), 46);
+  }
+  HIP_TRY(c, hipGetLastError(), 46);
+  std::swap(c->faces, c->simp_faces_alt);
+  std::swap(c->verts, c->simp_verts_alt);
+  std::swap(c->vbase, c->simp_vbase_alt);
+  *p_T = T;
+  *p_V = newV;
+  return 0;
 }
 
 static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
@@ -776,12 +1034,6 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   if (nlabels >= (1u << 27)) {  // label id shares the 64-bit weld key
     SET_ERR(c, "%u labels exceed the 2^27 per-chunk label limit", nlabels);
     return 15;
-  }
-  if (reduction_factor > 1) {
-    SET_ERR(c, "GPU simplifier not yet wired (reduction_factor=%u); "
-            "round-1 engine supports reduction_factor<=1",
-            reduction_factor);
-    return 16;
   }
 
   const uint64_t T = total_tris;
@@ -932,6 +1184,16 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   HIP_TRY(c, hipGetLastError(), 22);
   HIP_TRY(c, hipEventRecord(c->ev[6], s), 22);
 
+  // [6] per-label quadric simplification (mesh.py:376-381 semantics)
+  uint64_t Tcur = T, Vcur = total_verts;
+  if (reduction_factor > 1 && T > 0) {
+    int rc = run_simplify(c, nlabels, reduction_factor, max_error,
+                          &Tcur, &Vcur);
+    if (rc) return rc;
+  }
+  HIP_TRY(c, hipEventRecord(c->ev[8], s), 22);
+  const uint64_t NCX = 3 * Tcur;
+
   // [7] extract
   mg_meshset *ms = nullptr;
   if (flags_ & MG_FLAG_DEVICE_ONLY) {
@@ -941,18 +1203,20 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   } else {
     float *h_verts = nullptr;
     uint32_t *h_faces = nullptr;
-    HIP_TRY(c, hipHostMalloc((void **)&h_verts, total_verts * 12), 24);
-    if (hipHostMalloc((void **)&h_faces, NC * 4) != hipSuccess) {
+    HIP_TRY(c, hipHostMalloc((void **)&h_verts, Vcur * 12 + 12), 24);
+    if (hipHostMalloc((void **)&h_faces, NCX * 4 + 4) != hipSuccess) {
       (void)hipHostFree(h_verts);
       SET_ERR(c, "hipHostMalloc faces failed");
       return 24;
     }
     std::vector<uint32_t> h_tri_off(nlabels + 1), h_vbase(nlabels + 1);
     std::vector<uint64_t> h_label_values(nlabels);
-    HIP_TRY(c, hipMemcpyAsync(h_verts, c->verts.ptr, total_verts * 12,
-                              hipMemcpyDeviceToHost, s), 24);
-    HIP_TRY(c, hipMemcpyAsync(h_faces, c->faces.ptr, NC * 4,
-                              hipMemcpyDeviceToHost, s), 24);
+    if (Vcur > 0)
+      HIP_TRY(c, hipMemcpyAsync(h_verts, c->verts.ptr, Vcur * 12,
+                                hipMemcpyDeviceToHost, s), 24);
+    if (NCX > 0)
+      HIP_TRY(c, hipMemcpyAsync(h_faces, c->faces.ptr, NCX * 4,
+                                hipMemcpyDeviceToHost, s), 24);
     HIP_TRY(c, hipMemcpyAsync(h_tri_off.data(), c->tri_off.ptr,
                               (nlabels + 1) * 4, hipMemcpyDeviceToHost, s), 24);
     HIP_TRY(c, hipMemcpyAsync(h_vbase.data(), c->vbase.ptr, (nlabels + 1) * 4,
@@ -994,8 +1258,8 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   c->stats.ms_emit = ev_ms(c, 3, 4);
   c->stats.ms_partition = ev_ms(c, 4, 5);
   c->stats.ms_weld = ev_ms(c, 5, 6);
-  c->stats.ms_simplify = 0.0;
-  c->stats.ms_d2h = ev_ms(c, 6, 7);
+  c->stats.ms_simplify = ev_ms(c, 6, 8);
+  c->stats.ms_d2h = ev_ms(c, 8, 7);
   c->stats.ms_total = ev_ms(c, 0, 7);
   return 0;
 }

@@ -1,0 +1,286 @@
+// simplify.hip — per-label quadric edge-collapse simplifier on the GPU.
+//
+// Restates zmesh's Mesher.get(id, reduction_factor, max_error,
+// voxel_centered) simplification step
+// (/root/reference/igneous/tasks/mesh/mesh.py:376-381) with the SAME
+// deterministic matched-pair independent-set schedule as the CPU oracle
+// (oracle/simplify.c) — bit-exact: identical f32 expressions (plane from
+// cross product + correctly-rounded sqrtf/div, unit-weight quadrics summed
+// in ascending face order, midpoint placement, cost = (Qu+Qw)(m) <=
+// max_error^2, per-vertex cheapest edge with smaller-peer tie-break,
+// matched pairs collapse, rounds until a label hits its triangle target
+// or stops shrinking).
+//
+// All labels simplify SIMULTANEOUSLY: faces stay label-partitioned, the
+// vertex space is label-disjoint (vbase slices), so per-label rounds are
+// independent; a finished label freezes while others continue.
+//
+// Included by meshgine.hip (single TU).
+
+struct SimpPlane {
+  float nx, ny, nz, d;
+};
+
+// quadric helpers — textually identical arithmetic to oracle/simplify.c
+__device__ __forceinline__ void sq_add_plane(float *q, float a, float b,
+                                             float c_, float d, float w) {
+  q[0] += w * a * a; q[1] += w * a * b; q[2] += w * a * c_; q[3] += w * a * d;
+  q[4] += w * b * b; q[5] += w * b * c_; q[6] += w * b * d;
+  q[7] += w * c_ * c_; q[8] += w * c_ * d;
+  q[9] += w * d * d;
+}
+
+__device__ __forceinline__ float sq_eval(const float *q, float x, float y,
+                                         float z) {
+  return q[0]*x*x + 2.0f*q[1]*x*y + 2.0f*q[2]*x*z + 2.0f*q[3]*x
+       + q[4]*y*y + 2.0f*q[5]*y*z + 2.0f*q[6]*y
+       + q[7]*z*z + 2.0f*q[8]*z
+       + q[9];
+}
+
+// [S1] per-face plane (recomputed each round; verts move)
+__global__ void k_face_planes(const uint32_t *__restrict__ faces_g,
+                              const float *__restrict__ verts,
+                              const uint8_t *__restrict__ active_lab,
+                              const uint32_t *__restrict__ flab,
+                              SimpPlane *__restrict__ fq,
+                              uint8_t *__restrict__ fvalid,
+                              uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  if (!active_lab[flab[t]]) { fvalid[t] = 0; return; }
+  uint32_t i0 = faces_g[3*t], i1 = faces_g[3*t+1], i2 = faces_g[3*t+2];
+  const float *p0 = verts + 3*i0, *p1 = verts + 3*i1, *p2 = verts + 3*i2;
+  float ux = p1[0]-p0[0], uy = p1[1]-p0[1], uz = p1[2]-p0[2];
+  float vx = p2[0]-p0[0], vy = p2[1]-p0[1], vz = p2[2]-p0[2];
+  float nx = uy*vz - uz*vy, ny = uz*vx - ux*vz, nz = ux*vy - uy*vx;
+  float len = sqrtf(nx*nx + ny*ny + nz*nz);
+  if (len <= 0.0f) { fvalid[t] = 0; return; }
+  float inv = 1.0f / len;
+  nx *= inv; ny *= inv; nz *= inv;
+  float d = -(nx*p0[0] + ny*p0[1] + nz*p0[2]);
+  fq[t] = SimpPlane{nx, ny, nz, d};
+  fvalid[t] = 1;
+}
+
+// [S2] emit (vertex, face) pairs for quadric accumulation; inactive
+// labels get the sentinel key (sorted to the end, skipped)
+__global__ void k_emit_vf_pairs(const uint32_t *__restrict__ faces_g,
+                                const uint8_t *__restrict__ active_lab,
+                                const uint32_t *__restrict__ flab,
+                                uint32_t *__restrict__ pk,
+                                uint32_t *__restrict__ pv,
+                                uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  bool act = active_lab[flab[t]];
+  #pragma unroll
+  for (int v = 0; v < 3; ++v) {
+    pk[3*t + v] = act ? faces_g[3*t + v] : 0xFFFFFFFFu;
+    pv[3*t + v] = (uint32_t)t;
+  }
+}
+
+// [S3] per-vertex quadric: walk the vertex's pair segment in ascending
+// face order (stable sort preserves it) — the oracle's summation order
+__global__ void k_accum_quadrics(const uint32_t *__restrict__ pk,
+                                 const uint32_t *__restrict__ pv,
+                                 const SimpPlane *__restrict__ fq,
+                                 const uint8_t *__restrict__ fvalid,
+                                 float *__restrict__ Q,
+                                 uint64_t npairs) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= npairs) return;
+  uint32_t v = pk[i];
+  if (v == 0xFFFFFFFFu) return;
+  if (i > 0 && pk[i-1] == v) return;  // not a segment head
+  float q[10];
+  #pragma unroll
+  for (int k = 0; k < 10; ++k) q[k] = 0.0f;
+  for (uint64_t j = i; j < npairs && pk[j] == v; ++j) {
+    uint32_t f = pv[j];
+    if (!fvalid[f]) continue;
+    SimpPlane p = fq[f];
+    sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
+  }
+  #pragma unroll
+  for (int k = 0; k < 10; ++k) Q[10ull*v + k] = q[k];
+}
+
+// [S4] per-vertex cheapest incident edge (cost-bits<<32 | peer, min)
+__global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
+                            const uint8_t *__restrict__ active_lab,
+                            const uint32_t *__restrict__ flab,
+                            const float *__restrict__ verts,
+                            const float *__restrict__ Q,
+                            unsigned long long *__restrict__ pick,
+                            float max_cost, uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  if (!active_lab[flab[t]]) return;
+  #pragma unroll
+  for (int e = 0; e < 3; ++e) {
+    uint32_t a = faces_g[3*t + e], b = faces_g[3*t + (e+1)%3];
+    if (a == b) continue;
+    uint32_t u = a < b ? a : b, w = a < b ? b : a;
+    float mx = 0.5f*(verts[3*u]+verts[3*w]);
+    float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+    float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+    float S[10];
+    #pragma unroll
+    for (int k = 0; k < 10; ++k) S[k] = Q[10ull*u + k] + Q[10ull*w + k];
+    float cost = sq_eval(S, mx, my, mz);
+    if (cost < 0.0f) cost = 0.0f;
+    if (cost > max_cost) continue;
+    uint32_t cb = __float_as_uint(cost);
+    atomicMin(&pick[u], ((unsigned long long)cb << 32) | w);
+    atomicMin(&pick[w], ((unsigned long long)cb << 32) | u);
+  }
+}
+
+// [S5] matched pairs collapse to the midpoint; u (smaller id) survives
+__global__ void k_collapse(const unsigned long long *__restrict__ pick,
+                           float *__restrict__ verts,
+                           uint32_t *__restrict__ remap,
+                           uint64_t nverts) {
+  uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (u >= nverts) return;
+  unsigned long long pu = pick[u];
+  if (pu == ~0ull) return;
+  uint32_t w = (uint32_t)pu;
+  if (w <= u) return;
+  unsigned long long pw = pick[w];
+  if (pw == ~0ull || (uint32_t)pw != u) return;
+  verts[3*u]   = 0.5f*(verts[3*u]+verts[3*w]);
+  verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+  verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+  remap[w] = (uint32_t)u;
+}
+
+// [S6] remap face corners in place; keep flag for non-degenerates
+__global__ void k_remap_faces(uint32_t *__restrict__ faces_g,
+                              const uint32_t *__restrict__ remap,
+                              uint32_t *__restrict__ keep,
+                              uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  uint32_t i0 = remap[faces_g[3*t]], i1 = remap[faces_g[3*t+1]],
+           i2 = remap[faces_g[3*t+2]];
+  faces_g[3*t] = i0; faces_g[3*t+1] = i1; faces_g[3*t+2] = i2;
+  keep[t] = (i0 != i1 && i1 != i2 && i0 != i2) ? 1u : 0u;
+}
+
+// [S7] stable compaction of kept faces (+ labels); per-label new counts
+__global__ void k_compact_faces(const uint32_t *__restrict__ faces_g,
+                                const uint32_t *__restrict__ flab,
+                                const uint32_t *__restrict__ keep,
+                                const uint32_t *__restrict__ keep_scan,
+                                uint32_t *__restrict__ faces_out,
+                                uint32_t *__restrict__ flab_out,
+                                uint32_t *__restrict__ nt_new,
+                                uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  if (!keep[t]) return;
+  uint64_t o = keep_scan[t];
+  faces_out[3*o] = faces_g[3*t];
+  faces_out[3*o+1] = faces_g[3*t+1];
+  faces_out[3*o+2] = faces_g[3*t+2];
+  uint32_t lab = flab[t];
+  flab_out[o] = lab;
+  atomicAdd(&nt_new[lab], 1u);
+}
+
+// [S8] per-label round bookkeeping
+__global__ void k_update_active(const uint32_t *__restrict__ nt_new,
+                                uint32_t *__restrict__ nt_cur,
+                                const uint32_t *__restrict__ target,
+                                uint8_t *__restrict__ active_lab,
+                                uint32_t *__restrict__ any_active,
+                                uint32_t nlabels) {
+  uint32_t l = blockIdx.x * blockDim.x + threadIdx.x;
+  if (l >= nlabels) return;
+  if (!active_lab[l]) return;
+  uint32_t nn = nt_new[l];
+  bool progress = nn != nt_cur[l];
+  nt_cur[l] = nn;
+  bool act = progress && nn > target[l];
+  active_lab[l] = act ? 1 : 0;
+  if (act) atomicExch(any_active, 1u);
+}
+
+__global__ void k_init_simplify(const uint32_t *__restrict__ tri_off,
+                                uint32_t *__restrict__ nt_cur,
+                                uint32_t *__restrict__ target,
+                                uint8_t *__restrict__ active_lab,
+                                uint32_t reduction_factor,
+                                uint32_t nlabels) {
+  uint32_t l = blockIdx.x * blockDim.x + threadIdx.x;
+  if (l >= nlabels) return;
+  uint32_t nt = tri_off[l+1] - tri_off[l];
+  uint32_t tg = nt / reduction_factor;
+  if (tg < 1) tg = 1;
+  nt_cur[l] = nt;
+  target[l] = tg;
+  active_lab[l] = (nt > tg) ? 1 : 0;
+}
+
+// faces local -> global vertex ids (vbase offset per label)
+__global__ void k_globalize_faces(uint32_t *__restrict__ faces,
+                                  const uint4 *__restrict__ recs_sorted,
+                                  const uint32_t *__restrict__ vbase,
+                                  uint32_t *__restrict__ flab,
+                                  uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  uint32_t lab = recs_sorted[t].w;
+  flab[t] = lab;
+  uint32_t base = vbase[lab];
+  faces[3*t] += base; faces[3*t+1] += base; faces[3*t+2] += base;
+}
+
+// ---- final referenced-vertex compaction ------------------------------
+
+__global__ void k_mark_ref(const uint32_t *__restrict__ faces_g,
+                           uint32_t *__restrict__ ref, uint64_t ncorners) {
+  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= ncorners) return;
+  ref[faces_g[i]] = 1u;  // idempotent store, order-free
+}
+
+__global__ void k_scatter_verts(const float *__restrict__ verts,
+                                const uint32_t *__restrict__ ref,
+                                const uint32_t *__restrict__ newid,
+                                float *__restrict__ verts_out,
+                                uint64_t nverts) {
+  uint64_t v = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (v >= nverts) return;
+  if (!ref[v]) return;
+  uint64_t o = newid[v];
+  verts_out[3*o] = verts[3*v];
+  verts_out[3*o+1] = verts[3*v+1];
+  verts_out[3*o+2] = verts[3*v+2];
+}
+
+__global__ void k_new_vbase(const uint32_t *__restrict__ vbase_old,
+                            const uint32_t *__restrict__ newid,
+                            uint32_t *__restrict__ vbase_new,
+                            uint32_t nlabels, uint32_t total_new) {
+  uint32_t l = blockIdx.x * blockDim.x + threadIdx.x;
+  if (l > nlabels) return;
+  vbase_new[l] = (l == nlabels) ? total_new : newid[vbase_old[l]];
+}
+
+__global__ void k_localize_faces(const uint32_t *__restrict__ faces_g,
+                                 const uint32_t *__restrict__ newid,
+                                 const uint32_t *__restrict__ flab,
+                                 const uint32_t *__restrict__ vbase_new,
+                                 uint32_t *__restrict__ faces_out,
+                                 uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  uint32_t base = vbase_new[flab[t]];
+  faces_out[3*t] = newid[faces_g[3*t]] - base;
+  faces_out[3*t+1] = newid[faces_g[3*t+1]] - base;
+  faces_out[3*t+2] = newid[faces_g[3*t+2]] - base;
+}
