@@ -36,6 +36,7 @@
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
+#include <cstdlib>
 #include <mutex>
 #include <string>
 #include <vector>
@@ -750,7 +751,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   }
   HIP_TRY(c, hipGetLastError(), 40);
 
-  for (int round = 0; round < 256; ++round) {
+  for (int round = 0; round < 65536; ++round) {
     // any label still active?
     HIP_TRY(c, hipMemsetAsync(d_any, 0, 4, s), 41);
     {
@@ -801,6 +802,22 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                        faces_g, active, flab, verts,
                        (const float *)c->simp_Q.ptr,
                        (unsigned long long *)c->simp_pick.ptr, max_cost, T);
+    if (round == 0 && getenv("MG_DEBUG_SIMPLIFY")) {
+      float hq[40]; unsigned long long hp[8]; uint32_t hf[12];
+      (void)hipMemcpyAsync(hq, c->simp_Q.ptr, sizeof(hq), hipMemcpyDeviceToHost, s);
+      (void)hipMemcpyAsync(hp, c->simp_pick.ptr, sizeof(hp), hipMemcpyDeviceToHost, s);
+      (void)hipMemcpyAsync(hf, faces_g, sizeof(hf), hipMemcpyDeviceToHost, s);
+      (void)hipStreamSynchronize(s);
+      fprintf(stderr, "[mg] faces0-3:");
+      for (int k = 0; k < 12; ++k) fprintf(stderr, " %u", hf[k]);
+      fprintf(stderr, "\n[mg] Q0: ");
+      for (int k = 0; k < 10; ++k) fprintf(stderr, "%a ", hq[k]);
+      fprintf(stderr, "\n[mg] Q1: ");
+      for (int k = 10; k < 20; ++k) fprintf(stderr, "%a ", hq[k]);
+      fprintf(stderr, "\n[mg] pick0-7:");
+      for (int k = 0; k < 8; ++k) fprintf(stderr, " %llx", hp[k]);
+      fprintf(stderr, "\n");
+    }
     {
       uint64_t nbv = (V + blk - 1) / blk;
       hipLaunchKernelGGL(k_iota, dim3((uint32_t)nbv), dim3(blk), 0, s,

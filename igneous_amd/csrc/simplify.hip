@@ -133,6 +133,10 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
     if (cost < 0.0f) cost = 0.0f;
     if (cost > max_cost) continue;
     uint32_t cb = __float_as_uint(cost);
+    // per-edge tie jitter — identical to oracle/simplify.c
+    uint32_t hsh = u ^ (w * 2654435761u);
+    hsh ^= hsh >> 16; hsh *= 2246822519u; hsh ^= hsh >> 13;
+    cb ^= (hsh & 7u);
     atomicMin(&pick[u], ((unsigned long long)cb << 32) | w);
     atomicMin(&pick[w], ((unsigned long long)cb << 32) | u);
   }

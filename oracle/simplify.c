@@ -27,7 +27,9 @@
 #include <stdint.h>
 #include <stdlib.h>
 #include <string.h>
+#include <stdlib.h>
 #include <math.h>
+#include <stdio.h>
 
 typedef struct { float q[10]; } quad10; /* symmetric 4x4: a2,ab,ac,ad,b2,bc,bd,c2,cd,d2 */
 
@@ -98,6 +100,14 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
         if (cost < 0.0f) cost = 0.0f;
         if (cost > max_cost) continue;
         uint32_t cb; memcpy(&cb, &cost, 4);
+        /* deterministic per-edge jitter on the 3 low cost bits: breaks
+         * the equal-cost pick chains of flat regions (which would give
+         * O(1/sqrt(n)) matches per round) into random-preference
+         * matchings (~1/degree of vertices collapse per round). Part of
+         * the canonical contract; the HIP kernel mirrors it exactly. */
+        uint32_t hsh = u ^ (w * 2654435761u);
+        hsh ^= hsh >> 16; hsh *= 2246822519u; hsh ^= hsh >> 13;
+        cb ^= (hsh & 7u);
         uint64_t enc_u = ((uint64_t)cb << 32) | w;
         uint64_t enc_w = ((uint64_t)cb << 32) | u;
         if (enc_u < pick[u]) pick[u] = enc_u;
@@ -129,6 +139,9 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
       out++;
     }
     if (out < nt) progress = 1;
+    if (getenv("OMC_DEBUG"))
+      fprintf(stderr, "[omc] round: collapses=%u nt %u -> %u\n",
+              collapses, nt, out);
     nt = out;
   }
 
