@@ -799,7 +799,8 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     }
     HIP_TRY(c, hipMemsetAsync(c->simp_pick.ptr, 0xFF, V * 8, s), 43);
     hipLaunchKernelGGL(k_edge_pick, dim3((uint32_t)nbt), dim3(blk), 0, s,
-                       faces_g, active, flab, verts,
+                       faces_g, active, flab,
+                       (const uint32_t *)c->vbase.ptr, verts,
                        (const float *)c->simp_Q.ptr,
                        (unsigned long long *)c->simp_pick.ptr, max_cost, T);
     if (round == 0 && getenv("MG_DEBUG_SIMPLIFY")) {

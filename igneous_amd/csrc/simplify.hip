@@ -111,13 +111,16 @@ __global__ void k_accum_quadrics(const uint32_t *__restrict__ pk,
 __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
                             const uint8_t *__restrict__ active_lab,
                             const uint32_t *__restrict__ flab,
+                            const uint32_t *__restrict__ vbase,
                             const float *__restrict__ verts,
                             const float *__restrict__ Q,
                             unsigned long long *__restrict__ pick,
                             float max_cost, uint64_t ntris) {
   uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= ntris) return;
-  if (!active_lab[flab[t]]) return;
+  uint32_t lab = flab[t];
+  if (!active_lab[lab]) return;
+  const uint32_t vb = vbase[lab];
   #pragma unroll
   for (int e = 0; e < 3; ++e) {
     uint32_t a = faces_g[3*t + e], b = faces_g[3*t + (e+1)%3];
@@ -133,8 +136,10 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
     if (cost < 0.0f) cost = 0.0f;
     if (cost > max_cost) continue;
     uint32_t cb = __float_as_uint(cost);
-    // per-edge tie jitter — identical to oracle/simplify.c
-    uint32_t hsh = u ^ (w * 2654435761u);
+    // per-edge tie jitter — identical to oracle/simplify.c, which works
+    // in label-LOCAL vertex ids: subtract the label's vertex base
+    uint32_t ul = u - vb, wl = w - vb;
+    uint32_t hsh = ul ^ (wl * 2654435761u);
     hsh ^= hsh >> 16; hsh *= 2246822519u; hsh ^= hsh >> 13;
     cb ^= (hsh & 7u);
     atomicMin(&pick[u], ((unsigned long long)cb << 32) | w);
