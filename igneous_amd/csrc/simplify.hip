@@ -7,9 +7,10 @@
 // (oracle/simplify.c) — bit-exact: identical f32 expressions (plane from
 // cross product + correctly-rounded sqrtf/div, unit-weight quadrics summed
 // in ascending face order, midpoint placement, cost = (Qu+Qw)(m) <=
-// max_error^2, per-vertex cheapest edge with smaller-peer tie-break,
-// matched pairs collapse, rounds until a label hits its triangle target
-// or stops shrinking).
+// max_error^2, per-vertex cheapest edge with deterministic per-edge tie
+// jitter on the 3 low cost bits + smaller-peer tie-break, matched pairs
+// collapse, rounds until a label hits its triangle target or stops
+// shrinking).
 //
 // All labels simplify SIMULTANEOUSLY: faces stay label-partitioned, the
 // vertex space is label-disjoint (vbase slices), so per-label rounds are
