@@ -532,7 +532,7 @@ struct mg_ctx {
       simp_fq, simp_valid, simp_pk, simp_pk_alt, simp_pv, simp_pv_alt,
       simp_Q, simp_pick, simp_remap, simp_flab, simp_flab_alt,
       simp_faces_alt, simp_verts_alt, simp_vbase_alt, simp_meta, simp_ref,
-      simp_keep, simp_keep_scan;
+      simp_keep, simp_keep_scan, simp_park, simp_first, simp_troff_final;
   uint64_t lh_slots = 1ull << 20;
   hipEvent_t ev[16] = {};
 };
@@ -619,7 +619,8 @@ void mg_destroy(mg_ctx *c) {
                   &c->simp_remap, &c->simp_flab, &c->simp_flab_alt,
                   &c->simp_faces_alt, &c->simp_verts_alt, &c->simp_vbase_alt,
                   &c->simp_meta, &c->simp_ref, &c->simp_keep,
-                  &c->simp_keep_scan}) {
+                  &c->simp_keep_scan, &c->simp_park, &c->simp_first,
+                  &c->simp_troff_final}) {
     if (b->ptr) (void)hipFree(b->ptr);
   }
   for (auto &e : c->ev) if (e) (void)hipEventDestroy(e);
@@ -733,6 +734,9 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   if (ensure(c, c->simp_remap, V * 4)) return 40;
   if (ensure(c, c->simp_keep, T * 4)) return 40;
   if (ensure(c, c->simp_keep_scan, T * 4)) return 40;
+  if (ensure(c, c->simp_park, 3 * T * 4)) return 40;
+  if (ensure(c, c->simp_first, (L + 1) * 4)) return 40;
+  if (ensure(c, c->simp_troff_final, (L + 1) * 4)) return 40;
 
   uint32_t *faces_g = (uint32_t *)c->faces.ptr;
   float *verts = (float *)c->verts.ptr;
@@ -869,6 +873,118 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     faces_g = (uint32_t *)c->faces.ptr;
     T = kept;
     if (T == 0) break;
+
+    // park faces of labels that just went inactive: working set keeps
+    // only active labels (late rounds touch only the active tail)
+    {
+      uint64_t nbt2 = (T + blk - 1) / blk;
+      uint32_t *flab2 = (uint32_t *)c->simp_flab.ptr;
+      hipLaunchKernelGGL(k_flag_active_faces, dim3((uint32_t)nbt2),
+                         dim3(blk), 0, s, flab2, active,
+                         (uint32_t *)c->simp_keep.ptr, T);
+      size_t tmp = 0;
+      hipError_t e = rocprim::exclusive_scan(
+          nullptr, tmp, (uint32_t *)c->simp_keep.ptr,
+          (uint32_t *)c->simp_keep_scan.ptr, 0u, T,
+          rocprim::plus<uint32_t>(), s);
+      if (e != hipSuccess) { SET_ERR(c, "park scan size query"); return 47; }
+      if (ensure(c, c->scan_tmp, tmp)) return 47;
+      e = rocprim::exclusive_scan(
+          c->scan_tmp.ptr, tmp, (uint32_t *)c->simp_keep.ptr,
+          (uint32_t *)c->simp_keep_scan.ptr, 0u, T,
+          rocprim::plus<uint32_t>(), s);
+      if (e != hipSuccess) { SET_ERR(c, "park scan"); return 47; }
+      hipLaunchKernelGGL(k_first_label_idx, dim3((uint32_t)nbt2), dim3(blk),
+                         0, s, flab2, (uint32_t *)c->simp_first.ptr, T);
+      hipLaunchKernelGGL(k_park_scatter, dim3((uint32_t)nbt2), dim3(blk), 0,
+                         s, faces_g, flab2,
+                         (const uint32_t *)c->simp_keep.ptr,
+                         (const uint32_t *)c->simp_keep_scan.ptr,
+                         (const uint32_t *)c->simp_first.ptr,
+                         (const uint32_t *)c->tri_off.ptr,
+                         (uint32_t *)c->simp_faces_alt.ptr,
+                         (uint32_t *)c->simp_flab_alt.ptr,
+                         (uint32_t *)c->simp_park.ptr, T);
+      hipLaunchKernelGGL(k_last_sum, dim3(1), dim3(1), 0, s,
+                         (const uint32_t *)c->simp_keep_scan.ptr,
+                         (const uint32_t *)c->simp_keep.ptr, T,
+                         (uint32_t *)c->lh_misc.ptr + 4);
+      uint32_t act_total = 0;
+      HIP_TRY(c, hipMemcpyAsync(&act_total, (uint32_t *)c->lh_misc.ptr + 4,
+                                4, hipMemcpyDeviceToHost, s), 47);
+      HIP_TRY(c, hipStreamSynchronize(s), 47);
+      std::swap(c->faces, c->simp_faces_alt);
+      std::swap(c->simp_flab, c->simp_flab_alt);
+      faces_g = (uint32_t *)c->faces.ptr;
+      T = act_total;
+      if (T == 0) break;
+    }
+  }
+
+  // any faces still in the working set (round-cap exit): force-park them
+  if (T > 0) {
+    HIP_TRY(c, hipMemsetAsync(active, 0, L, s), 48);
+    uint64_t nbt2 = (T + blk - 1) / blk;
+    uint32_t *flab2 = (uint32_t *)c->simp_flab.ptr;
+    hipLaunchKernelGGL(k_flag_active_faces, dim3((uint32_t)nbt2), dim3(blk),
+                       0, s, flab2, active, (uint32_t *)c->simp_keep.ptr, T);
+    size_t tmp = 0;
+    rocprim::exclusive_scan(nullptr, tmp, (uint32_t *)c->simp_keep.ptr,
+                            (uint32_t *)c->simp_keep_scan.ptr, 0u, T,
+                            rocprim::plus<uint32_t>(), s);
+    if (ensure(c, c->scan_tmp, tmp)) return 48;
+    rocprim::exclusive_scan(c->scan_tmp.ptr, tmp,
+                            (uint32_t *)c->simp_keep.ptr,
+                            (uint32_t *)c->simp_keep_scan.ptr, 0u, T,
+                            rocprim::plus<uint32_t>(), s);
+    hipLaunchKernelGGL(k_first_label_idx, dim3((uint32_t)nbt2), dim3(blk),
+                       0, s, flab2, (uint32_t *)c->simp_first.ptr, T);
+    hipLaunchKernelGGL(k_park_scatter, dim3((uint32_t)nbt2), dim3(blk), 0,
+                       s, faces_g, flab2,
+                       (const uint32_t *)c->simp_keep.ptr,
+                       (const uint32_t *)c->simp_keep_scan.ptr,
+                       (const uint32_t *)c->simp_first.ptr,
+                       (const uint32_t *)c->tri_off.ptr,
+                       (uint32_t *)c->simp_faces_alt.ptr,
+                       (uint32_t *)c->simp_flab_alt.ptr,
+                       (uint32_t *)c->simp_park.ptr, T);
+    HIP_TRY(c, hipGetLastError(), 48);
+  }
+
+  // final face array: per-label slices from the park store, lid order
+  {
+    size_t tmp = 0;
+    hipError_t e = rocprim::exclusive_scan(
+        nullptr, tmp, nt_cur, (uint32_t *)c->simp_troff_final.ptr, 0u, L,
+        rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "final troff scan size"); return 49; }
+    if (ensure(c, c->scan_tmp, tmp)) return 49;
+    e = rocprim::exclusive_scan(
+        c->scan_tmp.ptr, tmp, nt_cur, (uint32_t *)c->simp_troff_final.ptr,
+        0u, L, rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "final troff scan"); return 49; }
+    hipLaunchKernelGGL(k_last_sum, dim3(1), dim3(1), 0, s,
+                       (const uint32_t *)c->simp_troff_final.ptr, nt_cur, L,
+                       (uint32_t *)c->lh_misc.ptr + 6);
+    HIP_TRY(c, hipMemcpyAsync((uint32_t *)c->simp_troff_final.ptr + L,
+                              (uint32_t *)c->lh_misc.ptr + 6, 4,
+                              hipMemcpyDeviceToDevice, s), 49);
+    uint32_t final_total = 0;
+    HIP_TRY(c, hipMemcpyAsync(&final_total, (uint32_t *)c->lh_misc.ptr + 6,
+                              4, hipMemcpyDeviceToHost, s), 49);
+    HIP_TRY(c, hipStreamSynchronize(s), 49);
+    T = final_total;
+    if (T > 0) {
+      uint64_t nbt2 = (T + blk - 1) / blk;
+      hipLaunchKernelGGL(k_gather_final, dim3((uint32_t)nbt2), dim3(blk), 0,
+                         s, (const uint32_t *)c->simp_park.ptr,
+                         (const uint32_t *)c->tri_off.ptr,
+                         (const uint32_t *)c->simp_troff_final.ptr,
+                         (uint32_t *)c->faces.ptr,
+                         (uint32_t *)c->simp_flab.ptr, (uint32_t)L, T);
+    }
+    faces_g = (uint32_t *)c->faces.ptr;
+    std::swap(c->tri_off, c->simp_troff_final);
   }
 
   // final: drop unreferenced vertices (stable), re-localize faces
@@ -917,22 +1033,6 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                          (const uint32_t *)c->simp_vbase_alt.ptr,
                          (uint32_t *)c->simp_faces_alt.ptr, T);
     }
-  }
-  // tri_off = exclusive scan of per-label counts, [L] = T
-  {
-    size_t tmp = 0;
-    hipError_t e = rocprim::exclusive_scan(
-        nullptr, tmp, nt_cur, (uint32_t *)c->tri_off.ptr, 0u, L,
-        rocprim::plus<uint32_t>(), s);
-    if (e != hipSuccess) { SET_ERR(c, "ntoff scan size query"); return 46; }
-    if (ensure(c, c->scan_tmp, tmp)) return 46;
-    e = rocprim::exclusive_scan(
-        c->scan_tmp.ptr, tmp, nt_cur, (uint32_t *)c->tri_off.ptr, 0u, L,
-        rocprim::plus<uint32_t>(), s);
-    if (e != hipSuccess) { SET_ERR(c, "ntoff scan"); return 46; }
-    uint32_t Tu = (uint32_t)T;
-    HIP_TRY(c, hipMemcpyAsync((uint32_t *)c->tri_off.ptr + L, &Tu, 4,
-                              hipMemcpyHostToDevice, s), 46);
   }
   HIP_TRY(c, hipGetLastError(), 46);
   std::swap(c->faces, c->simp_faces_alt);

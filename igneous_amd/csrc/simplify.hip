@@ -294,3 +294,76 @@ __global__ void k_localize_faces(const uint32_t *__restrict__ faces_g,
   faces_out[3*t+1] = newid[faces_g[3*t+1]] - base;
   faces_out[3*t+2] = newid[faces_g[3*t+2]] - base;
 }
+
+// ---- working-set parking: a label that goes inactive leaves the round
+// loop entirely; its faces are stored verbatim (in face order) at the
+// label's ORIGINAL tri_off offset in the park store, and the working
+// array keeps only active labels' faces. Late rounds then touch only the
+// shrinking active tail instead of all 155M faces.
+
+__global__ void k_flag_active_faces(const uint32_t *__restrict__ flab,
+                                    const uint8_t *__restrict__ active_lab,
+                                    uint32_t *__restrict__ aflag,
+                                    uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  aflag[t] = active_lab[flab[t]] ? 1u : 0u;
+}
+
+__global__ void k_first_label_idx(const uint32_t *__restrict__ flab,
+                                  uint32_t *__restrict__ first,
+                                  uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  if (t == 0 || flab[t] != flab[t - 1]) first[flab[t]] = (uint32_t)t;
+}
+
+__global__ void k_park_scatter(const uint32_t *__restrict__ faces_g,
+                               const uint32_t *__restrict__ flab,
+                               const uint32_t *__restrict__ aflag,
+                               const uint32_t *__restrict__ apos,
+                               const uint32_t *__restrict__ first,
+                               const uint32_t *__restrict__ orig_tri_off,
+                               uint32_t *__restrict__ work_faces,
+                               uint32_t *__restrict__ work_flab,
+                               uint32_t *__restrict__ park_faces,
+                               uint64_t ntris) {
+  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ntris) return;
+  uint32_t lab = flab[t];
+  if (aflag[t]) {
+    uint64_t o = apos[t];
+    work_faces[3*o] = faces_g[3*t];
+    work_faces[3*o+1] = faces_g[3*t+1];
+    work_faces[3*o+2] = faces_g[3*t+2];
+    work_flab[o] = lab;
+  } else {
+    uint64_t o = (uint64_t)orig_tri_off[lab] + ((uint32_t)t - first[lab]);
+    park_faces[3*o] = faces_g[3*t];
+    park_faces[3*o+1] = faces_g[3*t+1];
+    park_faces[3*o+2] = faces_g[3*t+2];
+  }
+}
+
+// assemble the final compact face array from the park store
+__global__ void k_gather_final(const uint32_t *__restrict__ park_faces,
+                               const uint32_t *__restrict__ orig_tri_off,
+                               const uint32_t *__restrict__ final_tri_off,
+                               uint32_t *__restrict__ faces_out,
+                               uint32_t *__restrict__ flab_out,
+                               uint32_t nlabels, uint64_t final_total) {
+  uint64_t f = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (f >= final_total) return;
+  // binary search: label whose [final_tri_off[l], final_tri_off[l+1]) holds f
+  uint32_t lo = 0, hi = nlabels;
+  while (hi - lo > 1) {
+    uint32_t mid = (lo + hi) >> 1;
+    if ((uint64_t)final_tri_off[mid] <= f) lo = mid; else hi = mid;
+  }
+  uint32_t l = lo;
+  uint64_t src = (uint64_t)orig_tri_off[l] + (f - final_tri_off[l]);
+  faces_out[3*f] = park_faces[3*src];
+  faces_out[3*f+1] = park_faces[3*src+1];
+  faces_out[3*f+2] = park_faces[3*src+2];
+  flab_out[f] = l;
+}
