@@ -42,52 +42,59 @@ RESOLUTION = (16.0, 16.0, 40.0)
 PEAK_HBM = 8.0e12  # B/s, MI355X spec (MI355X_MICROARCH.md)
 
 
-def _oracle_baseline_worker(args):
-    sub, res = args
+_BASE_DATA = None   # fork-shared (copy-on-write) chunk for baseline workers
+_BASE_RED = 0
+
+
+def _oracle_baseline_worker(idx):
+    """Mesh one 64^3 (+1 overlap) subchunk of the fork-shared chunk —
+    the reference's process-per-chunk --parallel worker model."""
     import oracle  # oracle/ is on sys.path (checker/baseline leg only)
-    t0 = time.perf_counter()
-    oracle.mesh_chunk(sub, resolution=res)
-    return time.perf_counter() - t0, sub.size
+    n = SHAPE[0] // 64
+    z, rem = divmod(idx, n * n)
+    y, x = divmod(rem, n)
+    sub = np.asfortranarray(
+        _BASE_DATA[x * 64:x * 64 + 65,
+                   y * 64:y * 64 + 65,
+                   z * 64:z * 64 + 65])
+    oracle.mesh_chunk(sub, resolution=RESOLUTION,
+                      reduction_factor=_BASE_RED, max_error=40.0)
+    return sub.size
 
 
-def cpu_baseline(data: np.ndarray, budget_s: float = 20.0) -> dict:
+def cpu_baseline(data: np.ndarray, reduction: int = 0,
+                 budget_s: float = 25.0) -> dict:
     """Time the CPU oracle on a bounded sample of the same chunk:
-    process-per-subchunk over all host cores (the reference's --parallel
-    worker model), ~budget_s of wall time."""
+    process-per-subchunk over all host cores, ~budget_s of wall time.
+    Data is fork-shared so workers pay no serialization."""
+    global _BASE_DATA, _BASE_RED
     sys.path.insert(0, os.path.join(REPO, "oracle"))
     import oracle
     oracle.build()
+    _BASE_DATA = data
+    _BASE_RED = reduction
     cores = os.cpu_count() or 1
-    # sample: 128^3 (+1 overlap) subchunks of the real chunk
-    subs = []
-    n = SHAPE[0] // 128
-    for z in range(n):
-        for y in range(n):
-            for x in range(n):
-                sub = np.asfortranarray(
-                    data[x * 128:x * 128 + 129,
-                         y * 128:y * 128 + 129,
-                         z * 128:z * 128 + 129])
-                subs.append(sub)
+    n = SHAPE[0] // 64
+    total_chunks = n ** 3  # 512 tasks of 65^3
     # calibrate with one subchunk, single process
-    t_one, vox_one = _oracle_baseline_worker((subs[0], RESOLUTION))
-    per_chunk = t_one
-    target_chunks = max(cores, min(len(subs),
-                                   int(budget_s / per_chunk * cores)))
-    chosen = subs[:target_chunks]
+    t0 = time.perf_counter()
+    _oracle_baseline_worker(0)
+    per_chunk = time.perf_counter() - t0
+    target = max(cores, min(total_chunks,
+                            int(budget_s / per_chunk * cores)))
+    chosen = list(range(target))
     t0 = time.perf_counter()
     with mp.get_context("fork").Pool(cores) as pool:
-        pool.map(_oracle_baseline_worker,
-                 [(s, RESOLUTION) for s in chosen])
+        vox = sum(pool.map(_oracle_baseline_worker, chosen, chunksize=1))
     elapsed = time.perf_counter() - t0
-    vox = sum(s.size for s in chosen)
     return {
         "value": round(vox / elapsed / 1e6, 2),
         "unit": "Mvox/s",
         "cores": cores,
         "kind": "port",
-        "sample": (f"{len(chosen)} x 129^3 subchunks of the same 512^3 "
-                   f"chunk, {cores}-process pool, {elapsed:.1f}s"),
+        "sample": (f"{len(chosen)} x 65^3 subchunks of the same 512^3 "
+                   f"chunk (reduction_factor={reduction}), "
+                   f"{cores}-process pool, {elapsed:.1f}s"),
     }
 
 
@@ -187,7 +194,7 @@ def main():
 
     cpu = None
     if not args.no_cpu_baseline and rank == 0 and world == 1:
-        cpu = cpu_baseline(data)
+        cpu = cpu_baseline(data, reduction=red)
 
     line = {
         "metric": "Mvoxels/s meshed (512^3 uint64 seg chunk)",
