@@ -806,7 +806,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       uint64_t nbv = (V + blk - 1) / blk;
       hipLaunchKernelGGL(k_iota, dim3((uint32_t)nbv), dim3(blk), 0, s,
                          (uint32_t *)c->simp_remap.ptr, V);
-      for (int sub = 0; sub < 4; ++sub) {
+      for (int sub = 0; sub < 2; ++sub) {
         HIP_TRY(c, hipMemsetAsync(c->simp_pick.ptr, 0xFF, V * 8, s), 43);
         HIP_TRY(c, hipMemsetAsync((uint32_t *)c->lh_misc.ptr + 7, 0, 4, s),
                 43);

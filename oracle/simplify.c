@@ -81,7 +81,7 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
       quad_add_plane(&Q[i1], nx, ny, nz, d, 1.0f);
       quad_add_plane(&Q[i2], nx, ny, nz, d, 1.0f);
     }
-    /* 2+3. FOUR matching sub-rounds per quadric recompute: each
+    /* 2+3. TWO matching sub-rounds per quadric recompute: each
      * sub-round picks every live vertex's cheapest incident edge
      * (corners resolved through the cumulative remap), collapses the
      * mutual pairs to their midpoints, and the next sub-round re-picks
@@ -92,7 +92,7 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
      * the canonical contract; the HIP kernels mirror it exactly. */
     for (uint32_t v = 0; v < nv; v++) remap[v] = v;
     uint32_t collapses = 0;
-    for (int sub = 0; sub < 4; sub++) {
+    for (int sub = 0; sub < 2; sub++) {
       for (uint32_t v = 0; v < nv; v++) pick[v] = UINT64_MAX;
       for (uint32_t t = 0; t < nt; t++) {
         for (int e = 0; e < 3; e++) {
