@@ -801,33 +801,12 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                          (const uint8_t *)c->simp_valid.ptr,
                          (float *)c->simp_Q.ptr, NP);
     }
-    // 4 matching sub-rounds on the round's quadrics (remap cumulative)
-    {
-      uint64_t nbv = (V + blk - 1) / blk;
-      hipLaunchKernelGGL(k_iota, dim3((uint32_t)nbv), dim3(blk), 0, s,
-                         (uint32_t *)c->simp_remap.ptr, V);
-      for (int sub = 0; sub < 2; ++sub) {
-        HIP_TRY(c, hipMemsetAsync(c->simp_pick.ptr, 0xFF, V * 8, s), 43);
-        HIP_TRY(c, hipMemsetAsync((uint32_t *)c->lh_misc.ptr + 7, 0, 4, s),
-                43);
-        hipLaunchKernelGGL(k_edge_pick, dim3((uint32_t)nbt), dim3(blk), 0, s,
-                           faces_g, active, flab,
-                           (const uint32_t *)c->vbase.ptr,
-                           (const uint32_t *)c->simp_remap.ptr, verts,
-                           (const float *)c->simp_Q.ptr,
-                           (unsigned long long *)c->simp_pick.ptr, max_cost,
-                           T);
-        hipLaunchKernelGGL(k_collapse, dim3((uint32_t)nbv), dim3(blk), 0, s,
-                           (const unsigned long long *)c->simp_pick.ptr,
-                           verts, (uint32_t *)c->simp_remap.ptr,
-                           (uint32_t *)c->lh_misc.ptr + 7, V);
-        uint32_t ncol = 0;
-        HIP_TRY(c, hipMemcpyAsync(&ncol, (uint32_t *)c->lh_misc.ptr + 7, 4,
-                                  hipMemcpyDeviceToHost, s), 43);
-        HIP_TRY(c, hipStreamSynchronize(s), 43);
-        if (!ncol) break;
-      }
-    }
+    HIP_TRY(c, hipMemsetAsync(c->simp_pick.ptr, 0xFF, V * 8, s), 43);
+    hipLaunchKernelGGL(k_edge_pick, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       faces_g, active, flab,
+                       (const uint32_t *)c->vbase.ptr, verts,
+                       (const float *)c->simp_Q.ptr,
+                       (unsigned long long *)c->simp_pick.ptr, max_cost, T);
     if (round == 0 && getenv("MG_DEBUG_SIMPLIFY")) {
       float hq[40]; unsigned long long hp[8]; uint32_t hf[12];
       (void)hipMemcpyAsync(hq, c->simp_Q.ptr, sizeof(hq), hipMemcpyDeviceToHost, s);
@@ -843,6 +822,14 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       fprintf(stderr, "\n[mg] pick0-7:");
       for (int k = 0; k < 8; ++k) fprintf(stderr, " %llx", hp[k]);
       fprintf(stderr, "\n");
+    }
+    {
+      uint64_t nbv = (V + blk - 1) / blk;
+      hipLaunchKernelGGL(k_iota, dim3((uint32_t)nbv), dim3(blk), 0, s,
+                         (uint32_t *)c->simp_remap.ptr, V);
+      hipLaunchKernelGGL(k_collapse, dim3((uint32_t)nbv), dim3(blk), 0, s,
+                         (const unsigned long long *)c->simp_pick.ptr, verts,
+                         (uint32_t *)c->simp_remap.ptr, V);
     }
     hipLaunchKernelGGL(k_remap_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
                        faces_g, (const uint32_t *)c->simp_remap.ptr,

@@ -108,20 +108,11 @@ __global__ void k_accum_quadrics(const uint32_t *__restrict__ pk,
   for (int k = 0; k < 10; ++k) Q[10ull*v + k] = q[k];
 }
 
-__device__ __forceinline__ uint32_t simp_resolve(
-    const uint32_t *__restrict__ remap, uint32_t x) {
-  // cumulative-collapse chains are at most sub-round deep (<= 4)
-  while (remap[x] != x) x = remap[x];
-  return x;
-}
-
-// [S4] per-vertex cheapest incident edge (cost-bits<<32 | peer, min);
-// corners resolved through the cumulative intra-round remap (sub-rounds)
+// [S4] per-vertex cheapest incident edge (cost-bits<<32 | peer, min)
 __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
                             const uint8_t *__restrict__ active_lab,
                             const uint32_t *__restrict__ flab,
                             const uint32_t *__restrict__ vbase,
-                            const uint32_t *__restrict__ remap,
                             const float *__restrict__ verts,
                             const float *__restrict__ Q,
                             unsigned long long *__restrict__ pick,
@@ -131,12 +122,9 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
   uint32_t lab = flab[t];
   if (!active_lab[lab]) return;
   const uint32_t vb = vbase[lab];
-  uint32_t fc[3] = {simp_resolve(remap, faces_g[3*t]),
-                    simp_resolve(remap, faces_g[3*t+1]),
-                    simp_resolve(remap, faces_g[3*t+2])};
   #pragma unroll
   for (int e = 0; e < 3; ++e) {
-    uint32_t a = fc[e], b = fc[(e+1)%3];
+    uint32_t a = faces_g[3*t + e], b = faces_g[3*t + (e+1)%3];
     if (a == b) continue;
     uint32_t u = a < b ? a : b, w = a < b ? b : a;
     float mx = 0.5f*(verts[3*u]+verts[3*w]);
@@ -160,14 +148,10 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
   }
 }
 
-// [S5] matched pairs collapse to the midpoint; u (smaller id) survives.
-// Picks reference only live (resolved) vertices, so matched pairs are
-// disjoint and the parallel writes race-free; counts collapses so the
-// host can stop sub-rounds early.
+// [S5] matched pairs collapse to the midpoint; u (smaller id) survives
 __global__ void k_collapse(const unsigned long long *__restrict__ pick,
                            float *__restrict__ verts,
                            uint32_t *__restrict__ remap,
-                           uint32_t *__restrict__ n_collapsed,
                            uint64_t nverts) {
   uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (u >= nverts) return;
@@ -181,19 +165,17 @@ __global__ void k_collapse(const unsigned long long *__restrict__ pick,
   verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
   verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
   remap[w] = (uint32_t)u;
-  atomicAdd(n_collapsed, 1u);
 }
 
-// [S6] remap face corners in place (full chain resolve); keep flag
+// [S6] remap face corners in place; keep flag for non-degenerates
 __global__ void k_remap_faces(uint32_t *__restrict__ faces_g,
                               const uint32_t *__restrict__ remap,
                               uint32_t *__restrict__ keep,
                               uint64_t ntris) {
   uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= ntris) return;
-  uint32_t i0 = simp_resolve(remap, faces_g[3*t]),
-           i1 = simp_resolve(remap, faces_g[3*t+1]),
-           i2 = simp_resolve(remap, faces_g[3*t+2]);
+  uint32_t i0 = remap[faces_g[3*t]], i1 = remap[faces_g[3*t+1]],
+           i2 = remap[faces_g[3*t+2]];
   faces_g[3*t] = i0; faces_g[3*t+1] = i1; faces_g[3*t+2] = i2;
   keep[t] = (i0 != i1 && i1 != i2 && i0 != i2) ? 1u : 0u;
 }

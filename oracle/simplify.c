@@ -81,67 +81,59 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
       quad_add_plane(&Q[i1], nx, ny, nz, d, 1.0f);
       quad_add_plane(&Q[i2], nx, ny, nz, d, 1.0f);
     }
-    /* 2+3. TWO matching sub-rounds per quadric recompute: each
-     * sub-round picks every live vertex's cheapest incident edge
-     * (corners resolved through the cumulative remap), collapses the
-     * mutual pairs to their midpoints, and the next sub-round re-picks
-     * among survivors with the SAME (round-start) quadrics. Cost as raw
-     * f32 bits (>= 0, so bit order == value order); deterministic
-     * per-edge jitter on the 3 low cost bits breaks the equal-cost pick
-     * chains of flat regions into random-preference matchings. Part of
-     * the canonical contract; the HIP kernels mirror it exactly. */
+    /* 2. per-vertex best incident edge: encode (costbits<<32 | peer) and
+     * take min. Cost as raw f32 bits (all costs >= 0 so bit order == value
+     * order); tie-break by smaller peer index. Deterministic: min over
+     * edges is order-independent. */
+    for (uint32_t v = 0; v < nv; v++) pick[v] = UINT64_MAX;
+    for (uint32_t t = 0; t < nt; t++) {
+      for (int e = 0; e < 3; e++) {
+        uint32_t a = faces[3*t + e], b = faces[3*t + (e+1)%3];
+        if (a == b) continue;
+        uint32_t u = a < b ? a : b, w = a < b ? b : a;
+        float mx = 0.5f*(verts[3*u]+verts[3*w]);
+        float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+        float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+        quad10 S;
+        for (int k = 0; k < 10; k++) S.q[k] = Q[u].q[k] + Q[w].q[k];
+        float cost = quad_eval(&S, mx, my, mz);
+        if (cost < 0.0f) cost = 0.0f;
+        if (cost > max_cost) continue;
+        uint32_t cb; memcpy(&cb, &cost, 4);
+        /* deterministic per-edge jitter on the 3 low cost bits: breaks
+         * the equal-cost pick chains of flat regions (which would give
+         * O(1/sqrt(n)) matches per round) into random-preference
+         * matchings (~1/degree of vertices collapse per round). Part of
+         * the canonical contract; the HIP kernel mirrors it exactly. */
+        uint32_t hsh = u ^ (w * 2654435761u);
+        hsh ^= hsh >> 16; hsh *= 2246822519u; hsh ^= hsh >> 13;
+        cb ^= (hsh & 7u);
+        uint64_t enc_u = ((uint64_t)cb << 32) | w;
+        uint64_t enc_w = ((uint64_t)cb << 32) | u;
+        if (enc_u < pick[u]) pick[u] = enc_u;
+        if (enc_w < pick[w]) pick[w] = enc_w;
+      }
+    }
+    /* 3. matched pairs collapse to midpoint (u<v keeps u) */
     for (uint32_t v = 0; v < nv; v++) remap[v] = v;
     uint32_t collapses = 0;
-    for (int sub = 0; sub < 2; sub++) {
-      for (uint32_t v = 0; v < nv; v++) pick[v] = UINT64_MAX;
-      for (uint32_t t = 0; t < nt; t++) {
-        for (int e = 0; e < 3; e++) {
-          uint32_t a = faces[3*t + e], b = faces[3*t + (e+1)%3];
-          while (remap[a] != a) a = remap[a];
-          while (remap[b] != b) b = remap[b];
-          if (a == b) continue;
-          uint32_t u = a < b ? a : b, w = a < b ? b : a;
-          float mx = 0.5f*(verts[3*u]+verts[3*w]);
-          float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
-          float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
-          quad10 S;
-          for (int k = 0; k < 10; k++) S.q[k] = Q[u].q[k] + Q[w].q[k];
-          float cost = quad_eval(&S, mx, my, mz);
-          if (cost < 0.0f) cost = 0.0f;
-          if (cost > max_cost) continue;
-          uint32_t cb; memcpy(&cb, &cost, 4);
-          uint32_t hsh = u ^ (w * 2654435761u);
-          hsh ^= hsh >> 16; hsh *= 2246822519u; hsh ^= hsh >> 13;
-          cb ^= (hsh & 7u);
-          uint64_t enc_u = ((uint64_t)cb << 32) | w;
-          uint64_t enc_w = ((uint64_t)cb << 32) | u;
-          if (enc_u < pick[u]) pick[u] = enc_u;
-          if (enc_w < pick[w]) pick[w] = enc_w;
-        }
-      }
-      uint32_t sub_collapses = 0;
-      for (uint32_t u = 0; u < nv; u++) {
-        if (pick[u] == UINT64_MAX) continue;
-        uint32_t w = (uint32_t)pick[u];
-        if (w <= u) continue;             /* handle each pair from its min end */
-        if (pick[w] == UINT64_MAX || (uint32_t)pick[w] != u) continue;
-        verts[3*u]   = 0.5f*(verts[3*u]+verts[3*w]);
-        verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
-        verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
-        remap[w] = u;
-        sub_collapses++;
-      }
-      collapses += sub_collapses;
-      if (!sub_collapses) break;
+    for (uint32_t u = 0; u < nv; u++) {
+      if (pick[u] == UINT64_MAX) continue;
+      uint32_t w = (uint32_t)pick[u];
+      if (w <= u) continue;               /* handle each pair from its min end */
+      if (pick[w] == UINT64_MAX || (uint32_t)pick[w] != u) continue; /* not matched */
+      if (remap[u] != u || remap[w] != w) continue; /* already touched */
+      verts[3*u]   = 0.5f*(verts[3*u]+verts[3*w]);
+      verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+      verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+      remap[w] = u;
+      collapses++;
     }
     if (!collapses) break;
-    /* 4. rewrite faces through the cumulative remap, drop degenerates */
+    /* 4. rewrite faces, drop degenerates */
     uint32_t out = 0;
     for (uint32_t t = 0; t < nt; t++) {
-      uint32_t i0 = faces[3*t], i1 = faces[3*t+1], i2 = faces[3*t+2];
-      while (remap[i0] != i0) i0 = remap[i0];
-      while (remap[i1] != i1) i1 = remap[i1];
-      while (remap[i2] != i2) i2 = remap[i2];
+      uint32_t i0 = remap[faces[3*t]], i1 = remap[faces[3*t+1]], i2 = remap[faces[3*t+2]];
       if (i0 == i1 || i1 == i2 || i0 == i2) continue;
       faces[3*out] = i0; faces[3*out+1] = i1; faces[3*out+2] = i2;
       out++;
