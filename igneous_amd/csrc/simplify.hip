@@ -122,17 +122,34 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
   uint32_t lab = flab[t];
   if (!active_lab[lab]) return;
   const uint32_t vb = vbase[lab];
+  // stage the 3 corner quadrics + positions in registers once per face
+  // (the a<b swap below defeats CSE of the Q loads otherwise)
+  uint32_t fc[3] = {faces_g[3*t], faces_g[3*t+1], faces_g[3*t+2]};
+  float cq[3][10], cp[3][3];
+  #pragma unroll
+  for (int ci = 0; ci < 3; ++ci) {
+    #pragma unroll
+    for (int k = 0; k < 10; ++k) cq[ci][k] = Q[10ull*fc[ci] + k];
+    #pragma unroll
+    for (int k = 0; k < 3; ++k) cp[ci][k] = verts[3ull*fc[ci] + k];
+  }
   #pragma unroll
   for (int e = 0; e < 3; ++e) {
-    uint32_t a = faces_g[3*t + e], b = faces_g[3*t + (e+1)%3];
+    const int ea = e, eb = (e + 1) % 3;   // compile-time after unroll
+    uint32_t a = fc[ea], b = fc[eb];
     if (a == b) continue;
-    uint32_t u = a < b ? a : b, w = a < b ? b : a;
-    float mx = 0.5f*(verts[3*u]+verts[3*w]);
-    float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
-    float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+    // canonical u < w; addition is commutative-safe here only because
+    // both operand orders are evaluated identically (x+y) — keep the
+    // oracle's Q[u]+Q[w] order via selects on compile-time indices
+    bool fwd = a < b;
+    uint32_t u = fwd ? a : b, w = fwd ? b : a;
+    float mx = 0.5f*((fwd ? cp[ea][0] : cp[eb][0]) + (fwd ? cp[eb][0] : cp[ea][0]));
+    float my = 0.5f*((fwd ? cp[ea][1] : cp[eb][1]) + (fwd ? cp[eb][1] : cp[ea][1]));
+    float mz = 0.5f*((fwd ? cp[ea][2] : cp[eb][2]) + (fwd ? cp[eb][2] : cp[ea][2]));
     float S[10];
     #pragma unroll
-    for (int k = 0; k < 10; ++k) S[k] = Q[10ull*u + k] + Q[10ull*w + k];
+    for (int k = 0; k < 10; ++k)
+      S[k] = (fwd ? cq[ea][k] : cq[eb][k]) + (fwd ? cq[eb][k] : cq[ea][k]);
     float cost = sq_eval(S, mx, my, mz);
     if (cost < 0.0f) cost = 0.0f;
     if (cost > max_cost) continue;
