@@ -143,40 +143,6 @@ __device__ __forceinline__ void load_corners(const T *__restrict__ labels,
   c[6] = p[sxy + sx];     c[7] = p[sxy + sx + 1];
 }
 
-// Cooperative stage of the 4 voxel rows a 64-cell segment's corners live
-// in (65 voxels each), into this wave's private LDS slice. Coalesced
-// wave-wide loads replace 8 scalar latency-bound loads per cell; corners
-// are then conflict-free ds_reads (lane, lane+1).
-template <typename T>
-__device__ __forceinline__ void stage_rows(const T *__restrict__ labels,
-                                           int64_t sx, int64_t sxy,
-                                           int64_t cx0, int64_t cy,
-                                           int64_t cz, int lane,
-                                           T (*lds)[65]) {
-  const T *base = labels + cx0 + cy * sx + cz * sxy;
-  const int64_t off[4] = {0, sx, sxy, sxy + sx};
-  const int maxx = (int)((sx - cx0) < 65 ? (sx - cx0) : 65);
-  #pragma unroll
-  for (int i = 0; i < 5; ++i) {
-    int idx = lane + i * WAVE;
-    if (idx >= 4 * 65) break;
-    int r = idx / 65, x = idx - r * 65;
-    if (x < maxx) lds[r][x] = base[off[r] + x];
-  }
-  // order this wave's LDS writes before its reads (per-wave slice: no
-  // cross-wave dependence, so no block barrier)
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-}
-
-template <typename T>
-__device__ __forceinline__ void corners_from_lds(const T (*lds)[65],
-                                                 int lane, T c[8]) {
-  c[0] = lds[0][lane];     c[1] = lds[0][lane + 1];
-  c[2] = lds[1][lane];     c[3] = lds[1][lane + 1];
-  c[4] = lds[2][lane];     c[5] = lds[2][lane + 1];
-  c[6] = lds[3][lane];     c[7] = lds[3][lane + 1];
-}
-
 template <typename T>
 __device__ __forceinline__ uint32_t cell_tri_count(const T c[8],
                                                    const uint8_t *cnt_tab) {
@@ -211,7 +177,6 @@ template <typename T>
 __global__ void k_count(const T *__restrict__ labels, GridDims g,
                         uint32_t *__restrict__ segcnt, LabelHash lh) {
   __shared__ uint8_t s_cnt[256];
-  __shared__ T s_rows[4][4][65];  // [wave][row][voxel]
   for (int k = threadIdx.x; k < 256; k += blockDim.x)
     s_cnt[k] = MC_TRI_COUNT[k];
   __syncthreads();
@@ -227,12 +192,10 @@ __global__ void k_count(const T *__restrict__ labels, GridDims g,
     const int64_t cy = row % g.ncy;
     const int64_t cz = row / g.ncy;
     const int64_t cx = segx * WAVE + lane;
-    stage_rows(labels, g.sx, sxy, segx * WAVE, cy, cz, lane,
-               s_rows[wave_in_blk]);
     uint32_t cnt = 0;
     if (cx < g.ncx) {
       T c[8];
-      corners_from_lds(s_rows[wave_in_blk], lane, c);
+      load_corners(labels, g.sx, sxy, cx, cy, cz, c);
       if (!(c[0] == c[1] && c[0] == c[2] && c[0] == c[3] && c[0] == c[4] &&
             c[0] == c[5] && c[0] == c[6] && c[0] == c[7])) {
         #pragma unroll
@@ -274,7 +237,6 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
   __shared__ uint8_t s_cnt[256];
   __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
   __shared__ uint32_t s_comb[12];  // eoff*3 + axis, folded slot math
-  __shared__ T s_rows[4][4][65];   // [wave][row][voxel]
   for (int k = threadIdx.x; k < 256; k += blockDim.x) {
     s_cnt[k] = MC_TRI_COUNT[k];
     #pragma unroll
@@ -308,10 +270,8 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
     T c[8];
     uint32_t cnt = 0;
     bool active = false;
-    stage_rows(labels, g.sx, sxy, segx * WAVE, cy, cz, lane,
-               s_rows[wave_in_blk]);
     if (cx < g.ncx) {
-      corners_from_lds(s_rows[wave_in_blk], lane, c);
+      load_corners(labels, g.sx, sxy, cx, cy, cz, c);
       cnt = cell_tri_count(c, s_cnt);
       active = cnt > 0;
     }
