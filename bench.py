@@ -127,8 +127,15 @@ def main():
     from igneous_amd.engine import Engine
     from igneous_amd.synth import voronoi_labels
 
-    # per-rank chunk: same stats, different seed per rank (weak scaling)
-    data = voronoi_labels(SHAPE, K_SEEDS, SEED + rank, dtype=np.uint64)
+    # weak scaling: every rank meshes its own copy of the same synthetic
+    # chunk (chunks are independent; identical per-rank work). Rank 0
+    # generates (or loads) the cached chunk; other ranks wait, then read
+    # the cache — avoids N concurrent 30s generations.
+    if dist is not None and rank != 0:
+        dist.barrier()
+    data = voronoi_labels(SHAPE, K_SEEDS, SEED, dtype=np.uint64)
+    if dist is not None and rank == 0:
+        dist.barrier()
     eng = Engine.get(local_rank)
 
     red = 100 if args.simplify else 0
