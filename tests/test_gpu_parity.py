@@ -246,3 +246,27 @@ def test_simplify_512_runs(eng):
         v, f = got[lab]
         if f.shape[0]:
             assert f.max() < v.shape[0]
+
+
+def test_mesher_api_gpu(eng):
+    """The zmesh.Mesher drop-in API surface on the real engine
+    (mesh.py:151,245,374-381 call pattern)."""
+    import oracle
+    from igneous_amd import Mesher
+    data = np.zeros((20, 20, 20), dtype=np.uint64, order="F")
+    data[2:18, 2:18, 2:18] = 4
+    data[5:12, 5:12, 5:12] = 9
+    mesher = Mesher((16.0, 16.0, 40.0))
+    mesher.mesh(data, preserve_order=False)
+    ids = mesher.ids()
+    assert ids == [4, 9]
+    want = oracle.mesh_chunk(data, resolution=(16.0, 16.0, 40.0),
+                             reduction_factor=10, max_error=1e9)
+    for lab in ids:
+        m = mesher.get(lab, reduction_factor=10, max_error=1e9,
+                       voxel_centered=True)
+        assert np.array_equal(m.vertices, want[lab][0])
+        assert np.array_equal(m.faces, want[lab][1])
+    mesher.erase(4)
+    assert mesher.ids() == [9]
+    mesher.clear()
