@@ -107,6 +107,14 @@ def main():
     ap.add_argument("--simplify", action="store_true",
                     help="BASELINE config 5: simplification_factor=100, "
                          "max_error=40 through the quadric-collapse kernels")
+    ap.add_argument("--mode", choices=["chunk512", "chunks256"],
+                    default="chunk512",
+                    help="chunk512: the headline single-chunk device rate; "
+                         "chunks256: BASELINE configs[3] shape — a fan-out "
+                         "of 256^3 chunks through the FULL path (H2D + "
+                         "kernels + host extract) on --streams HIP "
+                         "streams/contexts per GPU")
+    ap.add_argument("--streams", type=int, default=3)
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -139,9 +147,34 @@ def main():
     eng = Engine.get(local_rank)
 
     red = 100 if args.simplify else 0
-    def step(skip_h2d=True):
-        eng.mesh_chunk(data, resolution=RESOLUTION, reduction_factor=red,
-                       max_error=40.0, device_only=True, skip_h2d=skip_h2d)
+
+    if args.mode == "chunks256":
+        # configs[3] shape: independent 256^3 chunks, full path incl.
+        # H2D and host extract, overlapped on per-thread HIP contexts
+        from concurrent.futures import ThreadPoolExecutor
+        from igneous_amd import engine as engine_mod
+        engine_mod.PER_THREAD_CTX = True
+        nchunks = 8
+        chunks = [voronoi_labels((256, 256, 256), 6250, 1000 + i,
+                                 dtype=np.uint64) for i in range(nchunks)]
+        pool = ThreadPoolExecutor(max_workers=args.streams)
+
+        def mesh_one(chunk):
+            e = engine_mod.Engine.get(local_rank)
+            return e.mesh_chunk(chunk, resolution=RESOLUTION,
+                                reduction_factor=red, max_error=40.0)
+
+        def step(skip_h2d=True):
+            list(pool.map(mesh_one, chunks))
+
+        step_vox = nchunks * 256 ** 3
+    else:
+        def step(skip_h2d=True):
+            eng.mesh_chunk(data, resolution=RESOLUTION, reduction_factor=red,
+                           max_error=40.0, device_only=True,
+                           skip_h2d=skip_h2d)
+
+        step_vox = int(np.prod(SHAPE))
 
     # warmup (first call stages the labels into HBM)
     step(skip_h2d=False)
@@ -167,8 +200,7 @@ def main():
         elapsed = float(t.item())
 
     stats = eng.stats()
-    nvox = int(np.prod(SHAPE))
-    total_vox = nvox * args.steps * world
+    total_vox = step_vox * args.steps * world
     mvox_s = total_vox / elapsed / 1e6
 
     if rank != 0:
@@ -217,7 +249,10 @@ def main():
         "dtype": "u64",
         "data": "synthetic",
         "config": {
-            "workload": WORKLOAD + (" + simplification_factor=100" if args.simplify else ""),
+            "workload": (WORKLOAD if args.mode == "chunk512" else
+                         f"{8} x 256^3 u64 chunks (configs[3] shape), full "
+                         f"H2D+extract path, {args.streams} streams/GPU")
+                        + (" + simplification_factor=100" if args.simplify else ""),
             "chunk": list(SHAPE),
             "labels": K_SEEDS,
             "resolution_nm": list(RESOLUTION),

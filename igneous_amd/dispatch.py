@@ -29,19 +29,34 @@ def shard_tasks(tasks: Iterable, rank: Optional[int] = None,
             yield task
 
 
+def _run_one(task) -> int:
+    if callable(getattr(task, "execute", None)):
+        task.execute()
+    else:
+        task()
+    return 1
+
+
 def execute_tasks(tasks: Iterable, progress: bool = False,
-                  barrier: bool = True) -> int:
+                  barrier: bool = True, streams: int = 1) -> int:
     """Execute this rank's shard of `tasks` on this rank's GPU
     (MESHGINE_DEVICE defaults to LOCAL_RANK inside the engine). Returns
     the number of tasks this rank executed. With torch.distributed
-    initialized (or WORLD_SIZE>1), synchronizes all ranks at the end."""
+    initialized (or WORLD_SIZE>1), synchronizes all ranks at the end.
+
+    streams>1: a thread pool with one HIP stream/context per thread —
+    in-flight chunks overlap H2D, kernels and D2H on one GPU (ctypes
+    releases the GIL during engine calls)."""
     n = 0
-    for task in shard_tasks(tasks):
-        if callable(getattr(task, "execute", None)):
-            task.execute()
-        else:
-            task()
-        n += 1
+    if streams > 1:
+        from concurrent.futures import ThreadPoolExecutor
+        from . import engine
+        engine.PER_THREAD_CTX = True
+        with ThreadPoolExecutor(max_workers=streams) as pool:
+            n = sum(pool.map(_run_one, shard_tasks(tasks)))
+    else:
+        for task in shard_tasks(tasks):
+            n += _run_one(task)
     if barrier:
         _, world = rank_world()
         if world > 1:

@@ -109,6 +109,13 @@ def load_library() -> ctypes.CDLL:
         return lib
 
 
+# When True, Engine.get() returns one context (and HIP stream) per
+# (device, thread): concurrent MeshTasks on one GPU overlap their
+# H2D / kernels / D2H — the reference's "one HIP stream per in-flight
+# chunk" fan-out. Set by igneous_amd.dispatch.
+PER_THREAD_CTX = False
+
+
 class Engine:
     """One mg_ctx on one HIP device."""
 
@@ -128,11 +135,13 @@ class Engine:
 
     @classmethod
     def get(cls, device_id: int = 0) -> "Engine":
+        key = (device_id, threading.get_ident()) if PER_THREAD_CTX \
+            else device_id
         with cls._cls_lock:
-            eng = cls._per_device.get(device_id)
+            eng = cls._per_device.get(key)
             if eng is None:
                 eng = cls(device_id)
-                cls._per_device[device_id] = eng
+                cls._per_device[key] = eng
             return eng
 
     def mesh_chunk(self, labels: np.ndarray, resolution=(1.0, 1.0, 1.0),
