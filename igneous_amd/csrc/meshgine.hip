@@ -516,6 +516,11 @@ struct DevBuf {
   size_t cap = 0;
 };
 
+struct HostBuf {
+  void *ptr = nullptr;
+  size_t cap = 0;
+};
+
 struct mg_ctx {
   int device = 0;
   hipStream_t stream = nullptr;
@@ -534,6 +539,7 @@ struct mg_ctx {
       simp_faces_alt, simp_verts_alt, simp_vbase_alt, simp_meta, simp_ref,
       simp_keep, simp_keep_scan, simp_park, simp_first, simp_troff_final;
   uint64_t lh_slots = 1ull << 20;
+  HostBuf h_verts, h_faces;  // pinned output staging, reused across calls
   hipEvent_t ev[16] = {};
 };
 
@@ -555,6 +561,24 @@ static thread_local std::string g_err;  // for ctx==NULL failures
       return retcode;                                         \
     }                                                         \
   } while (0)
+
+static int ensure_host(mg_ctx *c, HostBuf &b, size_t bytes) {
+  if (b.cap >= bytes) return 0;
+  if (b.ptr) (void)hipHostFree(b.ptr);
+  b.ptr = nullptr;
+  b.cap = 0;
+  size_t want = bytes + bytes / 4;
+  if (hipHostMalloc(&b.ptr, want) != hipSuccess) {
+    if (hipHostMalloc(&b.ptr, bytes) != hipSuccess) {
+      SET_ERR(c, "hipHostMalloc(%zu) failed", bytes);
+      return 1;
+    }
+    b.cap = bytes;
+    return 0;
+  }
+  b.cap = want;
+  return 0;
+}
 
 static int ensure(mg_ctx *c, DevBuf &b, size_t bytes) {
   if (b.cap >= bytes) return 0;
@@ -623,6 +647,8 @@ void mg_destroy(mg_ctx *c) {
                   &c->simp_troff_final}) {
     if (b->ptr) (void)hipFree(b->ptr);
   }
+  if (c->h_verts.ptr) (void)hipHostFree(c->h_verts.ptr);
+  if (c->h_faces.ptr) (void)hipHostFree(c->h_faces.ptr);
   for (auto &e : c->ev) if (e) (void)hipEventDestroy(e);
   if (c->stream) (void)hipStreamDestroy(c->stream);
   delete c;
@@ -639,12 +665,8 @@ int mg_get_stats(mg_ctx *c, mg_stats *out) {
 }
 
 void mg_meshset_free(mg_meshset *ms) {
-  if (!ms) return;
-  // layout: [mg_meshset][mg_mesh array][host buffer pointers]
-  void **bufs = (void **)((char *)ms + sizeof(mg_meshset) +
-                          sizeof(mg_mesh) * ms->nmeshes);
-  for (int i = 0; i < 2; ++i)
-    if (bufs[i]) (void)hipHostFree(bufs[i]);
+  // the flat vertex/face storage is ctx-owned (pinned, reused); the
+  // meshset is just the descriptor + mesh array
   free(ms);
 }
 
@@ -1319,14 +1341,10 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     ms = (mg_meshset *)calloc(1, sizeof(mg_meshset) + 2 * sizeof(void *));
     *out = ms;
   } else {
-    float *h_verts = nullptr;
-    uint32_t *h_faces = nullptr;
-    HIP_TRY(c, hipHostMalloc((void **)&h_verts, Vcur * 12 + 12), 24);
-    if (hipHostMalloc((void **)&h_faces, NCX * 4 + 4) != hipSuccess) {
-      (void)hipHostFree(h_verts);
-      SET_ERR(c, "hipHostMalloc faces failed");
-      return 24;
-    }
+    if (ensure_host(c, c->h_verts, Vcur * 12 + 12)) return 24;
+    if (ensure_host(c, c->h_faces, NCX * 4 + 4)) return 24;
+    float *h_verts = (float *)c->h_verts.ptr;
+    uint32_t *h_faces = (uint32_t *)c->h_faces.ptr;
     std::vector<uint32_t> h_tri_off(nlabels + 1), h_vbase(nlabels + 1);
     std::vector<uint64_t> h_label_values(nlabels);
     if (Vcur > 0)
@@ -1344,12 +1362,13 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     HIP_TRY(c, hipStreamSynchronize(s), 24);
 
     ms = (mg_meshset *)calloc(
-        1, sizeof(mg_meshset) + sizeof(mg_mesh) * nlabels + 2 * sizeof(void *));
+        1, sizeof(mg_meshset) + sizeof(mg_mesh) * nlabels);
     ms->nmeshes = nlabels;
     ms->meshes = (mg_mesh *)((char *)ms + sizeof(mg_meshset));
-    void **bufs = (void **)((char *)ms->meshes + sizeof(mg_mesh) * nlabels);
-    bufs[0] = h_verts;
-    bufs[1] = h_faces;
+    ms->verts_base = h_verts;
+    ms->faces_base = h_faces;
+    ms->total_verts = Vcur;
+    ms->total_tris = Tcur;
     // order meshes by ascending label value
     std::vector<uint32_t> idx(nlabels);
     for (uint32_t i = 0; i < nlabels; ++i) idx[i] = i;

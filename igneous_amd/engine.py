@@ -44,6 +44,10 @@ class _MgMeshSet(ctypes.Structure):
     _fields_ = [
         ("nmeshes", ctypes.c_uint32),
         ("meshes", ctypes.POINTER(_MgMesh)),
+        ("verts_base", ctypes.POINTER(ctypes.c_float)),
+        ("faces_base", ctypes.POINTER(ctypes.c_uint32)),
+        ("total_verts", ctypes.c_uint64),
+        ("total_tris", ctypes.c_uint64),
     ]
 
 
@@ -178,13 +182,29 @@ class Engine:
             result = {}
             try:
                 ms = out.contents
-                for i in range(ms.nmeshes):
-                    m = ms.meshes[i]
-                    v = np.ctypeslib.as_array(
-                        m.verts, shape=(m.nverts, 3)).copy()
-                    f = np.ctypeslib.as_array(
-                        m.faces, shape=(m.ntris, 3)).copy()
-                    result[int(m.label)] = (v, f)
+                if ms.nmeshes:
+                    # single copy of the flat storage, per-label views
+                    # (the flat buffers are ctx-owned pinned staging,
+                    # valid until the next call on this ctx)
+                    vbase_addr = ctypes.cast(
+                        ms.verts_base, ctypes.c_void_p).value
+                    fbase_addr = ctypes.cast(
+                        ms.faces_base, ctypes.c_void_p).value
+                    all_v = np.ctypeslib.as_array(
+                        ms.verts_base,
+                        shape=(int(ms.total_verts), 3)).copy()
+                    all_f = np.ctypeslib.as_array(
+                        ms.faces_base,
+                        shape=(int(ms.total_tris), 3)).copy()
+                    for i in range(ms.nmeshes):
+                        m = ms.meshes[i]
+                        vo = (ctypes.cast(m.verts, ctypes.c_void_p).value
+                              - vbase_addr) // 12
+                        fo = (ctypes.cast(m.faces, ctypes.c_void_p).value
+                              - fbase_addr) // 12
+                        result[int(m.label)] = (
+                            all_v[vo:vo + m.nverts],
+                            all_f[fo:fo + m.ntris])
             finally:
                 self.lib.mg_meshset_free(out)
         return result

@@ -76,7 +76,15 @@ typedef struct {
 typedef struct {
   uint32_t  nmeshes;
   mg_mesh  *meshes;      /* sorted by ascending label id */
-  /* opaque storage follows */
+  /* flat storage the per-mesh pointers alias (single-copy extraction):
+   * verts_base[3*total_verts], faces_base[3*total_tris]. The storage is
+   * OWNED BY THE CTX (pinned, reused): it stays valid until the next
+   * mg_mesh_chunk on the same ctx or mg_destroy. mg_meshset_free frees
+   * only the descriptor. */
+  float    *verts_base;
+  uint32_t *faces_base;
+  uint64_t  total_verts;
+  uint64_t  total_tris;
 } mg_meshset;
 
 /* Per-call kernel timing/stats, HIP-event measured on the engine stream.
