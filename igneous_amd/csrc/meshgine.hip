@@ -1361,14 +1361,19 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                               nlabels * 8, hipMemcpyDeviceToHost, s), 24);
     HIP_TRY(c, hipStreamSynchronize(s), 24);
 
-    ms = (mg_meshset *)calloc(
-        1, sizeof(mg_meshset) + sizeof(mg_mesh) * nlabels);
+    size_t meta_off = sizeof(mg_meshset) + sizeof(mg_mesh) * nlabels;
+    ms = (mg_meshset *)calloc(1, meta_off + (size_t)nlabels * 24);
     ms->nmeshes = nlabels;
     ms->meshes = (mg_mesh *)((char *)ms + sizeof(mg_meshset));
     ms->verts_base = h_verts;
     ms->faces_base = h_faces;
     ms->total_verts = Vcur;
     ms->total_tris = Tcur;
+    ms->labels_arr = (uint64_t *)((char *)ms + meta_off);
+    ms->voff_arr = (uint32_t *)(ms->labels_arr + nlabels);
+    ms->nv_arr = ms->voff_arr + nlabels;
+    ms->foff_arr = ms->nv_arr + nlabels;
+    ms->nf_arr = ms->foff_arr + nlabels;
     // order meshes by ascending label value
     std::vector<uint32_t> idx(nlabels);
     for (uint32_t i = 0; i < nlabels; ++i) idx[i] = i;
@@ -1383,6 +1388,11 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
       mm.ntris = h_tri_off[l + 1] - h_tri_off[l];
       mm.verts = h_verts + 3ull * h_vbase[l];
       mm.faces = h_faces + 3ull * h_tri_off[l];
+      ms->labels_arr[m] = mm.label;
+      ms->voff_arr[m] = h_vbase[l];
+      ms->nv_arr[m] = mm.nverts;
+      ms->foff_arr[m] = h_tri_off[l];
+      ms->nf_arr[m] = mm.ntris;
     }
     *out = ms;
   }

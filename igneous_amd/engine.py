@@ -48,6 +48,11 @@ class _MgMeshSet(ctypes.Structure):
         ("faces_base", ctypes.POINTER(ctypes.c_uint32)),
         ("total_verts", ctypes.c_uint64),
         ("total_tris", ctypes.c_uint64),
+        ("labels_arr", ctypes.POINTER(ctypes.c_uint64)),
+        ("voff_arr", ctypes.POINTER(ctypes.c_uint32)),
+        ("nv_arr", ctypes.POINTER(ctypes.c_uint32)),
+        ("foff_arr", ctypes.POINTER(ctypes.c_uint32)),
+        ("nf_arr", ctypes.POINTER(ctypes.c_uint32)),
     ]
 
 
@@ -182,29 +187,35 @@ class Engine:
             result = {}
             try:
                 ms = out.contents
-                if ms.nmeshes:
-                    # single copy of the flat storage, per-label views
-                    # (the flat buffers are ctx-owned pinned staging,
-                    # valid until the next call on this ctx)
-                    vbase_addr = ctypes.cast(
-                        ms.verts_base, ctypes.c_void_p).value
-                    fbase_addr = ctypes.cast(
-                        ms.faces_base, ctypes.c_void_p).value
-                    all_v = np.ctypeslib.as_array(
-                        ms.verts_base,
-                        shape=(int(ms.total_verts), 3)).copy()
-                    all_f = np.ctypeslib.as_array(
-                        ms.faces_base,
-                        shape=(int(ms.total_tris), 3)).copy()
-                    for i in range(ms.nmeshes):
-                        m = ms.meshes[i]
-                        vo = (ctypes.cast(m.verts, ctypes.c_void_p).value
-                              - vbase_addr) // 12
-                        fo = (ctypes.cast(m.faces, ctypes.c_void_p).value
-                              - fbase_addr) // 12
-                        result[int(m.label)] = (
-                            all_v[vo:vo + m.nverts],
-                            all_f[fo:fo + m.ntris])
+                n = int(ms.nmeshes)
+                if n:
+                    # one GIL-releasing memmove per flat buffer (the flat
+                    # buffers are ctx-owned pinned staging, valid until
+                    # the next call on this ctx), then per-label views
+                    tv, tt = int(ms.total_verts), int(ms.total_tris)
+                    all_v = np.empty((tv, 3), dtype=np.float32)
+                    all_f = np.empty((tt, 3), dtype=np.uint32)
+                    if tv:
+                        ctypes.memmove(all_v.ctypes.data, ms.verts_base,
+                                       tv * 12)
+                    if tt:
+                        ctypes.memmove(all_f.ctypes.data, ms.faces_base,
+                                       tt * 12)
+                    labels = np.ctypeslib.as_array(
+                        ms.labels_arr, shape=(n,)).tolist()
+                    voff = np.ctypeslib.as_array(
+                        ms.voff_arr, shape=(n,)).tolist()
+                    nv = np.ctypeslib.as_array(
+                        ms.nv_arr, shape=(n,)).tolist()
+                    foff = np.ctypeslib.as_array(
+                        ms.foff_arr, shape=(n,)).tolist()
+                    nf = np.ctypeslib.as_array(
+                        ms.nf_arr, shape=(n,)).tolist()
+                    result = {
+                        lab: (all_v[vo:vo + kv], all_f[fo:fo + kf])
+                        for lab, vo, kv, fo, kf
+                        in zip(labels, voff, nv, foff, nf)
+                    }
             finally:
                 self.lib.mg_meshset_free(out)
         return result

@@ -85,6 +85,14 @@ typedef struct {
   uint32_t *faces_base;
   uint64_t  total_verts;
   uint64_t  total_tris;
+  /* flat per-mesh metadata (same order as meshes[]; ctx-independent,
+   * stored in the descriptor allocation): hosts without struct-walking
+   * FFI can slice the flat storage from these. */
+  uint64_t *labels_arr;   /* [nmeshes] */
+  uint32_t *voff_arr;     /* [nmeshes] vertex offset into verts_base/3 */
+  uint32_t *nv_arr;       /* [nmeshes] */
+  uint32_t *foff_arr;     /* [nmeshes] face offset into faces_base/3 */
+  uint32_t *nf_arr;       /* [nmeshes] */
 } mg_meshset;
 
 /* Per-call kernel timing/stats, HIP-event measured on the engine stream.
