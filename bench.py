@@ -161,8 +161,10 @@ def main():
 
         def mesh_one(chunk):
             e = engine_mod.Engine.get(local_rank)
+            # results discarded before the thread's next call: zero-copy
             return e.mesh_chunk(chunk, resolution=RESOLUTION,
-                                reduction_factor=red, max_error=40.0)
+                                reduction_factor=red, max_error=40.0,
+                                copy=False)
 
         def step(skip_h2d=True):
             list(pool.map(mesh_one, chunks))

@@ -157,10 +157,16 @@ class Engine:
                    reduction_factor: int = 0, max_error: float = 40.0,
                    voxel_centered: bool = True,
                    device_only: bool = False,
-                   skip_h2d: bool = False) -> dict:
+                   skip_h2d: bool = False,
+                   copy: bool = True) -> dict:
         """GPU counterpart of oracle.mesh_chunk: F-order (sx,sy,sz)
         uint32/uint64 labels -> {label: (verts (V,3) f32 nm, faces (F,3) u32)},
-        labels ascending. Under device_only returns {} (stats still filled)."""
+        labels ascending. Under device_only returns {} (stats still filled).
+
+        copy=False returns VIEWS into this context's pinned staging
+        buffers — zero host copies, but the arrays are only valid until
+        the next mesh_chunk on this Engine. Use when results are consumed
+        (encoded/written) before the next call, as MeshTask does."""
         labels = np.asfortranarray(labels)
         if labels.dtype == np.uint32:
             dtype = MG_U32
@@ -193,14 +199,20 @@ class Engine:
                     # buffers are ctx-owned pinned staging, valid until
                     # the next call on this ctx), then per-label views
                     tv, tt = int(ms.total_verts), int(ms.total_tris)
-                    all_v = np.empty((tv, 3), dtype=np.float32)
-                    all_f = np.empty((tt, 3), dtype=np.uint32)
-                    if tv:
-                        ctypes.memmove(all_v.ctypes.data, ms.verts_base,
-                                       tv * 12)
-                    if tt:
-                        ctypes.memmove(all_f.ctypes.data, ms.faces_base,
-                                       tt * 12)
+                    if copy:
+                        all_v = np.empty((tv, 3), dtype=np.float32)
+                        all_f = np.empty((tt, 3), dtype=np.uint32)
+                        if tv:
+                            ctypes.memmove(all_v.ctypes.data, ms.verts_base,
+                                           tv * 12)
+                        if tt:
+                            ctypes.memmove(all_f.ctypes.data, ms.faces_base,
+                                           tt * 12)
+                    else:
+                        all_v = np.ctypeslib.as_array(
+                            ms.verts_base, shape=(max(tv, 1), 3))[:tv]
+                        all_f = np.ctypeslib.as_array(
+                            ms.faces_base, shape=(max(tt, 1), 3))[:tt]
                     labels = np.ctypeslib.as_array(
                         ms.labels_arr, shape=(n,)).tolist()
                     voff = np.ctypeslib.as_array(
@@ -230,10 +242,12 @@ class Engine:
 
 def mesh_chunk(labels: np.ndarray, resolution=(1.0, 1.0, 1.0),
                reduction_factor: int = 0, max_error: float = 40.0,
-               voxel_centered: bool = True, device_id: Optional[int] = None) -> dict:
+               voxel_centered: bool = True, device_id: Optional[int] = None,
+               copy: bool = True) -> dict:
     """Module-level product mesher (the default MeshTask path)."""
     if device_id is None:
         device_id = int(os.environ.get("MESHGINE_DEVICE",
                                        os.environ.get("LOCAL_RANK", "0")))
     return Engine.get(device_id).mesh_chunk(
-        labels, resolution, reduction_factor, max_error, voxel_centered)
+        labels, resolution, reduction_factor, max_error, voxel_centered,
+        copy=copy)

@@ -173,13 +173,26 @@ class MeshTask(RegisteredTask):
         data = data[..., 0]
 
         mesher = _get_mesher()
-        raw = mesher(
-            data,
-            resolution=tuple(float(r) for r in self._volume.resolution),
-            reduction_factor=int(opts['simplification_factor'] or 0),
-            max_error=float(opts['max_simplification_error']),
-            voxel_centered=True,
-        )
+        # copy=False where supported: this task consumes (shifts, encodes,
+        # uploads) every mesh before its next engine call, so zero-copy
+        # views into the engine's staging buffers are safe
+        try:
+            raw = mesher(
+                data,
+                resolution=tuple(float(r) for r in self._volume.resolution),
+                reduction_factor=int(opts['simplification_factor'] or 0),
+                max_error=float(opts['max_simplification_error']),
+                voxel_centered=True,
+                copy=False,
+            )
+        except TypeError:
+            raw = mesher(
+                data,
+                resolution=tuple(float(r) for r in self._volume.resolution),
+                reduction_factor=int(opts['simplification_factor'] or 0),
+                max_error=float(opts['max_simplification_error']),
+                voxel_centered=True,
+            )
         del data
         meshes = {
             int(label): Mesh(verts, faces, id=int(label))
