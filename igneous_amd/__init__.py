@@ -11,6 +11,7 @@ from .meshes import Mesh
 from .queue import LocalTaskQueue, RegisteredTask
 from .tasks import (
     MeshTask, MeshManifestPrefixTask, MeshManifestFilesystemTask,
+    TransferMeshFilesTask, DeleteMeshFilesTask,
 )
 from .task_creation import create_meshing_tasks
 from .volume import PrecomputedVolume

@@ -2,5 +2,7 @@ from .mesh import (
     MeshTask,
     MeshManifestPrefixTask,
     MeshManifestFilesystemTask,
+    TransferMeshFilesTask,
+    DeleteMeshFilesTask,
     set_mesher,
 )

@@ -378,3 +378,28 @@ def MeshManifestPrefixTask(layer_path: str, prefix: str, lod: int = 0,
     cf.put_jsons(
         (f"{mesh_dir}/{segid}:{lod}", {"fragments": frags})
         for segid, frags in segids.items())
+
+
+def TransferMeshFilesTask(src: str, dest: str, prefix: str,
+                          mesh_dir=None):
+    """Copy mesh files between layers (reference mesh.py:726-739)."""
+    from ..volume import PrecomputedVolume
+    cv_src = PrecomputedVolume(src)
+    cv_dest = PrecomputedVolume(dest)
+    sdir = mesh_dir or cv_src.info.get('mesh', 'mesh')
+    ddir = mesh_dir or cv_dest.info.get('mesh', sdir)
+    cf_src = CloudFiles(f"{src.rstrip('/')}/{sdir}")
+    cf_dest = CloudFiles(f"{dest.rstrip('/')}/{ddir}")
+    for name in cf_src.list(prefix=prefix):
+        data = cf_src.get(name)
+        if data is not None:
+            cf_dest.put(name, data)
+
+
+def DeleteMeshFilesTask(cloudpath: str, prefix: str, mesh_dir=None):
+    """Delete mesh files under a prefix (reference mesh.py:741-749)."""
+    from ..volume import PrecomputedVolume
+    cv = PrecomputedVolume(cloudpath)
+    mdir = mesh_dir or cv.info.get('mesh', 'mesh')
+    cf = CloudFiles(f"{cloudpath.rstrip('/')}/{mdir}")
+    cf.delete(list(cf.list(prefix=prefix)))

@@ -290,3 +290,21 @@ def test_chunk_seams_stitch(tmp_layer_path, oracle_mesher):
     seam0 = {tuple(v) for v in m0.vertices[np.isclose(m0.vertices[:, 0], 10.0)]}
     seam1 = {tuple(v) for v in m1.vertices[np.isclose(m1.vertices[:, 0], 10.0)]}
     assert seam0 and seam0 == seam1
+
+
+def test_transfer_and_delete_mesh_files(tmp_path, oracle_mesher):
+    """Mirror of reference TransferMeshFilesTask/DeleteMeshFilesTask
+    (mesh.py:726-749)."""
+    from igneous_amd.tasks import TransferMeshFilesTask, DeleteMeshFilesTask
+    src = f"file://{tmp_path}/src"
+    dst = f"file://{tmp_path}/dst"
+    _make_box_layer(src)
+    PrecomputedVolume.from_numpy(
+        np.zeros((8, 8, 8), np.uint32), dst, mesh_dir="mesh")
+    MeshTask(shape=(64, 64, 64), offset=(0, 0, 0), layer_path=src,
+             mip=0, simplification_factor=0).execute()
+    TransferMeshFilesTask(src=src, dest=dst, prefix="1:")
+    cf_dst = CloudFiles(dst)
+    assert cf_dst.get("mesh/1:0:0-64_0-64_0-64") is not None
+    DeleteMeshFilesTask(cloudpath=dst, prefix="1:")
+    assert list(cf_dst.list("mesh/")) == []
