@@ -537,7 +537,8 @@ struct mg_ctx {
       simp_fq, simp_valid, simp_pk, simp_pk_alt, simp_pv, simp_pv_alt,
       simp_Q, simp_pick, simp_remap, simp_flab, simp_flab_alt,
       simp_faces_alt, simp_verts_alt, simp_vbase_alt, simp_meta, simp_ref,
-      simp_keep, simp_keep_scan, simp_park, simp_first, simp_troff_final;
+      simp_keep, simp_keep_scan, simp_park, simp_first, simp_troff_final,
+      simp_deg, simp_adj;
   uint64_t lh_slots = 1ull << 20;
   HostBuf h_verts, h_faces;  // pinned output staging, reused across calls
   hipEvent_t ev[16] = {};
@@ -644,7 +645,7 @@ void mg_destroy(mg_ctx *c) {
                   &c->simp_faces_alt, &c->simp_verts_alt, &c->simp_vbase_alt,
                   &c->simp_meta, &c->simp_ref, &c->simp_keep,
                   &c->simp_keep_scan, &c->simp_park, &c->simp_first,
-                  &c->simp_troff_final}) {
+                  &c->simp_troff_final, &c->simp_deg, &c->simp_adj}) {
     if (b->ptr) (void)hipFree(b->ptr);
   }
   if (c->h_verts.ptr) (void)hipHostFree(c->h_verts.ptr);
@@ -759,6 +760,8 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   if (ensure(c, c->simp_park, 3 * T * 4)) return 40;
   if (ensure(c, c->simp_first, (L + 1) * 4)) return 40;
   if (ensure(c, c->simp_troff_final, (L + 1) * 4)) return 40;
+  if (ensure(c, c->simp_deg, V * 4)) return 40;
+  if (ensure(c, c->simp_adj, V * 4)) return 40;
 
   uint32_t *faces_g = (uint32_t *)c->faces.ptr;
   float *verts = (float *)c->verts.ptr;
@@ -776,6 +779,72 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                        active, reduction_factor, (uint32_t)L);
   }
   HIP_TRY(c, hipGetLastError(), 40);
+
+  // labels up to SIMP_BIG_CAP faces simplify entirely inside one
+  // workgroup each (cache-resident round loop, no global sorts); only
+  // bigger labels go through the global-rounds machinery below.
+  const uint32_t SIMP_BIG_CAP = 65536;
+  {
+    hipLaunchKernelGGL(k_simplify_label, dim3((uint32_t)L), dim3(256), 0, s,
+                       faces_g, (uint32_t *)c->simp_faces_alt.ptr,
+                       (const uint32_t *)c->tri_off.ptr,
+                       (const uint32_t *)c->vbase.ptr,
+                       verts, (float *)c->simp_Q.ptr,
+                       (unsigned long long *)c->simp_pick.ptr,
+                       (uint32_t *)c->simp_remap.ptr,
+                       (uint32_t *)c->simp_deg.ptr,
+                       (uint32_t *)c->simp_adj.ptr,
+                       (uint32_t *)c->simp_pk.ptr,
+                       (SimpPlane *)c->simp_fq.ptr,
+                       (uint8_t *)c->simp_valid.ptr,
+                       nt_cur, target, active,
+                       (uint32_t *)c->simp_park.ptr,
+                       max_cost, (uint32_t)L, SIMP_BIG_CAP);
+    HIP_TRY(c, hipGetLastError(), 40);
+  }
+  // drop the per-label-kernel labels from the working set; park big
+  // never-active labels (their final faces are their originals)
+  {
+    uint64_t nbt2 = (T + blk - 1) / blk;
+    uint32_t *flab2 = (uint32_t *)c->simp_flab.ptr;
+    hipLaunchKernelGGL(k_flag_active_faces, dim3((uint32_t)nbt2), dim3(blk),
+                       0, s, flab2, active, (uint32_t *)c->simp_keep.ptr, T);
+    size_t tmp = 0;
+    hipError_t e = rocprim::exclusive_scan(
+        nullptr, tmp, (uint32_t *)c->simp_keep.ptr,
+        (uint32_t *)c->simp_keep_scan.ptr, 0u, T,
+        rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "prepark scan size"); return 50; }
+    if (ensure(c, c->scan_tmp, tmp)) return 50;
+    e = rocprim::exclusive_scan(
+        c->scan_tmp.ptr, tmp, (uint32_t *)c->simp_keep.ptr,
+        (uint32_t *)c->simp_keep_scan.ptr, 0u, T,
+        rocprim::plus<uint32_t>(), s);
+    if (e != hipSuccess) { SET_ERR(c, "prepark scan"); return 50; }
+    hipLaunchKernelGGL(k_first_label_idx, dim3((uint32_t)nbt2), dim3(blk),
+                       0, s, flab2, (uint32_t *)c->simp_first.ptr, T);
+    hipLaunchKernelGGL(k_park_scatter, dim3((uint32_t)nbt2), dim3(blk), 0,
+                       s, faces_g, flab2,
+                       (const uint32_t *)c->simp_keep.ptr,
+                       (const uint32_t *)c->simp_keep_scan.ptr,
+                       (const uint32_t *)c->simp_first.ptr,
+                       (const uint32_t *)c->tri_off.ptr,
+                       (uint32_t *)c->simp_faces_alt.ptr,
+                       (uint32_t *)c->simp_flab_alt.ptr,
+                       (uint32_t *)c->simp_park.ptr, SIMP_BIG_CAP, T);
+    hipLaunchKernelGGL(k_last_sum, dim3(1), dim3(1), 0, s,
+                       (const uint32_t *)c->simp_keep_scan.ptr,
+                       (const uint32_t *)c->simp_keep.ptr, T,
+                       (uint32_t *)c->lh_misc.ptr + 4);
+    uint32_t act_total = 0;
+    HIP_TRY(c, hipMemcpyAsync(&act_total, (uint32_t *)c->lh_misc.ptr + 4,
+                              4, hipMemcpyDeviceToHost, s), 50);
+    HIP_TRY(c, hipStreamSynchronize(s), 50);
+    std::swap(c->faces, c->simp_faces_alt);
+    std::swap(c->simp_flab, c->simp_flab_alt);
+    faces_g = (uint32_t *)c->faces.ptr;
+    T = act_total;
+  }
 
   for (int round = 0; round < 65536; ++round) {
     // any label still active?
@@ -926,7 +995,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                          (const uint32_t *)c->tri_off.ptr,
                          (uint32_t *)c->simp_faces_alt.ptr,
                          (uint32_t *)c->simp_flab_alt.ptr,
-                         (uint32_t *)c->simp_park.ptr, T);
+                         (uint32_t *)c->simp_park.ptr, SIMP_BIG_CAP, T);
       hipLaunchKernelGGL(k_last_sum, dim3(1), dim3(1), 0, s,
                          (const uint32_t *)c->simp_keep_scan.ptr,
                          (const uint32_t *)c->simp_keep.ptr, T,
@@ -969,7 +1038,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                        (const uint32_t *)c->tri_off.ptr,
                        (uint32_t *)c->simp_faces_alt.ptr,
                        (uint32_t *)c->simp_flab_alt.ptr,
-                       (uint32_t *)c->simp_park.ptr, T);
+                       (uint32_t *)c->simp_park.ptr, SIMP_BIG_CAP, T);
     HIP_TRY(c, hipGetLastError(), 48);
   }
 

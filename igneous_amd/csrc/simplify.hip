@@ -344,10 +344,15 @@ __global__ void k_park_scatter(const uint32_t *__restrict__ faces_g,
                                uint32_t *__restrict__ work_faces,
                                uint32_t *__restrict__ work_flab,
                                uint32_t *__restrict__ park_faces,
+                               uint32_t skip_cap,  // labels <= this size
+                               // were parked by k_simplify_label: drop
                                uint64_t ntris) {
   uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= ntris) return;
   uint32_t lab = flab[t];
+  if (!aflag[t] &&
+      orig_tri_off[lab + 1] - orig_tri_off[lab] <= skip_cap)
+    return;  // small label: its park slice is already final
   if (aflag[t]) {
     uint64_t o = apos[t];
     work_faces[3*o] = faces_g[3*t];
@@ -383,4 +388,243 @@ __global__ void k_gather_final(const uint32_t *__restrict__ park_faces,
   faces_out[3*f+1] = park_faces[3*src+1];
   faces_out[3*f+2] = park_faces[3*src+2];
   flab_out[f] = l;
+}
+
+// ---------------------------------------------------------------------------
+// Per-label simplification: one workgroup runs a label's ENTIRE round
+// loop over its own slices of the global arrays (faces/verts/Q/pick/
+// remap + a per-label CSR). A label's working set (~100-300 KB) stays
+// cache-resident and there are no global sorts, no host round trips and
+// no device-wide synchronization — all 50k labels simplify concurrently.
+// Labels larger than `big_cap` faces are left to the global-rounds path.
+//
+// Arithmetic and schedule are IDENTICAL to oracle/simplify.c per label
+// (same plane/quadric/cost expressions, ascending-face quadric order via
+// the sorted CSR, jittered pick encoding on label-local ids, mutual-pick
+// matched collapse, stable compaction, same termination conditions).
+
+// block-wide exclusive scan of src[0..n) into dst (+base), returns total
+__device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
+                               uint32_t n, uint32_t base,
+                               uint32_t *s_sums /*257*/) {
+  const uint32_t tid = threadIdx.x;
+  const uint32_t chunk = (n + 255) / 256;
+  const uint32_t lo = tid * chunk;
+  const uint32_t hi = lo + chunk < n ? lo + chunk : n;
+  uint32_t sum = 0;
+  for (uint32_t i = lo; i < hi; ++i) sum += src[i];
+  s_sums[tid] = sum;
+  __syncthreads();
+  if (tid == 0) {
+    uint32_t acc = 0;
+    for (int k = 0; k < 256; ++k) {
+      uint32_t t = s_sums[k];
+      s_sums[k] = acc;
+      acc += t;
+    }
+    s_sums[256] = acc;
+  }
+  __syncthreads();
+  uint32_t run = base + s_sums[tid];
+  for (uint32_t i = lo; i < hi; ++i) {
+    uint32_t t = src[i];
+    dst[i] = run;
+    run += t;
+  }
+  uint32_t total = s_sums[256];
+  __syncthreads();
+  return total;
+}
+
+__global__ __launch_bounds__(256) void k_simplify_label(
+    uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
+    uint32_t *__restrict__ faces_tmp,     // same slicing (scratch)
+    const uint32_t *__restrict__ tri_off, // L+1 (original offsets)
+    const uint32_t *__restrict__ vbase,   // L+1
+    float *__restrict__ verts,
+    float *__restrict__ Q,                // 10 per vertex
+    unsigned long long *__restrict__ pick,
+    uint32_t *__restrict__ remap,
+    uint32_t *__restrict__ deg,           // per vertex (doubles as cursor)
+    uint32_t *__restrict__ adj_off,       // per vertex
+    uint32_t *__restrict__ cols,          // 3 per face slot (CSR payload)
+    SimpPlane *__restrict__ fq,           // per face (round scratch)
+    uint8_t *__restrict__ fvalid,         // per face (plane valid / keep)
+    uint32_t *__restrict__ nt_cur,
+    const uint32_t *__restrict__ target,
+    uint8_t *__restrict__ active,
+    uint32_t *__restrict__ park_faces,
+    float max_cost, uint32_t nlabels, uint32_t big_cap) {
+  const uint32_t b = blockIdx.x;
+  if (b >= nlabels) return;
+  const uint32_t f0 = tri_off[b];
+  const uint32_t nt0 = tri_off[b + 1] - f0;
+  if (nt0 > big_cap) return;  // global-rounds path handles big labels
+  if (!active[b]) {
+    // already at/below target: final faces = original faces; park them
+    for (uint32_t i = threadIdx.x; i < 3 * nt0; i += 256)
+      park_faces[3ull * f0 + i] = faces_g[3ull * f0 + i];
+    return;
+  }
+  const uint32_t v0 = vbase[b];
+  const uint32_t nv = vbase[b + 1] - v0;
+  const uint32_t tgt = target[b];
+  const uint32_t tid = threadIdx.x;
+  uint32_t *faces = faces_g + 3ull * f0;
+  uint32_t *ftmp = faces_tmp + 3ull * f0;
+  SimpPlane *pl = fq + f0;
+  uint8_t *valid = fvalid + f0;
+  uint32_t *dg = deg + v0;
+  uint32_t *aoff = adj_off + v0;
+  uint32_t *cl = cols + 3ull * f0;
+
+  __shared__ uint32_t s_sums[257];
+  __shared__ uint32_t s_nt, s_collapses;
+  if (tid == 0) s_nt = nt0;
+  __syncthreads();
+
+  for (int round = 0; round < 65536; ++round) {
+    const uint32_t nt = s_nt;
+    if (nt <= tgt) break;
+
+    // [1] face planes (oracle step 1 preamble)
+    for (uint32_t f = tid; f < nt; f += 256) {
+      uint32_t i0 = faces[3*f], i1 = faces[3*f+1], i2 = faces[3*f+2];
+      const float *p0 = verts + 3ull*i0, *p1 = verts + 3ull*i1,
+                  *p2 = verts + 3ull*i2;
+      float ux = p1[0]-p0[0], uy = p1[1]-p0[1], uz = p1[2]-p0[2];
+      float vx = p2[0]-p0[0], vy = p2[1]-p0[1], vz = p2[2]-p0[2];
+      float nx = uy*vz - uz*vy, ny = uz*vx - ux*vz, nz = ux*vy - uy*vx;
+      float len = sqrtf(nx*nx + ny*ny + nz*nz);
+      if (len <= 0.0f) { valid[f] = 0; continue; }
+      float inv = 1.0f / len;
+      nx *= inv; ny *= inv; nz *= inv;
+      float d = -(nx*p0[0] + ny*p0[1] + nz*p0[2]);
+      pl[f] = SimpPlane{nx, ny, nz, d};
+      valid[f] = 1;
+    }
+    // [2] CSR degrees
+    for (uint32_t v = tid; v < nv; v += 256) dg[v] = 0;
+    __syncthreads();
+    for (uint32_t f = tid; f < nt; f += 256) {
+      atomicAdd(&dg[faces[3*f] - v0], 1u);
+      atomicAdd(&dg[faces[3*f+1] - v0], 1u);
+      atomicAdd(&dg[faces[3*f+2] - v0], 1u);
+    }
+    __syncthreads();
+    // [3] offsets (local, base 0 into cl)
+    blk_exscan(dg, aoff, nv, 0, s_sums);
+    // [4] fill (cursor = deg reused)
+    for (uint32_t v = tid; v < nv; v += 256) dg[v] = aoff[v];
+    __syncthreads();
+    for (uint32_t f = tid; f < nt; f += 256) {
+      cl[atomicAdd(&dg[faces[3*f] - v0], 1u)] = f;
+      cl[atomicAdd(&dg[faces[3*f+1] - v0], 1u)] = f;
+      cl[atomicAdd(&dg[faces[3*f+2] - v0], 1u)] = f;
+    }
+    __syncthreads();
+    // [5] per-vertex: sort incident faces ascending (insertion sort),
+    // accumulate quadrics in that order (oracle step 1)
+    for (uint32_t v = tid; v < nv; v += 256) {
+      uint32_t lo = aoff[v];
+      uint32_t hi = dg[v];  // cursor ended at one-past-last
+      for (uint32_t i = lo + 1; i < hi; ++i) {
+        uint32_t x = cl[i];
+        uint32_t j = i;
+        while (j > lo && cl[j-1] > x) { cl[j] = cl[j-1]; --j; }
+        cl[j] = x;
+      }
+      float q[10];
+      #pragma unroll
+      for (int k = 0; k < 10; ++k) q[k] = 0.0f;
+      for (uint32_t i = lo; i < hi; ++i) {
+        uint32_t f = cl[i];
+        if (!valid[f]) continue;
+        SimpPlane p = pl[f];
+        sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
+      }
+      #pragma unroll
+      for (int k = 0; k < 10; ++k) Q[10ull*(v0+v) + k] = q[k];
+    }
+    // [6] picks (oracle step 2)
+    for (uint32_t v = tid; v < nv; v += 256) pick[v0 + v] = ~0ull;
+    __syncthreads();
+    for (uint32_t f = tid; f < nt; f += 256) {
+      uint32_t fc[3] = {faces[3*f], faces[3*f+1], faces[3*f+2]};
+      #pragma unroll
+      for (int e = 0; e < 3; ++e) {
+        uint32_t a = fc[e], bb = fc[(e+1)%3];
+        if (a == bb) continue;
+        uint32_t u = a < bb ? a : bb, w = a < bb ? bb : a;
+        float mx = 0.5f*(verts[3ull*u]+verts[3ull*w]);
+        float my = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
+        float mz = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
+        float S[10];
+        #pragma unroll
+        for (int k = 0; k < 10; ++k)
+          S[k] = Q[10ull*u + k] + Q[10ull*w + k];
+        float cost = sq_eval(S, mx, my, mz);
+        if (cost < 0.0f) cost = 0.0f;
+        if (cost > max_cost) continue;
+        uint32_t cb = __float_as_uint(cost);
+        uint32_t ul = u - v0, wl = w - v0;  // label-local ids (oracle)
+        uint32_t hsh = ul ^ (wl * 2654435761u);
+        hsh ^= hsh >> 16; hsh *= 2246822519u; hsh ^= hsh >> 13;
+        cb ^= (hsh & 7u);
+        atomicMin(&pick[u], ((unsigned long long)cb << 32) | w);
+        atomicMin(&pick[w], ((unsigned long long)cb << 32) | u);
+      }
+    }
+    // [7] matched-pair collapse (oracle step 3)
+    if (tid == 0) s_collapses = 0;
+    for (uint32_t v = tid; v < nv; v += 256) remap[v0 + v] = v0 + v;
+    __syncthreads();
+    for (uint32_t v = tid; v < nv; v += 256) {
+      uint32_t u = v0 + v;
+      unsigned long long pu = pick[u];
+      if (pu == ~0ull) continue;
+      uint32_t w = (uint32_t)pu;
+      if (w <= u) continue;
+      unsigned long long pw = pick[w];
+      if (pw == ~0ull || (uint32_t)pw != u) continue;
+      verts[3ull*u]   = 0.5f*(verts[3ull*u]+verts[3ull*w]);
+      verts[3ull*u+1] = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
+      verts[3ull*u+2] = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
+      remap[w] = u;
+      atomicAdd(&s_collapses, 1u);
+    }
+    __syncthreads();
+    if (s_collapses == 0) break;
+    // [8] rewrite + stable compact (oracle step 4); keep flag in valid[]
+    for (uint32_t f = tid; f < nt; f += 256) {
+      uint32_t i0 = remap[faces[3*f]], i1 = remap[faces[3*f+1]],
+               i2 = remap[faces[3*f+2]];
+      faces[3*f] = i0; faces[3*f+1] = i1; faces[3*f+2] = i2;
+      valid[f] = (i0 != i1 && i1 != i2 && i0 != i2) ? 1 : 0;
+    }
+    __syncthreads();
+    // compaction offsets over keep flags (u8 -> widen via deg? use
+    // per-face widen into cols as scratch: cols has 3*nt >= nt slots)
+    for (uint32_t f = tid; f < nt; f += 256) cl[f] = valid[f];
+    __syncthreads();
+    uint32_t kept = blk_exscan(cl, cl + nt, nt, 0, s_sums);
+    for (uint32_t f = tid; f < nt; f += 256) {
+      if (!valid[f]) continue;
+      uint32_t o = cl[nt + f];
+      ftmp[3*o] = faces[3*f]; ftmp[3*o+1] = faces[3*f+1];
+      ftmp[3*o+2] = faces[3*f+2];
+    }
+    __syncthreads();
+    for (uint32_t i = tid; i < 3*kept; i += 256) faces[i] = ftmp[i];
+    if (tid == 0) s_nt = kept;
+    __syncthreads();
+    if (kept == nt) break;  // no progress (oracle: progress == 0)
+  }
+  // park the final faces at the label's original offset
+  for (uint32_t i = tid; i < 3 * s_nt; i += 256)
+    park_faces[3ull * f0 + i] = faces[i];
+  if (tid == 0) {
+    nt_cur[b] = s_nt;
+    active[b] = 0;  // done: global rounds skip this label
+  }
 }
