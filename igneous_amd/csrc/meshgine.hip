@@ -785,13 +785,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   // bigger labels go through the global-rounds machinery below.
   const uint32_t SIMP_BIG_CAP = 65536;
   {
-    // occupancy throttle: a per-label working set is ~250 KB; capping
-    // resident blocks/CU keeps the aggregate inside the 256 MB L3
-    // instead of thrashing to HBM (dynamic-LDS padding, unused)
-    int blk_sz = 128;
-    if (const char *e = getenv("MG_SIMP_BLOCK")) blk_sz = atoi(e);
-    hipLaunchKernelGGL(k_simplify_label, dim3((uint32_t)L), dim3(blk_sz),
-                       0, s,
+    hipLaunchKernelGGL(k_simplify_label, dim3((uint32_t)L), dim3(256), 0, s,
                        faces_g, (uint32_t *)c->simp_faces_alt.ptr,
                        (const uint32_t *)c->tri_off.ptr,
                        (const uint32_t *)c->vbase.ptr,

@@ -408,8 +408,7 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
                                uint32_t n, uint32_t base,
                                uint32_t *s_sums /*257*/) {
   const uint32_t tid = threadIdx.x;
-  const uint32_t nthr = blockDim.x;
-  const uint32_t chunk = (n + nthr - 1) / nthr;
+  const uint32_t chunk = (n + 255) / 256;
   const uint32_t lo = tid * chunk;
   const uint32_t hi = lo + chunk < n ? lo + chunk : n;
   uint32_t sum = 0;
@@ -418,12 +417,12 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
   __syncthreads();
   if (tid == 0) {
     uint32_t acc = 0;
-    for (uint32_t k = 0; k < nthr; ++k) {
+    for (int k = 0; k < 256; ++k) {
       uint32_t t = s_sums[k];
       s_sums[k] = acc;
       acc += t;
     }
-    s_sums[nthr] = acc;
+    s_sums[256] = acc;
   }
   __syncthreads();
   uint32_t run = base + s_sums[tid];
@@ -432,14 +431,14 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
     dst[i] = run;
     run += t;
   }
-  uint32_t total = s_sums[nthr];
+  uint32_t total = s_sums[256];
   __syncthreads();
   return total;
 }
 
 __global__ __launch_bounds__(256) void k_simplify_label(
     uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
-    uint32_t *__restrict__ faces_tmp,     // same slicing (ping-pong)
+    uint32_t *__restrict__ faces_tmp,     // same slicing (scratch)
     const uint32_t *__restrict__ tri_off, // L+1 (original offsets)
     const uint32_t *__restrict__ vbase,   // L+1
     float *__restrict__ verts,
@@ -449,8 +448,8 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     uint32_t *__restrict__ deg,           // per vertex (doubles as cursor)
     uint32_t *__restrict__ adj_off,       // per vertex
     uint32_t *__restrict__ cols,          // 3 per face slot (CSR payload)
-    SimpPlane *__restrict__ fq,           // unused (kept for ABI stability)
-    uint8_t *__restrict__ fvalid,         // unused
+    SimpPlane *__restrict__ fq,           // per face (round scratch)
+    uint8_t *__restrict__ fvalid,         // per face (plane valid / keep)
     uint32_t *__restrict__ nt_cur,
     const uint32_t *__restrict__ target,
     uint8_t *__restrict__ active,
@@ -463,7 +462,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   if (nt0 > big_cap) return;  // global-rounds path handles big labels
   if (!active[b]) {
     // already at/below target: final faces = original faces; park them
-    for (uint32_t i = threadIdx.x; i < 3 * nt0; i += blockDim.x)
+    for (uint32_t i = threadIdx.x; i < 3 * nt0; i += 256)
       park_faces[3ull * f0 + i] = faces_g[3ull * f0 + i];
     return;
   }
@@ -471,9 +470,10 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   const uint32_t nv = vbase[b + 1] - v0;
   const uint32_t tgt = target[b];
   const uint32_t tid = threadIdx.x;
-  const uint32_t nthr = blockDim.x;
-  uint32_t *fbuf[2] = {faces_g + 3ull * f0, faces_tmp + 3ull * f0};
-  int cur = 0;
+  uint32_t *faces = faces_g + 3ull * f0;
+  uint32_t *ftmp = faces_tmp + 3ull * f0;
+  SimpPlane *pl = fq + f0;
+  uint8_t *valid = fvalid + f0;
   uint32_t *dg = deg + v0;
   uint32_t *aoff = adj_off + v0;
   uint32_t *cl = cols + 3ull * f0;
@@ -486,34 +486,48 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   for (int round = 0; round < 65536; ++round) {
     const uint32_t nt = s_nt;
     if (nt <= tgt) break;
-    uint32_t *faces = fbuf[cur];
 
-    // [1] CSR degrees
-    for (uint32_t v = tid; v < nv; v += nthr) dg[v] = 0;
+    // [1] face planes (oracle step 1 preamble)
+    for (uint32_t f = tid; f < nt; f += 256) {
+      uint32_t i0 = faces[3*f], i1 = faces[3*f+1], i2 = faces[3*f+2];
+      const float *p0 = verts + 3ull*i0, *p1 = verts + 3ull*i1,
+                  *p2 = verts + 3ull*i2;
+      float ux = p1[0]-p0[0], uy = p1[1]-p0[1], uz = p1[2]-p0[2];
+      float vx = p2[0]-p0[0], vy = p2[1]-p0[1], vz = p2[2]-p0[2];
+      float nx = uy*vz - uz*vy, ny = uz*vx - ux*vz, nz = ux*vy - uy*vx;
+      float len = sqrtf(nx*nx + ny*ny + nz*nz);
+      if (len <= 0.0f) { valid[f] = 0; continue; }
+      float inv = 1.0f / len;
+      nx *= inv; ny *= inv; nz *= inv;
+      float d = -(nx*p0[0] + ny*p0[1] + nz*p0[2]);
+      pl[f] = SimpPlane{nx, ny, nz, d};
+      valid[f] = 1;
+    }
+    // [2] CSR degrees
+    for (uint32_t v = tid; v < nv; v += 256) dg[v] = 0;
     __syncthreads();
-    for (uint32_t f = tid; f < nt; f += nthr) {
+    for (uint32_t f = tid; f < nt; f += 256) {
       atomicAdd(&dg[faces[3*f] - v0], 1u);
       atomicAdd(&dg[faces[3*f+1] - v0], 1u);
       atomicAdd(&dg[faces[3*f+2] - v0], 1u);
     }
     __syncthreads();
-    // [2] offsets (local, base 0 into cl); then fill (cursor = deg)
+    // [3] offsets (local, base 0 into cl)
     blk_exscan(dg, aoff, nv, 0, s_sums);
-    for (uint32_t v = tid; v < nv; v += nthr) dg[v] = aoff[v];
+    // [4] fill (cursor = deg reused)
+    for (uint32_t v = tid; v < nv; v += 256) dg[v] = aoff[v];
     __syncthreads();
-    for (uint32_t f = tid; f < nt; f += nthr) {
+    for (uint32_t f = tid; f < nt; f += 256) {
       cl[atomicAdd(&dg[faces[3*f] - v0], 1u)] = f;
       cl[atomicAdd(&dg[faces[3*f+1] - v0], 1u)] = f;
       cl[atomicAdd(&dg[faces[3*f+2] - v0], 1u)] = f;
     }
     __syncthreads();
-    // [3] per-vertex: sort incident faces ascending, accumulate quadrics
-    // in that order with the plane recomputed per use (identical f32
-    // expressions as oracle step 1 — recomputation is bit-stable), and
-    // reset this vertex's pick slot (fused; next phase is past a sync)
-    for (uint32_t v = tid; v < nv; v += nthr) {
+    // [5] per-vertex: sort incident faces ascending (insertion sort),
+    // accumulate quadrics in that order (oracle step 1)
+    for (uint32_t v = tid; v < nv; v += 256) {
       uint32_t lo = aoff[v];
-      uint32_t hi = dg[v];  // cursor ended one past last
+      uint32_t hi = dg[v];  // cursor ended at one-past-last
       for (uint32_t i = lo + 1; i < hi; ++i) {
         uint32_t x = cl[i];
         uint32_t j = i;
@@ -525,26 +539,17 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       for (int k = 0; k < 10; ++k) q[k] = 0.0f;
       for (uint32_t i = lo; i < hi; ++i) {
         uint32_t f = cl[i];
-        uint32_t i0 = faces[3*f], i1 = faces[3*f+1], i2 = faces[3*f+2];
-        const float *p0 = verts + 3ull*i0, *p1 = verts + 3ull*i1,
-                    *p2 = verts + 3ull*i2;
-        float ux = p1[0]-p0[0], uy = p1[1]-p0[1], uz = p1[2]-p0[2];
-        float vx = p2[0]-p0[0], vy = p2[1]-p0[1], vz = p2[2]-p0[2];
-        float nx = uy*vz - uz*vy, ny = uz*vx - ux*vz, nz = ux*vy - uy*vx;
-        float len = sqrtf(nx*nx + ny*ny + nz*nz);
-        if (len <= 0.0f) continue;
-        float inv = 1.0f / len;
-        nx *= inv; ny *= inv; nz *= inv;
-        float d = -(nx*p0[0] + ny*p0[1] + nz*p0[2]);
-        sq_add_plane(q, nx, ny, nz, d, 1.0f);
+        if (!valid[f]) continue;
+        SimpPlane p = pl[f];
+        sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
       }
       #pragma unroll
       for (int k = 0; k < 10; ++k) Q[10ull*(v0+v) + k] = q[k];
-      pick[v0 + v] = ~0ull;
     }
+    // [6] picks (oracle step 2)
+    for (uint32_t v = tid; v < nv; v += 256) pick[v0 + v] = ~0ull;
     __syncthreads();
-    // [4] picks (oracle step 2)
-    for (uint32_t f = tid; f < nt; f += nthr) {
+    for (uint32_t f = tid; f < nt; f += 256) {
       uint32_t fc[3] = {faces[3*f], faces[3*f+1], faces[3*f+2]};
       #pragma unroll
       for (int e = 0; e < 3; ++e) {
@@ -570,11 +575,11 @@ __global__ __launch_bounds__(256) void k_simplify_label(
         atomicMin(&pick[w], ((unsigned long long)cb << 32) | u);
       }
     }
-    // [5] matched-pair collapse (oracle step 3)
+    // [7] matched-pair collapse (oracle step 3)
     if (tid == 0) s_collapses = 0;
-    for (uint32_t v = tid; v < nv; v += nthr) remap[v0 + v] = v0 + v;
+    for (uint32_t v = tid; v < nv; v += 256) remap[v0 + v] = v0 + v;
     __syncthreads();
-    for (uint32_t v = tid; v < nv; v += nthr) {
+    for (uint32_t v = tid; v < nv; v += 256) {
       uint32_t u = v0 + v;
       unsigned long long pu = pick[u];
       if (pu == ~0ull) continue;
@@ -590,56 +595,36 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     }
     __syncthreads();
     if (s_collapses == 0) break;
-    // [6] rewrite + stable compact in two thread-chunked passes, writing
-    // the compacted faces into the OTHER buffer (round ping-pong)
-    uint32_t *fout = fbuf[cur ^ 1];
-    {
-      const uint32_t chunk = (nt + nthr - 1) / nthr;
-      const uint32_t lo = tid * chunk;
-      const uint32_t hi = lo + chunk < nt ? lo + chunk : nt;
-      uint32_t cnt = 0;
-      for (uint32_t f = lo; f < hi; ++f) {
-        uint32_t i0 = remap[faces[3*f]], i1 = remap[faces[3*f+1]],
-                 i2 = remap[faces[3*f+2]];
-        faces[3*f] = i0; faces[3*f+1] = i1; faces[3*f+2] = i2;
-        cnt += (i0 != i1 && i1 != i2 && i0 != i2) ? 1u : 0u;
-      }
-      s_sums[tid] = cnt;
-      __syncthreads();
-      if (tid == 0) {
-        uint32_t acc = 0;
-        for (uint32_t k = 0; k < nthr; ++k) {
-          uint32_t t = s_sums[k];
-          s_sums[k] = acc;
-          acc += t;
-        }
-        s_sums[nthr] = acc;
-      }
-      __syncthreads();
-      uint32_t o = s_sums[tid];
-      for (uint32_t f = lo; f < hi; ++f) {
-        uint32_t i0 = faces[3*f], i1 = faces[3*f+1], i2 = faces[3*f+2];
-        if (i0 == i1 || i1 == i2 || i0 == i2) continue;
-        fout[3*o] = i0; fout[3*o+1] = i1; fout[3*o+2] = i2;
-        ++o;
-      }
-      uint32_t kept = s_sums[nthr];
-      __syncthreads();
-      if (tid == 0) s_nt = kept;
-      __syncthreads();
-      cur ^= 1;
-      if (kept == nt) break;  // no progress (oracle: progress == 0)
+    // [8] rewrite + stable compact (oracle step 4); keep flag in valid[]
+    for (uint32_t f = tid; f < nt; f += 256) {
+      uint32_t i0 = remap[faces[3*f]], i1 = remap[faces[3*f+1]],
+               i2 = remap[faces[3*f+2]];
+      faces[3*f] = i0; faces[3*f+1] = i1; faces[3*f+2] = i2;
+      valid[f] = (i0 != i1 && i1 != i2 && i0 != i2) ? 1 : 0;
     }
+    __syncthreads();
+    // compaction offsets over keep flags (u8 -> widen via deg? use
+    // per-face widen into cols as scratch: cols has 3*nt >= nt slots)
+    for (uint32_t f = tid; f < nt; f += 256) cl[f] = valid[f];
+    __syncthreads();
+    uint32_t kept = blk_exscan(cl, cl + nt, nt, 0, s_sums);
+    for (uint32_t f = tid; f < nt; f += 256) {
+      if (!valid[f]) continue;
+      uint32_t o = cl[nt + f];
+      ftmp[3*o] = faces[3*f]; ftmp[3*o+1] = faces[3*f+1];
+      ftmp[3*o+2] = faces[3*f+2];
+    }
+    __syncthreads();
+    for (uint32_t i = tid; i < 3*kept; i += 256) faces[i] = ftmp[i];
+    if (tid == 0) s_nt = kept;
+    __syncthreads();
+    if (kept == nt) break;  // no progress (oracle: progress == 0)
   }
   // park the final faces at the label's original offset
-  {
-    uint32_t *faces = fbuf[cur];
-    for (uint32_t i = tid; i < 3 * s_nt; i += nthr)
-      park_faces[3ull * f0 + i] = faces[i];
-  }
+  for (uint32_t i = tid; i < 3 * s_nt; i += 256)
+    park_faces[3ull * f0 + i] = faces[i];
   if (tid == 0) {
     nt_cur[b] = s_nt;
     active[b] = 0;  // done: global rounds skip this label
   }
 }
-
