@@ -414,66 +414,6 @@ struct FirstOccur {
   }
 };
 
-// scan-output proxy: stores the exclusive-scan value (the corner's vertex
-// id if it is a first occurrence) into vtx_scan AND, for first
-// occurrences, immediately assigns the vertex: wvtx[slot] = id and the
-// decoded position -> verts. Fuses the former k_weld_verts pass into the
-// scan's single sweep (each output is written exactly once).
-struct WeldScanCtx {
-  uint32_t *vtx_scan;
-  const uint4 *recs;
-  const uint32_t *wminp;
-  uint32_t *wvtx;
-  float *verts;
-  uint32_t usx, usxy;
-  float rx, ry, rz, shift;
-};
-
-struct WeldScanRef {
-  const WeldScanCtx *c;
-  uint32_t i;
-  __device__ WeldScanRef &operator=(uint32_t v) {
-    c->vtx_scan[i] = v;
-    uint4 rec = c->recs[i / 3];
-    uint32_t slot = (i % 3 == 0) ? rec.x : ((i % 3 == 1) ? rec.y : rec.z);
-    if (c->wminp[slot] == ~i) {
-      c->wvtx[slot] = v;
-      uint32_t eslot = slot >> 1;
-      uint32_t axis = eslot % 3u;
-      uint32_t lin = eslot / 3u;
-      uint32_t vz = lin / c->usxy;
-      uint32_t rem = lin - vz * c->usxy;
-      uint32_t vy = rem / c->usx;
-      uint32_t vx = rem - vy * c->usx;
-      float dx = (float)(2 * vx + (axis == 0));
-      float dy = (float)(2 * vy + (axis == 1));
-      float dz = (float)(2 * vz + (axis == 2));
-      c->verts[3ull * v + 0] = (0.5f * dx + c->shift) * c->rx;
-      c->verts[3ull * v + 1] = (0.5f * dy + c->shift) * c->ry;
-      c->verts[3ull * v + 2] = (0.5f * dz + c->shift) * c->rz;
-    }
-    return *this;
-  }
-};
-
-struct WeldScanOut {
-  using value_type = uint32_t;
-  using reference = WeldScanRef;
-  using pointer = WeldScanRef *;
-  using difference_type = int64_t;
-  using iterator_category = std::random_access_iterator_tag;
-  const WeldScanCtx *c;
-  uint32_t i;
-  __device__ WeldScanRef operator*() const { return WeldScanRef{c, i}; }
-  __device__ WeldScanRef operator[](int64_t k) const {
-    return WeldScanRef{c, (uint32_t)(i + k)};
-  }
-  __host__ __device__ WeldScanOut operator+(int64_t k) const {
-    return WeldScanOut{c, (uint32_t)(i + k)};
-  }
-  __host__ __device__ WeldScanOut &operator++() { ++i; return *this; }
-};
-
 // total verts = vtx_scan[NC-1] + first_occur(NC-1)
 __global__ void k_total_verts(const uint32_t *__restrict__ vtx_scan,
                               const uint4 *__restrict__ recs,
@@ -1401,34 +1341,18 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     uint64_t nbt = (T + blk - 1) / blk;
     hipLaunchKernelGGL(k_weld_insert, dim3((uint32_t)nbt), dim3(blk), 0, s,
                        recs_sorted, wminp, T);
-    // scan first-occurrence flags -> vertex ids; the scan's output
-    // proxy also assigns first-occurrence vertices (id + position) in
-    // the same sweep. Needs verts storage BEFORE the total is known:
-    // upper bound = NC corners (real count read back right after).
-    if (ensure(c, c->verts, NC * 12)) return 20;
-    const float shift0 = voxel_centered ? 0.0f : 0.5f;
-    // device-side ctx for the output proxy (POD copied by value)
-    WeldScanCtx wctx{(uint32_t *)c->vtx_scan.ptr, recs_sorted, wminp, wvtx,
-                     (float *)c->verts.ptr, (uint32_t)g.sx,
-                     (uint32_t)(g.sx * g.sy), rx, ry, rz, shift0};
-    WeldScanOut wout{nullptr, 0};
-    // the ctx must live in memory visible to the device: keep it in a
-    // small device buffer
-    if (ensure(c, c->small, sizeof(WeldScanCtx))) return 20;
-    HIP_TRY(c, hipMemcpyAsync(c->small.ptr, &wctx, sizeof(wctx),
-                              hipMemcpyHostToDevice, s), 20);
-    wout.c = (const WeldScanCtx *)c->small.ptr;
+    // scan first-occurrence flags -> vertex ids (flags computed on the fly)
     FirstOccur fo{recs_sorted, wminp};
     auto it = rocprim::make_transform_iterator(
         rocprim::counting_iterator<uint32_t>(0), fo);
     size_t tmp_bytes = 0;
     hipError_t e = rocprim::exclusive_scan(
-        nullptr, tmp_bytes, it, wout, 0u, NC,
+        nullptr, tmp_bytes, it, (uint32_t *)c->vtx_scan.ptr, 0u, NC,
         rocprim::plus<uint32_t>(), s);
     if (e != hipSuccess) { SET_ERR(c, "weld scan size query failed"); return 20; }
     if (ensure(c, c->scan_tmp, tmp_bytes)) return 20;
     e = rocprim::exclusive_scan(
-        c->scan_tmp.ptr, tmp_bytes, it, wout, 0u, NC,
+        c->scan_tmp.ptr, tmp_bytes, it, (uint32_t *)c->vtx_scan.ptr, 0u, NC,
         rocprim::plus<uint32_t>(), s);
     if (e != hipSuccess) { SET_ERR(c, "weld scan failed"); return 20; }
     hipLaunchKernelGGL(k_total_verts, dim3(1), dim3(1), 0, s,
@@ -1443,11 +1367,19 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   }
   c->stats.total_verts = total_verts;
 
+  if (ensure(c, c->verts, total_verts * 12)) return 22;
   if (ensure(c, c->faces, NC * 4)) return 22;
   if (ensure(c, c->vbase, ((uint64_t)nlabels + 1) * 4)) return 22;
   {
     int blk = 256;
     uint64_t nbt = (T + blk - 1) / blk;
+    const float shift = voxel_centered ? 0.0f : 0.5f;
+    hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nbt), dim3(blk), 0, s,
+                       recs_sorted,
+                       (const uint32_t *)c->vtx_scan.ptr, wminp, wvtx,
+                       (float *)c->verts.ptr,
+                       (uint32_t)g.sx, (uint32_t)(g.sx * g.sy),
+                       rx, ry, rz, shift, T);
     uint32_t nbl = (nlabels + 1 + 255) / 256;
     hipLaunchKernelGGL(k_vbase, dim3(nbl), dim3(256), 0, s,
                        (const uint32_t *)c->tri_off.ptr,
