@@ -158,9 +158,12 @@ def main():
         chunks = [voronoi_labels((256, 256, 256), 6250, 1000 + i,
                                  dtype=np.uint64) for i in range(nchunks)]
         pool = ThreadPoolExecutor(max_workers=args.streams)
+        thread_engines = []
 
         def mesh_one(chunk):
             e = engine_mod.Engine.get(local_rank)
+            if e not in thread_engines:
+                thread_engines.append(e)
             # results discarded before the thread's next call: zero-copy
             return e.mesh_chunk(chunk, resolution=RESOLUTION,
                                 reduction_factor=red, max_error=40.0,
@@ -201,7 +204,10 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    stats = eng.stats()
+    if args.mode == "chunks256" and thread_engines:
+        stats = thread_engines[0].stats()  # per-thread ctxs did the work
+    else:
+        stats = eng.stats()
     total_vox = step_vox * args.steps * world
     mvox_s = total_vox / elapsed / 1e6
 

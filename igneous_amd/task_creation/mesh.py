@@ -10,7 +10,7 @@ from __future__ import annotations
 from time import strftime
 from typing import Optional
 
-from ..lib import Bbox, Vec
+from ..lib import Vec
 from ..storage import CloudFiles
 from ..tasks.mesh import MeshTask
 from ..volume import PrecomputedVolume

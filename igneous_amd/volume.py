@@ -13,8 +13,6 @@ Written from scratch; cloud-volume is not a dependency of this package.
 """
 from __future__ import annotations
 
-import json
-import os
 from typing import Optional
 
 import numpy as np
