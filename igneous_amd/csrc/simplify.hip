@@ -415,17 +415,18 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
   for (uint32_t i = lo; i < hi; ++i) sum += src[i];
   s_sums[tid] = sum;
   __syncthreads();
-  if (tid == 0) {
-    uint32_t acc = 0;
-    for (int k = 0; k < 256; ++k) {
-      uint32_t t = s_sums[k];
-      s_sums[k] = acc;
-      acc += t;
-    }
-    s_sums[256] = acc;
+  // Hillis-Steele inclusive scan over the 256 partials (log steps)
+  #pragma unroll
+  for (uint32_t st = 1; st < 256; st <<= 1) {
+    uint32_t x = (tid >= st) ? s_sums[tid - st] : 0;
+    __syncthreads();
+    s_sums[tid] += x;
+    __syncthreads();
   }
+  uint32_t excl = s_sums[tid] - sum;      // exclusive from inclusive
+  if (tid == 255) s_sums[256] = s_sums[255];
   __syncthreads();
-  uint32_t run = base + s_sums[tid];
+  uint32_t run = base + excl;
   for (uint32_t i = lo; i < hi; ++i) {
     uint32_t t = src[i];
     dst[i] = run;
@@ -528,20 +529,59 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     for (uint32_t v = tid; v < nv; v += 256) {
       uint32_t lo = aoff[v];
       uint32_t hi = dg[v];  // cursor ended at one-past-last
-      for (uint32_t i = lo + 1; i < hi; ++i) {
-        uint32_t x = cl[i];
-        uint32_t j = i;
-        while (j > lo && cl[j-1] > x) { cl[j] = cl[j-1]; --j; }
-        cl[j] = x;
-      }
+      uint32_t d = hi - lo;
       float q[10];
       #pragma unroll
       for (int k = 0; k < 10; ++k) q[k] = 0.0f;
-      for (uint32_t i = lo; i < hi; ++i) {
-        uint32_t f = cl[i];
-        if (!valid[f]) continue;
-        SimpPlane p = pl[f];
-        sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
+      if (d <= 16) {
+        // register path: one batched load of the face list, bitonic
+        // sort network (compile-time indices, no global RMW chains).
+        // Sorted order is unique (face ids distinct), so this matches
+        // the oracle's insertion sort exactly.
+        uint32_t fl[16];
+        #pragma unroll
+        for (int k = 0; k < 16; ++k)
+          fl[k] = (k < (int)d) ? cl[lo + k] : 0xFFFFFFFFu;
+        #pragma unroll
+        for (int ksz = 2; ksz <= 16; ksz <<= 1) {
+          #pragma unroll
+          for (int j = ksz >> 1; j > 0; j >>= 1) {
+            #pragma unroll
+            for (int i = 0; i < 16; ++i) {
+              int l = i ^ j;
+              if (l > i) {
+                bool up = ((i & ksz) == 0);
+                uint32_t a = fl[i], b2 = fl[l];
+                bool sw = up ? (a > b2) : (a < b2);
+                fl[i] = sw ? b2 : a;
+                fl[l] = sw ? a : b2;
+              }
+            }
+          }
+        }
+        #pragma unroll
+        for (int k = 0; k < 16; ++k) {
+          if (k < (int)d) {
+            uint32_t f = fl[k];
+            if (valid[f]) {
+              SimpPlane p = pl[f];
+              sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
+            }
+          }
+        }
+      } else {
+        for (uint32_t i = lo + 1; i < hi; ++i) {
+          uint32_t x = cl[i];
+          uint32_t j = i;
+          while (j > lo && cl[j-1] > x) { cl[j] = cl[j-1]; --j; }
+          cl[j] = x;
+        }
+        for (uint32_t i = lo; i < hi; ++i) {
+          uint32_t f = cl[i];
+          if (!valid[f]) continue;
+          SimpPlane p = pl[f];
+          sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
+        }
       }
       #pragma unroll
       for (int k = 0; k < 10; ++k) Q[10ull*(v0+v) + k] = q[k];
