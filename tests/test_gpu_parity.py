@@ -299,3 +299,26 @@ def test_dispatch_streams_gpu(tmp_path):
     assert sorted(names_a) == sorted(names_b) and names_a
     for n in names_a:
         assert cf_a.get(n) == cf_b.get(n), f"{n} differs across stream modes"
+
+
+def test_default_task_shape_449(eng):
+    """The reference's default task shape is 448^3 (+1vx high padding ->
+    449^3 input, task_creation/mesh.py:161). Property checks at that
+    shape: valid meshes, half-grid vertices, determinism."""
+    from igneous_amd.synth import voronoi_labels
+    data = np.zeros((449, 449, 449), dtype=np.uint64, order="F")
+    inner = voronoi_labels((448, 448, 448), 33000, 42, dtype=np.uint64)
+    data[:448, :448, :448] = inner
+    res = (16.0, 16.0, 40.0)
+    got = eng.mesh_chunk(data, resolution=res)
+    assert len(got) > 30000
+    labs = sorted(got)[:: max(1, len(got) // 10)]
+    for lab in labs:
+        v, f = got[lab]
+        assert f.max() < v.shape[0]
+        k = v / (0.5 * np.array(res, dtype=np.float32))
+        assert np.allclose(k, np.round(k), atol=0)
+    again = eng.mesh_chunk(data, resolution=res)
+    for lab in labs:
+        assert np.array_equal(got[lab][0], again[lab][0])
+        assert np.array_equal(got[lab][1], again[lab][1])
