@@ -361,10 +361,11 @@ __global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
 // window (the label-partitioned stream is spatially coherent, so ~half
 // the corners repeat a slot seen moments earlier) — cutting the global
 // atomic count roughly in half.
-#define WI_LDS_SLOTS 2048
-__global__ __launch_bounds__(256) void k_weld_insert(
+template <int BLK>
+__global__ __launch_bounds__(BLK) void k_weld_insert(
     const uint4 *__restrict__ recs_sorted, uint32_t *__restrict__ wminp,
     uint64_t ntris) {
+  constexpr int WI_LDS_SLOTS = BLK * 8;  // ~0.375 load at 3 corners/thread
   __shared__ uint32_t lkey[WI_LDS_SLOTS];
   __shared__ uint32_t lval[WI_LDS_SLOTS];
   for (int k = threadIdx.x; k < WI_LDS_SLOTS; k += blockDim.x) {
@@ -381,7 +382,8 @@ __global__ __launch_bounds__(256) void k_weld_insert(
       uint32_t slot = s[v];
       uint32_t enc = ~(uint32_t)(3 * t + v);
       uint32_t key = slot + 1;
-      uint32_t h = (slot * 2654435761u) >> (32 - 11);
+      constexpr int LG = (BLK == 256 ? 11 : (BLK == 512 ? 12 : 13));
+      uint32_t h = (slot * 2654435761u) >> (32 - LG);
       bool placed = false;
       for (int probe = 0; probe < 32; ++probe) {
         uint32_t cur = lkey[h];
@@ -1339,8 +1341,21 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   {
     int blk = 256;
     uint64_t nbt = (T + blk - 1) / blk;
-    hipLaunchKernelGGL(k_weld_insert, dim3((uint32_t)nbt), dim3(blk), 0, s,
-                       recs_sorted, wminp, T);
+    int wi_blk = 1024;
+    if (const char *e = getenv("MG_WELD_INSERT_BLK")) wi_blk = atoi(e);
+    if (wi_blk == 256) {
+      uint64_t nb2 = (T + 255) / 256;
+      hipLaunchKernelGGL(k_weld_insert<256>, dim3((uint32_t)nb2), dim3(256),
+                         0, s, recs_sorted, wminp, T);
+    } else if (wi_blk == 512) {
+      uint64_t nb2 = (T + 511) / 512;
+      hipLaunchKernelGGL(k_weld_insert<512>, dim3((uint32_t)nb2), dim3(512),
+                         0, s, recs_sorted, wminp, T);
+    } else {
+      uint64_t nb2 = (T + 1023) / 1024;
+      hipLaunchKernelGGL(k_weld_insert<1024>, dim3((uint32_t)nb2),
+                         dim3(1024), 0, s, recs_sorted, wminp, T);
+    }
     // scan first-occurrence flags -> vertex ids (flags computed on the fly)
     FirstOccur fo{recs_sorted, wminp};
     auto it = rocprim::make_transform_iterator(
