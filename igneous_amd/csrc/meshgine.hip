@@ -361,20 +361,27 @@ __global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
 // window (the label-partitioned stream is spatially coherent, so ~half
 // the corners repeat a slot seen moments earlier) — cutting the global
 // atomic count roughly in half.
-template <int BLK>
+template <int BLK, int TPT>
 __global__ __launch_bounds__(BLK) void k_weld_insert(
     const uint4 *__restrict__ recs_sorted, uint32_t *__restrict__ wminp,
     uint64_t ntris) {
-  constexpr int WI_LDS_SLOTS = BLK * 8;  // ~0.375 load at 3 corners/thread
+  // LDS table sized for ~BLK*TPT*3 corners at ~0.4 load; wider windows
+  // dedup more of a vertex's ~6 corner occurrences before the global
+  // atomics (the label-sorted stream is spatially coherent)
+  constexpr int WI_LDS_SLOTS = BLK * TPT * 4;
+  constexpr int LG = __builtin_ctz(WI_LDS_SLOTS);
   __shared__ uint32_t lkey[WI_LDS_SLOTS];
   __shared__ uint32_t lval[WI_LDS_SLOTS];
-  for (int k = threadIdx.x; k < WI_LDS_SLOTS; k += blockDim.x) {
+  for (int k = threadIdx.x; k < WI_LDS_SLOTS; k += BLK) {
     lkey[k] = 0;
     lval[k] = 0;
   }
   __syncthreads();
-  uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (t < ntris) {
+  const uint64_t span0 = (uint64_t)blockIdx.x * BLK * TPT;
+  #pragma unroll
+  for (int rep = 0; rep < TPT; ++rep) {
+    uint64_t t = span0 + (uint64_t)rep * BLK + threadIdx.x;
+    if (t >= ntris) break;
     uint4 rec = recs_sorted[t];
     const uint32_t s[3] = {rec.x, rec.y, rec.z};
     #pragma unroll
@@ -382,7 +389,6 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
       uint32_t slot = s[v];
       uint32_t enc = ~(uint32_t)(3 * t + v);
       uint32_t key = slot + 1;
-      constexpr int LG = (BLK == 256 ? 11 : (BLK == 512 ? 12 : 13));
       uint32_t h = (slot * 2654435761u) >> (32 - LG);
       bool placed = false;
       for (int probe = 0; probe < 32; ++probe) {
@@ -399,7 +405,7 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
     }
   }
   __syncthreads();
-  for (int k = threadIdx.x; k < WI_LDS_SLOTS; k += blockDim.x) {
+  for (int k = threadIdx.x; k < WI_LDS_SLOTS; k += BLK) {
     uint32_t key = lkey[k];
     if (key) atomicMax(&wminp[key - 1], lval[k]);
   }
@@ -1341,19 +1347,19 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   {
     int blk = 256;
     uint64_t nbt = (T + blk - 1) / blk;
-    int wi_blk = 1024;
-    if (const char *e = getenv("MG_WELD_INSERT_BLK")) wi_blk = atoi(e);
-    if (wi_blk == 256) {
-      uint64_t nb2 = (T + 255) / 256;
-      hipLaunchKernelGGL(k_weld_insert<256>, dim3((uint32_t)nb2), dim3(256),
-                         0, s, recs_sorted, wminp, T);
-    } else if (wi_blk == 512) {
-      uint64_t nb2 = (T + 511) / 512;
-      hipLaunchKernelGGL(k_weld_insert<512>, dim3((uint32_t)nb2), dim3(512),
-                         0, s, recs_sorted, wminp, T);
-    } else {
+    int wi_cfg = 2;  // 0: 1024x1, 1: 1024x2, 2: 1024x4
+    if (const char *e = getenv("MG_WELD_INSERT_CFG")) wi_cfg = atoi(e);
+    if (wi_cfg == 0) {
       uint64_t nb2 = (T + 1023) / 1024;
-      hipLaunchKernelGGL(k_weld_insert<1024>, dim3((uint32_t)nb2),
+      hipLaunchKernelGGL((k_weld_insert<1024, 1>), dim3((uint32_t)nb2),
+                         dim3(1024), 0, s, recs_sorted, wminp, T);
+    } else if (wi_cfg == 1) {
+      uint64_t nb2 = (T + 2047) / 2048;
+      hipLaunchKernelGGL((k_weld_insert<1024, 2>), dim3((uint32_t)nb2),
+                         dim3(1024), 0, s, recs_sorted, wminp, T);
+    } else {
+      uint64_t nb2 = (T + 4095) / 4096;
+      hipLaunchKernelGGL((k_weld_insert<1024, 4>), dim3((uint32_t)nb2),
                          dim3(1024), 0, s, recs_sorted, wminp, T);
     }
     // scan first-occurrence flags -> vertex ids (flags computed on the fly)
