@@ -411,32 +411,55 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
   }
 }
 
-// scan-input functor: 1 where this corner is its key's first occurrence
-struct FirstOccur {
-  const uint4 *recs;
-  const uint32_t *wminp;
-  __device__ uint32_t operator()(uint32_t i) const {
-    uint4 rec = recs[i / 3];
+// [5b] first-occurrence flags as a bit array: each wave handles 64
+// consecutive corners and ballots the flags into one u64 word. Vertex
+// ids then come from a word-granular scan + popcount instead of a full
+// per-element scan array.
+__global__ void k_weld_flag_bits(const uint4 *__restrict__ recs_sorted,
+                                 const uint32_t *__restrict__ wminp,
+                                 unsigned long long *__restrict__ bits,
+                                 uint64_t ncorners, uint64_t nwords) {
+  uint64_t gid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint64_t word = gid / 64;
+  if (word >= nwords) return;
+  uint32_t lane = (uint32_t)(gid & 63);
+  uint64_t i = word * 64 + lane;
+  bool flag = false;
+  if (i < ncorners) {
+    uint4 rec = recs_sorted[i / 3];
     uint32_t slot = (i % 3 == 0) ? rec.x : ((i % 3 == 1) ? rec.y : rec.z);
-    return (wminp[slot] == ~i) ? 1u : 0u;
+    flag = (wminp[slot] == ~(uint32_t)i);
+  }
+  unsigned long long m = __ballot(flag);
+  if (lane == 0) bits[word] = m;
+}
+
+struct PopcWord {
+  __device__ uint32_t operator()(unsigned long long w) const {
+    return (uint32_t)__popcll(w);
   }
 };
 
-// total verts = vtx_scan[NC-1] + first_occur(NC-1)
-__global__ void k_total_verts(const uint32_t *__restrict__ vtx_scan,
-                              const uint4 *__restrict__ recs,
-                              const uint32_t *__restrict__ wminp,
-                              uint64_t ncorners, uint32_t *out) {
-  uint32_t i = (uint32_t)(ncorners - 1);
-  uint4 rec = recs[i / 3];
-  uint32_t slot = (i % 3 == 0) ? rec.x : ((i % 3 == 1) ? rec.y : rec.z);
-  *out = vtx_scan[i] + ((wminp[slot] == ~i) ? 1u : 0u);
+// vertex id of corner i (must be below its word's scan base + rank)
+__device__ __forceinline__ uint32_t vtx_id_of(
+    const uint32_t *__restrict__ wscan,
+    const unsigned long long *__restrict__ bits, uint64_t i) {
+  uint64_t w = i >> 6;
+  unsigned long long mask = (1ull << (i & 63)) - 1;
+  return wscan[w] + (uint32_t)__popcll(bits[w] & mask);
+}
+
+__global__ void k_total_verts(const uint32_t *__restrict__ wscan,
+                              const unsigned long long *__restrict__ bits,
+                              uint64_t nwords, uint32_t *out) {
+  *out = wscan[nwords - 1] + (uint32_t)__popcll(bits[nwords - 1]);
 }
 
 // [5d] first occurrences: record vertex id in the table, write the vertex
 // (doubled coordinates decoded from the slot: voxel, axis, +1 on axis)
 __global__ void k_weld_verts(const uint4 *__restrict__ recs_sorted,
-                             const uint32_t *__restrict__ vtx_scan,
+                             const uint32_t *__restrict__ wscan,
+                             const unsigned long long *__restrict__ bits,
                              const uint32_t *__restrict__ wminp,
                              uint32_t *__restrict__ wvtx,
                              float *__restrict__ verts,
@@ -452,7 +475,7 @@ __global__ void k_weld_verts(const uint4 *__restrict__ recs_sorted,
     uint32_t i = (uint32_t)(3 * t + v);
     uint32_t slot = s[v];
     if (wminp[slot] != ~i) continue;
-    uint32_t vid = vtx_scan[i];
+    uint32_t vid = vtx_id_of(wscan, bits, i);
     wvtx[slot] = vid;
     uint32_t eslot = slot >> 1;
     uint32_t axis = eslot % 3u;
@@ -472,13 +495,14 @@ __global__ void k_weld_verts(const uint4 *__restrict__ recs_sorted,
 
 // [5e] per-label vertex bases: vbase[l] = vtx_scan at the label's first corner
 __global__ void k_vbase(const uint32_t *__restrict__ tri_off,
-                        const uint32_t *__restrict__ vtx_scan,
+                        const uint32_t *__restrict__ wscan,
+                        const unsigned long long *__restrict__ bits,
                         uint32_t *__restrict__ vbase,
                         uint32_t nlabels, uint64_t total_verts) {
   uint32_t l = blockIdx.x * blockDim.x + threadIdx.x;
   if (l > nlabels) return;
   if (l == nlabels) vbase[l] = (uint32_t)total_verts;
-  else vbase[l] = vtx_scan[3ull * tri_off[l]];
+  else vbase[l] = vtx_id_of(wscan, bits, 3ull * tri_off[l]);
 }
 
 // [5f] faces: per-label local vertex indices (the record carries its
@@ -1347,7 +1371,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   {
     int blk = 256;
     uint64_t nbt = (T + blk - 1) / blk;
-    int wi_cfg = 2;  // 0: 1024x1, 1: 1024x2, 2: 1024x4
+    int wi_cfg = 1;  // 0: 1024x1, 1: 1024x2, 2: 1024x4
     if (const char *e = getenv("MG_WELD_INSERT_CFG")) wi_cfg = atoi(e);
     if (wi_cfg == 0) {
       uint64_t nb2 = (T + 1023) / 1024;
@@ -1362,23 +1386,30 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
       hipLaunchKernelGGL((k_weld_insert<1024, 4>), dim3((uint32_t)nb2),
                          dim3(1024), 0, s, recs_sorted, wminp, T);
     }
-    // scan first-occurrence flags -> vertex ids (flags computed on the fly)
-    FirstOccur fo{recs_sorted, wminp};
-    auto it = rocprim::make_transform_iterator(
-        rocprim::counting_iterator<uint32_t>(0), fo);
+    // first-occurrence flags as a bit array + word-granular scan
+    const uint64_t nwords = (NC + 63) / 64;
+    unsigned long long *bits =
+        (unsigned long long *)c->vtx_scan.ptr;          // nwords * 8
+    uint32_t *wscan = (uint32_t *)c->vtx_scan.ptr + 2 * nwords;  // nwords * 4
+    // (vtx_scan buffer is NC*4 bytes >= nwords*12)
+    {
+      uint64_t nbw = (nwords * 64 + blk - 1) / blk;
+      hipLaunchKernelGGL(k_weld_flag_bits, dim3((uint32_t)nbw), dim3(blk),
+                         0, s, recs_sorted, wminp, bits, NC, nwords);
+    }
+    auto it = rocprim::make_transform_iterator(bits, PopcWord{});
     size_t tmp_bytes = 0;
     hipError_t e = rocprim::exclusive_scan(
-        nullptr, tmp_bytes, it, (uint32_t *)c->vtx_scan.ptr, 0u, NC,
+        nullptr, tmp_bytes, it, wscan, 0u, nwords,
         rocprim::plus<uint32_t>(), s);
     if (e != hipSuccess) { SET_ERR(c, "weld scan size query failed"); return 20; }
     if (ensure(c, c->scan_tmp, tmp_bytes)) return 20;
     e = rocprim::exclusive_scan(
-        c->scan_tmp.ptr, tmp_bytes, it, (uint32_t *)c->vtx_scan.ptr, 0u, NC,
+        c->scan_tmp.ptr, tmp_bytes, it, wscan, 0u, nwords,
         rocprim::plus<uint32_t>(), s);
     if (e != hipSuccess) { SET_ERR(c, "weld scan failed"); return 20; }
     hipLaunchKernelGGL(k_total_verts, dim3(1), dim3(1), 0, s,
-                       (const uint32_t *)c->vtx_scan.ptr,
-                       recs_sorted, wminp, NC,
+                       wscan, bits, nwords,
                        (uint32_t *)c->lh_misc.ptr + 3);
     uint32_t tv = 0;
     HIP_TRY(c, hipMemcpyAsync(&tv, (uint32_t *)c->lh_misc.ptr + 3, 4,
@@ -1395,16 +1426,19 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     int blk = 256;
     uint64_t nbt = (T + blk - 1) / blk;
     const float shift = voxel_centered ? 0.0f : 0.5f;
+    const uint64_t nwords = (NC + 63) / 64;
+    const unsigned long long *bits =
+        (const unsigned long long *)c->vtx_scan.ptr;
+    const uint32_t *wscan = (const uint32_t *)c->vtx_scan.ptr + 2 * nwords;
     hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nbt), dim3(blk), 0, s,
-                       recs_sorted,
-                       (const uint32_t *)c->vtx_scan.ptr, wminp, wvtx,
+                       recs_sorted, wscan, bits, wminp, wvtx,
                        (float *)c->verts.ptr,
                        (uint32_t)g.sx, (uint32_t)(g.sx * g.sy),
                        rx, ry, rz, shift, T);
     uint32_t nbl = (nlabels + 1 + 255) / 256;
     hipLaunchKernelGGL(k_vbase, dim3(nbl), dim3(256), 0, s,
                        (const uint32_t *)c->tri_off.ptr,
-                       (const uint32_t *)c->vtx_scan.ptr,
+                       wscan, bits,
                        (uint32_t *)c->vbase.ptr, nlabels, total_verts);
     hipLaunchKernelGGL(k_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
                        recs_sorted, wvtx,
