@@ -16,15 +16,17 @@
 //   [1] k_count        wave-per-64-cell-segment triangle counts
 //                      (+ label hash build)                    -> segcnt
 //   [2] scan           rocPRIM exclusive scan                  -> segoff, T
-//   [3] k_emit         recompute counts, wave prefix, write
-//                      (label_idx, 3x weld keys) per triangle in canonical
-//                      global order
-//   [4] partition      rocPRIM radix-sort by label_idx (stable) + gather
-//   [5] weld           hash insert w/ atomicMin(first-pos), flag first
-//                      occurrences, scan -> vertex ids, emit vertices,
-//                      rewrite faces as per-label indices
-//   [6] (optional) per-label quadric simplification
-//   [7] D2H slices + meshset assembly (host pinned memory)
+//   [3] k_emit         recompute counts, wave prefix, write one 16-B
+//                      record (3 weld slots + label id) per triangle in
+//                      canonical global order (case tables in LDS)
+//   [4] partition      rocPRIM stable radix-sort by label id + 16-B gather
+//   [5] weld           direct-addressed (edge,side) table: atomicMax(~pos)
+//                      first-seen positions (LDS pre-dedup), bit-packed
+//                      first-occurrence flags + word-granular scan ->
+//                      vertex ids, vertex emit, per-label face indices
+//   [6] (optional) per-label quadric simplification (one workgroup per
+//                      label; global matched-pair rounds for huge labels)
+//   [7] D2H slices + meshset assembly (ctx-owned pinned staging)
 //
 // Determinism: every kernel's output is a pure function of its inputs —
 // atomics are only used for first-position minima (order-free), hash slot
@@ -564,8 +566,8 @@ struct mg_ctx {
       lh_keys, lh_vals, lh_misc,
       tri_label, tri_label_alt, order, order_alt, tri_keys, keys_sorted,
       tri_off, sort_tmp,
-      wh_keys, wh_minp, wh_vtx, flags, vtx_scan, verts, faces, vbase,
-      label_values, small,
+      wh_keys, wh_vtx, vtx_scan, verts, faces, vbase,
+      label_values,
       simp_fq, simp_valid, simp_pk, simp_pk_alt, simp_pv, simp_pv_alt,
       simp_Q, simp_pick, simp_remap, simp_flab, simp_flab_alt,
       simp_faces_alt, simp_verts_alt, simp_vbase_alt, simp_meta, simp_ref,
@@ -668,9 +670,8 @@ void mg_destroy(mg_ctx *c) {
                   &c->lh_keys, &c->lh_vals, &c->lh_misc, &c->tri_label,
                   &c->tri_label_alt, &c->order, &c->order_alt, &c->tri_keys,
                   &c->keys_sorted, &c->tri_off, &c->sort_tmp, &c->wh_keys,
-                  &c->wh_minp, &c->wh_vtx, &c->flags, &c->vtx_scan,
+                  &c->wh_vtx, &c->vtx_scan,
                   &c->verts, &c->faces, &c->vbase, &c->label_values,
-                  &c->small,
                   &c->simp_fq, &c->simp_valid, &c->simp_pk, &c->simp_pk_alt,
                   &c->simp_pv, &c->simp_pv_alt, &c->simp_Q, &c->simp_pick,
                   &c->simp_remap, &c->simp_flab, &c->simp_flab_alt,

@@ -30,11 +30,14 @@
  * Vertices returned are CHUNK-LOCAL nm coordinates (float32), exactly as
  * zmesh returns them to MeshTask before the Python-side offset is applied.
  *
- * Threading: one mg_ctx per device; calls on distinct ctxs may run
- * concurrently. Calls on one ctx serialize internally. No global state
- * besides the HIP runtime. Errors: non-zero int return + mg_last_error.
- * Ownership: the engine allocates all output storage; the caller frees a
- * result with mg_meshset_free.
+ * Threading: one mg_ctx per device (or several per device for
+ * overlapping streams); calls on distinct ctxs may run concurrently.
+ * Calls on one ctx serialize internally. No global state besides the HIP
+ * runtime. Errors: non-zero int return + mg_last_error.
+ * Ownership: the meshset's flat vertex/face storage is CTX-OWNED pinned
+ * staging, reused by the ctx's next call (see mg_meshset below);
+ * mg_meshset_free frees the descriptor. Consume or copy results before
+ * the next mg_mesh_chunk on the same ctx.
  */
 #ifndef MESHGINE_H
 #define MESHGINE_H
