@@ -322,3 +322,50 @@ def test_default_task_shape_449(eng):
     for lab in labs:
         assert np.array_equal(got[lab][0], again[lab][0])
         assert np.array_equal(got[lab][1], again[lab][1])
+
+
+def test_label_hash_growth_260k_labels(eng):
+    """~260k distinct labels in a 64^3 chunk forces the label-hash
+    grow-and-retry path; still bit-exact vs the oracle."""
+    import oracle
+    rng = np.random.default_rng(99)
+    data = rng.integers(1, 1 << 40, size=(64, 64, 64)).astype(np.uint64)
+    data = np.asfortranarray(data)
+    res = (16.0, 16.0, 40.0)
+    got = eng.mesh_chunk(data, resolution=res)
+    want = oracle.mesh_chunk(data, resolution=res)
+    assert len(got) == len(want) and len(got) > 200_000
+    # full comparison is heavy at 260k labels; compare counts everywhere
+    # and geometry on a sample
+    for lab in want:
+        assert got[lab][0].shape == want[lab][0].shape
+        assert got[lab][1].shape == want[lab][1].shape
+    for lab in list(want)[:: max(1, len(want) // 500)]:
+        assert np.array_equal(got[lab][0], want[lab][0])
+        assert np.array_equal(got[lab][1], want[lab][1])
+
+
+def test_near_cap_dims(eng):
+    """1023x1023x255 (the reference's own 32-bit mesher bound shape class,
+    cli.py:1049-1052): indexing stays exact near the 32-bit weld-table
+    capacity."""
+    data = np.zeros((1023, 1023, 255), dtype=np.uint64, order="F")
+    data[100:900, 100:900, 50:200] = 7
+    data[400:600, 400:600, 100:150] = 9
+    got = eng.mesh_chunk(data, resolution=(16.0, 16.0, 40.0))
+    assert set(got) == {7, 9}
+    v7, f7 = got[7]
+    # closed box minus the label-9 hole boundary: validity + bbox
+    assert f7.max() < v7.shape[0]
+    assert np.allclose(v7.min(axis=0), [99.5 * 16, 99.5 * 16, 49.5 * 40])
+    assert np.allclose(v7.max(axis=0), [899.5 * 16, 899.5 * 16, 199.5 * 40])
+    v9, _ = got[9]
+    assert np.allclose(v9.min(axis=0), [399.5 * 16, 399.5 * 16, 99.5 * 40])
+
+
+def test_oversize_chunk_errors_cleanly(eng):
+    """Dims beyond the weld-table capacity raise with the documented
+    message instead of corrupting (2048^3 > 2047 per-axis cap)."""
+    bad = np.zeros((2048, 4, 4), dtype=np.uint32, order="F")
+    with pytest.raises(RuntimeError, match="dims out of range"):
+        eng.mesh_chunk(bad)
