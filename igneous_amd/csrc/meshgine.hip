@@ -145,47 +145,6 @@ __device__ __forceinline__ void load_corners(const T *__restrict__ labels,
   c[6] = p[sxy + sx];     c[7] = p[sxy + sx + 1];
 }
 
-// wave-cooperative corner load for a 64-cell x-segment: lane i's +x
-// corners are lane i+1's base corners — 4 loads per lane plus 4 boundary
-// loads on the last lane, the +x values delivered by __shfl_down. Every
-// lane with a valid VOXEL (vx < sx) loads, so shuffles are valid for
-// every lane with a valid CELL.
-template <typename T>
-__device__ __forceinline__ bool wave_load_corners(
-    const T *__restrict__ labels, int64_t sx, int64_t sxy,
-    int64_t cx0, int64_t cy, int64_t cz, int lane, int64_t ncx, T c[8]) {
-  const int64_t vx = cx0 + lane;
-  const T *p = labels + vx + cy * sx + cz * sxy;
-  T a0 = 0, a2 = 0, a4 = 0, a6 = 0;
-  if (vx < sx) {
-    a0 = p[0];
-    a2 = p[sx];
-    a4 = p[sxy];
-    a6 = p[sxy + sx];
-  }
-  T b1, b3, b5, b7;
-  if (sizeof(T) == 8) {
-    b1 = (T)__shfl_down((unsigned long long)a0, 1, WAVE);
-    b3 = (T)__shfl_down((unsigned long long)a2, 1, WAVE);
-    b5 = (T)__shfl_down((unsigned long long)a4, 1, WAVE);
-    b7 = (T)__shfl_down((unsigned long long)a6, 1, WAVE);
-  } else {
-    b1 = (T)__shfl_down((unsigned int)a0, 1, WAVE);
-    b3 = (T)__shfl_down((unsigned int)a2, 1, WAVE);
-    b5 = (T)__shfl_down((unsigned int)a4, 1, WAVE);
-    b7 = (T)__shfl_down((unsigned int)a6, 1, WAVE);
-  }
-  if (lane == WAVE - 1 && vx + 1 < sx) {
-    b1 = p[1];
-    b3 = p[sx + 1];
-    b5 = p[sxy + 1];
-    b7 = p[sxy + sx + 1];
-  }
-  c[0] = a0; c[1] = b1; c[2] = a2; c[3] = b3;
-  c[4] = a4; c[5] = b5; c[6] = a6; c[7] = b7;
-  return vx < ncx;  // this lane's CELL is valid
-}
-
 template <typename T>
 __device__ __forceinline__ uint32_t cell_tri_count(const T c[8],
                                                    const uint8_t *cnt_tab) {
@@ -236,10 +195,9 @@ __global__ void k_count(const T *__restrict__ labels, GridDims g,
     const int64_t cz = row / g.ncy;
     const int64_t cx = segx * WAVE + lane;
     uint32_t cnt = 0;
-    T c[8];
-    bool cell_ok = wave_load_corners(labels, g.sx, sxy, segx * WAVE, cy, cz,
-                                     lane, g.ncx, c);
-    if (cell_ok) {
+    if (cx < g.ncx) {
+      T c[8];
+      load_corners(labels, g.sx, sxy, cx, cy, cz, c);
       if (!(c[0] == c[1] && c[0] == c[2] && c[0] == c[3] && c[0] == c[4] &&
             c[0] == c[5] && c[0] == c[6] && c[0] == c[7])) {
         #pragma unroll
@@ -314,8 +272,8 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
     T c[8];
     uint32_t cnt = 0;
     bool active = false;
-    if (wave_load_corners(labels, g.sx, sxy, segx * WAVE, cy, cz, lane,
-                          g.ncx, c)) {
+    if (cx < g.ncx) {
+      load_corners(labels, g.sx, sxy, cx, cy, cz, c);
       cnt = cell_tri_count(c, s_cnt);
       active = cnt > 0;
     }
