@@ -475,12 +475,22 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   uint32_t *ftmp = faces_tmp + 3ull * f0;
   SimpPlane *pl = fq + f0;
   uint8_t *valid = fvalid + f0;
-  uint32_t *dg = deg + v0;
   uint32_t *aoff = adj_off + v0;
   uint32_t *cl = cols + 3ull * f0;
 
   __shared__ uint32_t s_sums[257];
   __shared__ uint32_t s_nt, s_collapses;
+  // the two ATOMIC-hot per-vertex arrays live in LDS when the label fits:
+  // neighboring faces' vertices share cache lines, so the global
+  // atomicAdd/atomicMin streams serialize on hot L2 lines — LDS atomics
+  // are bank-parallel. (~25 KB -> ~6 blocks/CU.)
+  constexpr uint32_t CAPV = 2048;
+  __shared__ uint32_t s_deg[CAPV];
+  __shared__ unsigned long long s_pick[CAPV];
+  const bool lds_mode = (nv <= CAPV);
+  uint32_t *dg = lds_mode ? s_deg : (deg + v0);
+  unsigned long long *pick_l =
+      lds_mode ? s_pick : (pick + v0);
   if (tid == 0) s_nt = nt0;
   __syncthreads();
 
@@ -587,7 +597,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       for (int k = 0; k < 10; ++k) Q[10ull*(v0+v) + k] = q[k];
     }
     // [6] picks (oracle step 2)
-    for (uint32_t v = tid; v < nv; v += 256) pick[v0 + v] = ~0ull;
+    for (uint32_t v = tid; v < nv; v += 256) pick_l[v] = ~0ull;
     __syncthreads();
     for (uint32_t f = tid; f < nt; f += 256) {
       uint32_t fc[3] = {faces[3*f], faces[3*f+1], faces[3*f+2]};
@@ -611,8 +621,8 @@ __global__ __launch_bounds__(256) void k_simplify_label(
         uint32_t hsh = ul ^ (wl * 2654435761u);
         hsh ^= hsh >> 16; hsh *= 2246822519u; hsh ^= hsh >> 13;
         cb ^= (hsh & 7u);
-        atomicMin(&pick[u], ((unsigned long long)cb << 32) | w);
-        atomicMin(&pick[w], ((unsigned long long)cb << 32) | u);
+        atomicMin(&pick_l[ul], ((unsigned long long)cb << 32) | w);
+        atomicMin(&pick_l[wl], ((unsigned long long)cb << 32) | u);
       }
     }
     // [7] matched-pair collapse (oracle step 3)
@@ -621,11 +631,11 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     __syncthreads();
     for (uint32_t v = tid; v < nv; v += 256) {
       uint32_t u = v0 + v;
-      unsigned long long pu = pick[u];
+      unsigned long long pu = pick_l[v];
       if (pu == ~0ull) continue;
       uint32_t w = (uint32_t)pu;
       if (w <= u) continue;
-      unsigned long long pw = pick[w];
+      unsigned long long pw = pick_l[w - v0];
       if (pw == ~0ull || (uint32_t)pw != u) continue;
       verts[3ull*u]   = 0.5f*(verts[3ull*u]+verts[3ull*w]);
       verts[3ull*u+1] = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
