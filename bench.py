@@ -127,10 +127,14 @@ def main():
     if world > 1:
         import torch.distributed as tdist
         dist = tdist
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # nccl (RCCL) wants one DISTINCT device per rank; fall back to
+        # gloo when ranks outnumber devices (single-GPU smoke of the
+        # distributed path)
+        ndev = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        backend = "nccl" if 0 < world <= ndev else "gloo"
         dist.init_process_group(backend=backend)
-        if torch.cuda.is_available():
-            torch.cuda.set_device(local_rank)
+        if ndev:
+            torch.cuda.set_device(local_rank % ndev)
 
     from igneous_amd.engine import Engine
     from igneous_amd.synth import voronoi_labels
@@ -144,7 +148,9 @@ def main():
     data = voronoi_labels(SHAPE, K_SEEDS, SEED, dtype=np.uint64)
     if dist is not None and rank == 0:
         dist.barrier()
-    eng = Engine.get(local_rank)
+    from igneous_amd.engine import load_library
+    ndev = max(1, load_library().mg_device_count())
+    eng = Engine.get(local_rank % ndev)
 
     red = 100 if args.simplify else 0
 
@@ -161,7 +167,7 @@ def main():
         thread_engines = []
 
         def mesh_one(chunk):
-            e = engine_mod.Engine.get(local_rank)
+            e = engine_mod.Engine.get(local_rank % ndev)
             if e not in thread_engines:
                 thread_engines.append(e)
             # results discarded before the thread's next call: zero-copy
