@@ -56,7 +56,13 @@ __global__ void k_face_planes(const uint32_t *__restrict__ faces_g,
   float vx = p2[0]-p0[0], vy = p2[1]-p0[1], vz = p2[2]-p0[2];
   float nx = uy*vz - uz*vy, ny = uz*vx - ux*vz, nz = ux*vy - uy*vx;
   float len = sqrtf(nx*nx + ny*ny + nz*nz);
-  if (len <= 0.0f) { fvalid[t] = 0; return; }
+  if (len <= 0.0f) {
+    // zero plane: accumulating it adds exact +0 products, bitwise
+    // identical to the oracle's skip (quadric sums are never -0)
+    fq[t] = SimpPlane{0.0f, 0.0f, 0.0f, 0.0f};
+    fvalid[t] = 0;
+    return;
+  }
   float inv = 1.0f / len;
   nx *= inv; ny *= inv; nz *= inv;
   float d = -(nx*p0[0] + ny*p0[1] + nz*p0[2]);
@@ -507,12 +513,14 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       float vx = p2[0]-p0[0], vy = p2[1]-p0[1], vz = p2[2]-p0[2];
       float nx = uy*vz - uz*vy, ny = uz*vx - ux*vz, nz = ux*vy - uy*vx;
       float len = sqrtf(nx*nx + ny*ny + nz*nz);
-      if (len <= 0.0f) { valid[f] = 0; continue; }
+      if (len <= 0.0f) {
+        pl[f] = SimpPlane{0.0f, 0.0f, 0.0f, 0.0f};  // +0 products = skip
+        continue;
+      }
       float inv = 1.0f / len;
       nx *= inv; ny *= inv; nz *= inv;
       float d = -(nx*p0[0] + ny*p0[1] + nz*p0[2]);
       pl[f] = SimpPlane{nx, ny, nz, d};
-      valid[f] = 1;
     }
     // [2] CSR degrees
     for (uint32_t v = tid; v < nv; v += 256) dg[v] = 0;
@@ -569,16 +577,16 @@ __global__ __launch_bounds__(256) void k_simplify_label(
             }
           }
         }
+        // gather all planes first (independent loads), then the ordered
+        // f32 accumulation from registers
+        SimpPlane ps[16];
         #pragma unroll
-        for (int k = 0; k < 16; ++k) {
-          if (k < (int)d) {
-            uint32_t f = fl[k];
-            if (valid[f]) {
-              SimpPlane p = pl[f];
-              sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
-            }
-          }
-        }
+        for (int k = 0; k < 16; ++k)
+          ps[k] = pl[k < (int)d ? fl[k] : fl[0]];
+        #pragma unroll
+        for (int k = 0; k < 16; ++k)
+          if (k < (int)d)
+            sq_add_plane(q, ps[k].nx, ps[k].ny, ps[k].nz, ps[k].d, 1.0f);
       } else {
         for (uint32_t i = lo + 1; i < hi; ++i) {
           uint32_t x = cl[i];
@@ -587,9 +595,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
           cl[j] = x;
         }
         for (uint32_t i = lo; i < hi; ++i) {
-          uint32_t f = cl[i];
-          if (!valid[f]) continue;
-          SimpPlane p = pl[f];
+          SimpPlane p = pl[cl[i]];
           sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
         }
       }
