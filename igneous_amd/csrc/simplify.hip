@@ -582,7 +582,9 @@ __global__ __launch_bounds__(256) void k_simplify_label(
         SimpPlane ps[16];
         #pragma unroll
         for (int k = 0; k < 16; ++k)
-          ps[k] = pl[k < (int)d ? fl[k] : fl[0]];
+          // pad with face 0 (nt >= 1 inside the loop); fl[k] is the
+          // 0xFFFFFFFF sort sentinel beyond d and must not be indexed
+          ps[k] = pl[k < (int)d ? fl[k] : 0u];
         #pragma unroll
         for (int k = 0; k < 16; ++k)
           if (k < (int)d)
