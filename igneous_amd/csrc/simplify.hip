@@ -603,10 +603,10 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       }
       #pragma unroll
       for (int k = 0; k < 10; ++k) Q[10ull*(v0+v) + k] = q[k];
+      pick_l[v] = ~0ull;  // fused pick reset (same-thread slot)
     }
-    // [6] picks (oracle step 2)
-    for (uint32_t v = tid; v < nv; v += 256) pick_l[v] = ~0ull;
     __syncthreads();
+    // [6] picks (oracle step 2)
     for (uint32_t f = tid; f < nt; f += 256) {
       uint32_t fc[3] = {faces[3*f], faces[3*f+1], faces[3*f+2]};
       #pragma unroll
