@@ -369,3 +369,17 @@ def test_oversize_chunk_errors_cleanly(eng):
     bad = np.zeros((2048, 4, 4), dtype=np.uint32, order="F")
     with pytest.raises(RuntimeError, match="dims out of range"):
         eng.mesh_chunk(bad)
+
+
+def test_u32_at_scale(eng):
+    """512^3 uint32 (the 2x-algorithmic-rate dtype): structural validity +
+    agreement with the u64 run of the same labels (ids < 2^28)."""
+    from igneous_amd.synth import voronoi_labels
+    d32 = voronoi_labels((512, 512, 512), 50000, 303, dtype=np.uint32)
+    res = (16.0, 16.0, 40.0)
+    got32 = eng.mesh_chunk(d32, resolution=res)
+    got64 = eng.mesh_chunk(d32.astype(np.uint64), resolution=res)
+    assert sorted(got32) == sorted(got64)
+    for lab in list(sorted(got32))[:: max(1, len(got32) // 100)]:
+        assert np.array_equal(got32[lab][0], got64[lab][0])
+        assert np.array_equal(got32[lab][1], got64[lab][1])
