@@ -33,12 +33,24 @@ class RegisteredTask:
         self._args = _jsonable(record)
 
     def payload(self) -> str:
-        return json.dumps({"class": type(self).__name__, "args": self._args})
+        return json.dumps({
+            "class": type(self).__name__,
+            "module": type(self).__module__,
+            "args": self._args,
+        })
 
     @staticmethod
     def deserialize(payload: str) -> "RegisteredTask":
         d = json.loads(payload)
-        cls = RegisteredTask._registry[d["class"]]
+        cls = RegisteredTask._registry.get(d["class"])
+        if cls is None and d.get("module"):
+            # worker processes (spawn) have an empty registry: import the
+            # defining module, which registers the class on import
+            import importlib
+            mod = importlib.import_module(d["module"])
+            cls = getattr(mod, d["class"], None) or                 RegisteredTask._registry.get(d["class"])
+        if cls is None:
+            raise KeyError(f"unknown task class {d['class']}")
         return cls(**d["args"])
 
     def execute(self):
