@@ -365,7 +365,10 @@ __global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
 // atomic count roughly in half.
 template <int BLK, int TPT>
 __global__ __launch_bounds__(BLK) void k_weld_insert(
-    const uint4 *__restrict__ recs_sorted, uint32_t *__restrict__ wminp,
+    const uint4 *__restrict__ tri_recs,   // emit order
+    const uint32_t *__restrict__ order,   // label partition permutation
+    uint4 *__restrict__ recs_sorted,      // gathered here (fused pass)
+    uint32_t *__restrict__ wminp,
     uint64_t ntris) {
   // LDS table sized for ~BLK*TPT*3 corners at ~0.4 load; wider windows
   // dedup more of a vertex's ~6 corner occurrences before the global
@@ -384,7 +387,8 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
   for (int rep = 0; rep < TPT; ++rep) {
     uint64_t t = span0 + (uint64_t)rep * BLK + threadIdx.x;
     if (t >= ntris) break;
-    uint4 rec = recs_sorted[t];
+    uint4 rec = tri_recs[order[t]];
+    recs_sorted[t] = rec;
     const uint32_t s[3] = {rec.x, rec.y, rec.z};
     #pragma unroll
     for (int v = 0; v < 3; ++v) {
@@ -1319,6 +1323,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   if (ensure(c, c->order_alt, T * 4)) return 19;
   if (ensure(c, c->tri_label_alt, T * 4)) return 19;
   if (ensure(c, c->keys_sorted, T * 16)) return 19;
+  uint32_t *order_sorted = nullptr;
   {
     int blk = 256;
     uint64_t nb = (T + blk - 1) / blk;
@@ -1340,10 +1345,8 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     e = rocprim::radix_sort_pairs(
         c->sort_tmp.ptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
     if (e != hipSuccess) { SET_ERR(c, "radix_sort failed"); return 19; }
-    // gather
-    hipLaunchKernelGGL(k_gather_recs, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (const uint4 *)c->tri_keys.ptr, d_vals.current(),
-                       (uint4 *)c->keys_sorted.ptr, T);
+    // (the 16-B record gather is fused into k_weld_insert below)
+    order_sorted = d_vals.current();
     // label ranges
     if (ensure(c, c->tri_off, ((uint64_t)nlabels + 1) * 4)) return 19;
     hipLaunchKernelGGL(k_label_ranges, dim3((uint32_t)nb), dim3(blk), 0, s,
@@ -1374,18 +1377,23 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     uint64_t nbt = (T + blk - 1) / blk;
     int wi_cfg = 1;  // 0: 1024x1, 1: 1024x2, 2: 1024x4
     if (const char *e = getenv("MG_WELD_INSERT_CFG")) wi_cfg = atoi(e);
+    uint4 *rs_mut = (uint4 *)c->keys_sorted.ptr;
+    const uint4 *tr = (const uint4 *)c->tri_keys.ptr;
     if (wi_cfg == 0) {
       uint64_t nb2 = (T + 1023) / 1024;
       hipLaunchKernelGGL((k_weld_insert<1024, 1>), dim3((uint32_t)nb2),
-                         dim3(1024), 0, s, recs_sorted, wminp, T);
+                         dim3(1024), 0, s, tr, order_sorted, rs_mut, wminp,
+                         T);
     } else if (wi_cfg == 1) {
       uint64_t nb2 = (T + 2047) / 2048;
       hipLaunchKernelGGL((k_weld_insert<1024, 2>), dim3((uint32_t)nb2),
-                         dim3(1024), 0, s, recs_sorted, wminp, T);
+                         dim3(1024), 0, s, tr, order_sorted, rs_mut, wminp,
+                         T);
     } else {
       uint64_t nb2 = (T + 4095) / 4096;
       hipLaunchKernelGGL((k_weld_insert<1024, 4>), dim3((uint32_t)nb2),
-                         dim3(1024), 0, s, recs_sorted, wminp, T);
+                         dim3(1024), 0, s, tr, order_sorted, rs_mut, wminp,
+                         T);
     }
     // first-occurrence flags as a bit array + word-granular scan
     const uint64_t nwords = (NC + 63) / 64;
