@@ -412,7 +412,8 @@ __global__ void k_gather_final(const uint32_t *__restrict__ park_faces,
 // block-wide exclusive scan of src[0..n) into dst (+base), returns total
 __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
                                uint32_t n, uint32_t base,
-                               uint32_t *s_sums /*257*/) {
+                               uint32_t *s_sums /*257*/,
+                               uint32_t *dst2 = nullptr) {
   const uint32_t tid = threadIdx.x;
   const uint32_t chunk = (n + 255) / 256;
   const uint32_t lo = tid * chunk;
@@ -436,7 +437,8 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
   for (uint32_t i = lo; i < hi; ++i) {
     uint32_t t = src[i];
     dst[i] = run;
-    run += t;
+    if (dst2) dst2[i] = run;  // optional cursor copy (src may alias
+    run += t;                 // dst2: t was read first)
   }
   uint32_t total = s_sums[256];
   __syncthreads();
@@ -504,9 +506,14 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     const uint32_t nt = s_nt;
     if (nt <= tgt) break;
 
-    // [1] face planes (oracle step 1 preamble)
+    // [1+2] face planes fused with the CSR degree count (one face pass)
+    for (uint32_t v = tid; v < nv; v += 256) dg[v] = 0;
+    __syncthreads();
     for (uint32_t f = tid; f < nt; f += 256) {
       uint32_t i0 = faces[3*f], i1 = faces[3*f+1], i2 = faces[3*f+2];
+      atomicAdd(&dg[i0 - v0], 1u);
+      atomicAdd(&dg[i1 - v0], 1u);
+      atomicAdd(&dg[i2 - v0], 1u);
       const float *p0 = verts + 3ull*i0, *p1 = verts + 3ull*i1,
                   *p2 = verts + 3ull*i2;
       float ux = p1[0]-p0[0], uy = p1[1]-p0[1], uz = p1[2]-p0[2];
@@ -522,20 +529,9 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       float d = -(nx*p0[0] + ny*p0[1] + nz*p0[2]);
       pl[f] = SimpPlane{nx, ny, nz, d};
     }
-    // [2] CSR degrees
-    for (uint32_t v = tid; v < nv; v += 256) dg[v] = 0;
     __syncthreads();
-    for (uint32_t f = tid; f < nt; f += 256) {
-      atomicAdd(&dg[faces[3*f] - v0], 1u);
-      atomicAdd(&dg[faces[3*f+1] - v0], 1u);
-      atomicAdd(&dg[faces[3*f+2] - v0], 1u);
-    }
-    __syncthreads();
-    // [3] offsets (local, base 0 into cl)
-    blk_exscan(dg, aoff, nv, 0, s_sums);
-    // [4] fill (cursor = deg reused)
-    for (uint32_t v = tid; v < nv; v += 256) dg[v] = aoff[v];
-    __syncthreads();
+    // [3] offsets; the scan also writes the fill cursors (dst2 = dg)
+    blk_exscan(dg, aoff, nv, 0, s_sums, dg);
     for (uint32_t f = tid; f < nt; f += 256) {
       cl[atomicAdd(&dg[faces[3*f] - v0], 1u)] = f;
       cl[atomicAdd(&dg[faces[3*f+1] - v0], 1u)] = f;
