@@ -821,6 +821,9 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   // workgroup each (cache-resident round loop, no global sorts); only
   // bigger labels go through the global-rounds machinery below.
   const uint32_t SIMP_BIG_CAP = 65536;
+  if (getenv("MG_SIMP_PROF"))
+    HIP_TRY(c, hipMemsetAsync((unsigned long long *)c->lh_misc.ptr + 8, 0,
+                              48, s), 40);
   {
     hipLaunchKernelGGL(k_simplify_label, dim3((uint32_t)L), dim3(256), 0, s,
                        faces_g, (uint32_t *)c->simp_faces_alt.ptr,
@@ -836,8 +839,23 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                        (uint8_t *)c->simp_valid.ptr,
                        nt_cur, target, active,
                        (uint32_t *)c->simp_park.ptr,
+                       getenv("MG_SIMP_PROF")
+                           ? (unsigned long long *)c->lh_misc.ptr + 8
+                           : nullptr,
                        max_cost, (uint32_t)L, SIMP_BIG_CAP);
     HIP_TRY(c, hipGetLastError(), 40);
+    if (getenv("MG_SIMP_PROF")) {
+      unsigned long long hp[6];
+      HIP_TRY(c, hipMemcpyAsync(hp, (unsigned long long *)c->lh_misc.ptr + 8,
+                                48, hipMemcpyDeviceToHost, s), 40);
+      HIP_TRY(c, hipStreamSynchronize(s), 40);
+      const char *nm[6] = {"loophead", "planes+deg", "scan+fill",
+                           "sortaccum", "picks", "collapse+compact"};
+      fprintf(stderr, "[mg simp phases, summed tid0 cycles]");
+      for (int k = 0; k < 6; ++k)
+        fprintf(stderr, " %s=%llu", nm[k], hp[k]);
+      fprintf(stderr, "\n");
+    }
   }
   // drop the per-label-kernel labels from the working set; park big
   // never-active labels (their final faces are their originals)

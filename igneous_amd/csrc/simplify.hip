@@ -463,6 +463,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     const uint32_t *__restrict__ target,
     uint8_t *__restrict__ active,
     uint32_t *__restrict__ park_faces,
+    unsigned long long *__restrict__ prof,  // 6 phase counters or null
     float max_cost, uint32_t nlabels, uint32_t big_cap) {
   const uint32_t b = blockIdx.x;
   if (b >= nlabels) return;
@@ -502,9 +503,20 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   if (tid == 0) s_nt = nt0;
   __syncthreads();
 
+  // env-gated phase profiling (thread 0 wall cycles per phase)
+  unsigned long long ph[6] = {0, 0, 0, 0, 0, 0};
+  unsigned long long t_last = prof ? __builtin_amdgcn_s_memtime() : 0;
+#define PHASE_MARK(k)                                          \
+  if (prof && tid == 0) {                                      \
+    unsigned long long now = __builtin_amdgcn_s_memtime();     \
+    ph[k] += now - t_last;                                     \
+    t_last = now;                                              \
+  }
+
   for (int round = 0; round < 65536; ++round) {
     const uint32_t nt = s_nt;
     if (nt <= tgt) break;
+    PHASE_MARK(0)  // loop head
 
     // [1+2] face planes fused with the CSR degree count (one face pass)
     for (uint32_t v = tid; v < nv; v += 256) dg[v] = 0;
@@ -530,6 +542,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       pl[f] = SimpPlane{nx, ny, nz, d};
     }
     __syncthreads();
+    PHASE_MARK(1)  // planes + degree count
     // [3] offsets; the scan also writes the fill cursors (dst2 = dg)
     blk_exscan(dg, aoff, nv, 0, s_sums, dg);
     for (uint32_t f = tid; f < nt; f += 256) {
@@ -538,6 +551,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       cl[atomicAdd(&dg[faces[3*f+2] - v0], 1u)] = f;
     }
     __syncthreads();
+    PHASE_MARK(2)  // offsets scan + CSR fill
     // [5] per-vertex: sort incident faces ascending (insertion sort),
     // accumulate quadrics in that order (oracle step 1)
     for (uint32_t v = tid; v < nv; v += 256) {
@@ -602,6 +616,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       pick_l[v] = ~0ull;  // fused pick reset (same-thread slot)
     }
     __syncthreads();
+    PHASE_MARK(3)  // sort + quadric accumulate
     // [6] picks (oracle step 2)
     for (uint32_t f = tid; f < nt; f += 256) {
       uint32_t fc[3] = {faces[3*f], faces[3*f+1], faces[3*f+2]};
@@ -629,6 +644,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
         atomicMin(&pick_l[wl], ((unsigned long long)cb << 32) | u);
       }
     }
+    PHASE_MARK(4)  // edge picks
     // [7] matched-pair collapse (oracle step 3)
     if (tid == 0) s_collapses = 0;
     for (uint32_t v = tid; v < nv; v += 256) remap[v0 + v] = v0 + v;
@@ -672,7 +688,14 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     for (uint32_t i = tid; i < 3*kept; i += 256) faces[i] = ftmp[i];
     if (tid == 0) s_nt = kept;
     __syncthreads();
+    PHASE_MARK(5)  // collapse + rewrite + compact
     if (kept == nt) break;  // no progress (oracle: progress == 0)
+  }
+#undef PHASE_MARK
+  if (prof && tid == 0) {
+    #pragma unroll
+    for (int k = 0; k < 6; ++k)
+      if (ph[k]) atomicAdd(&prof[k], ph[k]);
   }
   // park the final faces at the label's original offset
   for (uint32_t i = tid; i < 3 * s_nt; i += 256)
