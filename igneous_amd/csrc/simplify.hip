@@ -420,22 +420,18 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
   const uint32_t hi = lo + chunk < n ? lo + chunk : n;
   uint32_t sum = 0;
   for (uint32_t i = lo; i < hi; ++i) sum += src[i];
-  // two-barrier scan: wave-level shfl inclusive scan, then cross-wave
-  // offsets through 4 LDS words (16 barriers -> 2)
-  uint32_t incl = sum;
-  #pragma unroll
-  for (uint32_t st = 1; st < 64; st <<= 1) {
-    uint32_t y = __shfl_up(incl, st, 64);
-    if ((tid & 63) >= st) incl += y;
-  }
-  if ((tid & 63) == 63) s_sums[tid >> 6] = incl;
+  s_sums[tid] = sum;
   __syncthreads();
-  uint32_t wave_base = 0;
+  // Hillis-Steele inclusive scan over the 256 partials (log steps)
   #pragma unroll
-  for (uint32_t w = 0; w < 4; ++w)
-    if (w < (tid >> 6)) wave_base += s_sums[w];
-  uint32_t total = s_sums[0] + s_sums[1] + s_sums[2] + s_sums[3];
-  uint32_t excl = wave_base + incl - sum;
+  for (uint32_t st = 1; st < 256; st <<= 1) {
+    uint32_t x = (tid >= st) ? s_sums[tid - st] : 0;
+    __syncthreads();
+    s_sums[tid] += x;
+    __syncthreads();
+  }
+  uint32_t excl = s_sums[tid] - sum;      // exclusive from inclusive
+  if (tid == 255) s_sums[256] = s_sums[255];
   __syncthreads();
   uint32_t run = base + excl;
   for (uint32_t i = lo; i < hi; ++i) {
@@ -444,6 +440,7 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
     if (dst2) dst2[i] = run;  // optional cursor copy (src may alias
     run += t;                 // dst2: t was read first)
   }
+  uint32_t total = s_sums[256];
   __syncthreads();
   return total;
 }
