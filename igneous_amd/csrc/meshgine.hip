@@ -825,7 +825,10 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     HIP_TRY(c, hipMemsetAsync((unsigned long long *)c->lh_misc.ptr + 8, 0,
                               48, s), 40);
   {
-    hipLaunchKernelGGL(k_simplify_label, dim3((uint32_t)L), dim3(256), 0, s,
+    const char *clenv = getenv("MG_SIMP_CLLDS");  // "0" disables
+    auto *ksl = (clenv && clenv[0] == '0') ? k_simplify_label<false>
+                                           : k_simplify_label<true>;
+    hipLaunchKernelGGL(ksl, dim3((uint32_t)L), dim3(256), 0, s,
                        faces_g, (uint32_t *)c->simp_faces_alt.ptr,
                        (const uint32_t *)c->tri_off.ptr,
                        (const uint32_t *)c->vbase.ptr,
