@@ -64,10 +64,16 @@ def test_two_rank_gloo_dispatch(tmp_path):
                MESHGINE_REPO=REPO,
                MESHGINE_LAYER=layer,
                MASTER_ADDR="127.0.0.1")
+    # grab a free rendezvous port (a fixed one flakes when a straggler
+    # from an earlier run still holds it)
+    import socket
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        port = sk.getsockname()[1]
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run",
          "--nnodes=1", "--nproc-per-node", "2",
-         "--master-addr", "127.0.0.1", "--master-port", "29517",
+         "--master-addr", "127.0.0.1", "--master-port", str(port),
          str(script)],
         env=env, capture_output=True, text=True, timeout=300)
     assert out.returncode == 0, out.stderr[-2000:]
