@@ -506,6 +506,10 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   uint32_t *dg = lds_mode ? s_deg : (deg + v0);
   unsigned long long *pick_l =
       lds_mode ? s_pick : (pick + v0);
+  // ping-pong face buffers: rewrite reads fa, compaction scatters into
+  // fb, then the buffers swap — no copy-back pass. Parking at the end
+  // reads whichever buffer is current.
+  uint32_t *fa = faces, *fb = ftmp;
   if (tid == 0) s_nt = nt0;
   __syncthreads();
 
@@ -529,7 +533,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     for (uint32_t v = tid; v < nv; v += 256) dg[v] = 0;
     __syncthreads();
     for (uint32_t f = tid; f < nt; f += 256) {
-      uint32_t i0 = faces[3*f], i1 = faces[3*f+1], i2 = faces[3*f+2];
+      uint32_t i0 = fa[3*f], i1 = fa[3*f+1], i2 = fa[3*f+2];
       atomicAdd(&dg[i0 - v0], 1u);
       atomicAdd(&dg[i1 - v0], 1u);
       atomicAdd(&dg[i2 - v0], 1u);
@@ -554,15 +558,15 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     blk_exscan(dg, aoff, nv, 0, s_sums, dg);
     if (clmode) {
       for (uint32_t f = tid; f < nt; f += 256) {
-        s_cl[atomicAdd(&dg[faces[3*f] - v0], 1u)] = (uint16_t)f;
-        s_cl[atomicAdd(&dg[faces[3*f+1] - v0], 1u)] = (uint16_t)f;
-        s_cl[atomicAdd(&dg[faces[3*f+2] - v0], 1u)] = (uint16_t)f;
+        s_cl[atomicAdd(&dg[fa[3*f] - v0], 1u)] = (uint16_t)f;
+        s_cl[atomicAdd(&dg[fa[3*f+1] - v0], 1u)] = (uint16_t)f;
+        s_cl[atomicAdd(&dg[fa[3*f+2] - v0], 1u)] = (uint16_t)f;
       }
     } else {
       for (uint32_t f = tid; f < nt; f += 256) {
-        cl[atomicAdd(&dg[faces[3*f] - v0], 1u)] = f;
-        cl[atomicAdd(&dg[faces[3*f+1] - v0], 1u)] = f;
-        cl[atomicAdd(&dg[faces[3*f+2] - v0], 1u)] = f;
+        cl[atomicAdd(&dg[fa[3*f] - v0], 1u)] = f;
+        cl[atomicAdd(&dg[fa[3*f+1] - v0], 1u)] = f;
+        cl[atomicAdd(&dg[fa[3*f+2] - v0], 1u)] = f;
       }
     }
     __syncthreads();
@@ -647,7 +651,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     PHASE_MARK(3)  // sort + quadric accumulate
     // [6] picks (oracle step 2)
     for (uint32_t f = tid; f < nt; f += 256) {
-      uint32_t fc[3] = {faces[3*f], faces[3*f+1], faces[3*f+2]};
+      uint32_t fc[3] = {fa[3*f], fa[3*f+1], fa[3*f+2]};
       #pragma unroll
       for (int e = 0; e < 3; ++e) {
         uint32_t a = fc[e], bb = fc[(e+1)%3];
@@ -694,26 +698,24 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     __syncthreads();
     if (s_collapses == 0) break;
     // [8] rewrite + stable compact (oracle step 4); keep flag in valid[]
+    // and widened into cl[] in the same pass
     for (uint32_t f = tid; f < nt; f += 256) {
-      uint32_t i0 = remap[faces[3*f]], i1 = remap[faces[3*f+1]],
-               i2 = remap[faces[3*f+2]];
-      faces[3*f] = i0; faces[3*f+1] = i1; faces[3*f+2] = i2;
-      valid[f] = (i0 != i1 && i1 != i2 && i0 != i2) ? 1 : 0;
+      uint32_t i0 = remap[fa[3*f]], i1 = remap[fa[3*f+1]],
+               i2 = remap[fa[3*f+2]];
+      fa[3*f] = i0; fa[3*f+1] = i1; fa[3*f+2] = i2;
+      uint32_t keep = (i0 != i1 && i1 != i2 && i0 != i2) ? 1u : 0u;
+      valid[f] = (uint8_t)keep;
+      cl[f] = keep;  // widen into cols scratch (3*nt >= nt slots)
     }
-    __syncthreads();
-    // compaction offsets over keep flags (u8 -> widen via deg? use
-    // per-face widen into cols as scratch: cols has 3*nt >= nt slots)
-    for (uint32_t f = tid; f < nt; f += 256) cl[f] = valid[f];
     __syncthreads();
     uint32_t kept = blk_exscan(cl, cl + nt, nt, 0, s_sums);
     for (uint32_t f = tid; f < nt; f += 256) {
       if (!valid[f]) continue;
       uint32_t o = cl[nt + f];
-      ftmp[3*o] = faces[3*f]; ftmp[3*o+1] = faces[3*f+1];
-      ftmp[3*o+2] = faces[3*f+2];
+      fb[3*o] = fa[3*f]; fb[3*o+1] = fa[3*f+1]; fb[3*o+2] = fa[3*f+2];
     }
     __syncthreads();
-    for (uint32_t i = tid; i < 3*kept; i += 256) faces[i] = ftmp[i];
+    { uint32_t *t = fa; fa = fb; fb = t; }  // compacted faces now in fa
     if (tid == 0) s_nt = kept;
     __syncthreads();
     PHASE_MARK(5)  // collapse + rewrite + compact
@@ -727,7 +729,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   }
   // park the final faces at the label's original offset
   for (uint32_t i = tid; i < 3 * s_nt; i += 256)
-    park_faces[3ull * f0 + i] = faces[i];
+    park_faces[3ull * f0 + i] = fa[i];
   if (tid == 0) {
     nt_cur[b] = s_nt;
     active[b] = 0;  // done: global rounds skip this label
