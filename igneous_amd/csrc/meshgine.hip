@@ -821,6 +821,13 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   // workgroup each (cache-resident round loop, no global sorts); only
   // bigger labels go through the global-rounds machinery below.
   const uint32_t SIMP_BIG_CAP = 65536;
+  // sub-rounds per quadric recompute (contract constant; the oracle
+  // reads the same env knob, default 4 on both sides)
+  uint32_t simp_subs = 4;
+  {
+    const char *e = getenv("MG_SIMP_SUBS");
+    if (e && e[0]) { int v = atoi(e); simp_subs = v < 1 ? 1u : (uint32_t)v; }
+  }
   if (getenv("MG_SIMP_PROF"))
     HIP_TRY(c, hipMemsetAsync((unsigned long long *)c->lh_misc.ptr + 8, 0,
                               48, s), 40);
@@ -845,7 +852,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                        getenv("MG_SIMP_PROF")
                            ? (unsigned long long *)c->lh_misc.ptr + 8
                            : nullptr,
-                       max_cost, (uint32_t)L, SIMP_BIG_CAP);
+                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs);
     HIP_TRY(c, hipGetLastError(), 40);
     if (getenv("MG_SIMP_PROF")) {
       unsigned long long hp[6];

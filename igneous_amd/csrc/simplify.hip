@@ -465,7 +465,8 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     uint8_t *__restrict__ active,
     uint32_t *__restrict__ park_faces,
     unsigned long long *__restrict__ prof,  // 6 phase counters or null
-    float max_cost, uint32_t nlabels, uint32_t big_cap) {
+    float max_cost, uint32_t nlabels, uint32_t big_cap,
+    uint32_t subs) {
   const uint32_t b = blockIdx.x;
   if (b >= nlabels) return;
   const uint32_t f0 = tri_off[b];
@@ -524,8 +525,9 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   }
 
   for (int round = 0; round < 65536; ++round) {
-    const uint32_t nt = s_nt;
+    uint32_t nt = s_nt;
     if (nt <= tgt) break;
+    const uint32_t nt_group = nt;  // group-progress watermark
     const bool clmode = CLLDS && (nt <= CAPF);
     PHASE_MARK(0)  // loop head
 
@@ -649,6 +651,16 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     }
     __syncthreads();
     PHASE_MARK(3)  // sort + quadric accumulate
+    // sub-rounds: reuse merged quadrics (Q[u] += Q[w] on collapse) for
+    // up to `subs` pick/collapse/compact passes per recompute (oracle
+    // group structure; a full recompute resets the drift)
+    for (uint32_t sub = 0; sub < subs; ++sub) {
+    nt = s_nt;
+    if (nt <= tgt) break;
+    if (sub > 0) {
+      for (uint32_t v = tid; v < nv; v += 256) pick_l[v] = ~0ull;
+      __syncthreads();
+    }
     // [6] picks (oracle step 2)
     for (uint32_t f = tid; f < nt; f += 256) {
       uint32_t fc[3] = {fa[3*f], fa[3*f+1], fa[3*f+2]};
@@ -692,6 +704,8 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       verts[3ull*u]   = 0.5f*(verts[3ull*u]+verts[3ull*w]);
       verts[3ull*u+1] = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
       verts[3ull*u+2] = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
+      #pragma unroll
+      for (int k = 0; k < 10; ++k) Q[10ull*u + k] += Q[10ull*w + k];
       remap[w] = u;
       atomicAdd(&s_collapses, 1u);
     }
@@ -719,7 +733,9 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     if (tid == 0) s_nt = kept;
     __syncthreads();
     PHASE_MARK(5)  // collapse + rewrite + compact
-    if (kept == nt) break;  // no progress (oracle: progress == 0)
+    if (kept == nt) break;  // no face dropped this sub
+    }  // sub loop
+    if (s_nt == nt_group) break;  // group made no progress: terminate
   }
 #undef PHASE_MARK
   if (prof && tid == 0) {

@@ -20,8 +20,15 @@
  *    smaller peer index); an edge collapses iff both endpoints picked it
  *    AND cost <= max_error^2;
  *  - collapse moves both endpoints to the midpoint, remaps v->u (u<v),
- *    drops degenerate faces; rounds repeat until ntris <= target or no
- *    edge collapses in a round.
+ *    ADDS the loser's quadric into the winner (Q_u += Q_w, f32), and
+ *    drops degenerate faces;
+ *  - rounds are grouped: one full quadric recompute (step 1) is followed
+ *    by up to OMC_SUBS pick/collapse/compact sub-rounds that reuse the
+ *    merged quadrics (standard GH quadric merging; a full recompute
+ *    resets the drift every group). Labels that START above 65536 faces
+ *    use OMC_SUBS=1 (they run the engine's global-rounds path);
+ *  - iteration ends when ntris <= target or a group makes no progress
+ *    (the first sub-round of a group collapses nothing).
  */
 
 #include <stdint.h>
@@ -56,6 +63,16 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
   if (target < 1) target = 1;
   const float max_cost = max_error * max_error;
 
+  /* sub-rounds per quadric recompute; MG_SIMP_SUBS is a dev knob shared
+   * with the HIP engine (both default 4 -- part of the contract) */
+  uint32_t SUBS = 4;
+  {
+    const char *e = getenv("MG_SIMP_SUBS");
+    if (e && e[0]) SUBS = (uint32_t)atoi(e);
+    if (SUBS < 1) SUBS = 1;
+  }
+  if (nt > 65536) SUBS = 1; /* global-rounds path: plain handshake rounds */
+
   uint32_t *remap = (uint32_t*)malloc(nv * sizeof(uint32_t));
   quad10 *Q = (quad10*)malloc(nv * sizeof(quad10));
   /* pick[v]: encoded best edge for vertex v */
@@ -65,6 +82,7 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
   while (nt > target && progress) {
     progress = 0;
     /* 1. vertex quadrics, ascending face order */
+    {
     memset(Q, 0, nv * sizeof(quad10));
     for (uint32_t t = 0; t < nt; t++) {
       uint32_t i0 = faces[3*t], i1 = faces[3*t+1], i2 = faces[3*t+2];
@@ -81,6 +99,8 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
       quad_add_plane(&Q[i1], nx, ny, nz, d, 1.0f);
       quad_add_plane(&Q[i2], nx, ny, nz, d, 1.0f);
     }
+    }
+    for (uint32_t sub = 0; sub < SUBS && nt > target; sub++) {
     /* 2. per-vertex best incident edge: encode (costbits<<32 | peer) and
      * take min. Cost as raw f32 bits (all costs >= 0 so bit order == value
      * order); tie-break by smaller peer index. Deterministic: min over
@@ -126,6 +146,7 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
       verts[3*u]   = 0.5f*(verts[3*u]+verts[3*w]);
       verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
       verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+      for (int k = 0; k < 10; k++) Q[u].q[k] += Q[w].q[k];
       remap[w] = u;
       collapses++;
     }
@@ -140,9 +161,10 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
     }
     if (out < nt) progress = 1;
     if (getenv("OMC_DEBUG"))
-      fprintf(stderr, "[omc] round: collapses=%u nt %u -> %u\n",
-              collapses, nt, out);
+      fprintf(stderr, "[omc] round sub=%u: collapses=%u nt %u -> %u\n",
+              sub, collapses, nt, out);
     nt = out;
+    }
   }
 
   /* 5. compact vertices to those referenced, preserving index order */
