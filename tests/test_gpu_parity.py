@@ -228,6 +228,28 @@ def test_simplify_parity_multilabel(eng):
         _assert_meshsets_equal(got, want, f"simplify f={factor} e={err}")
 
 
+def test_simplify_parity_big_label_global_path(eng):
+    """A label that STARTS above 65536 faces takes the engine's
+    global-rounds simplify path (SIMP_BIG_CAP) — and, per the contract,
+    plain 1-sub-round groups on both sides — while small neighbors run
+    the per-label kernel with sub-round groups. Bit-exact vs the oracle
+    across the path split and the face-parking machinery."""
+    import oracle
+    data = np.zeros((136, 136, 136), dtype=np.uint64, order="F")
+    data[1:132, 1:132, 1:132] = 7       # ~2*6*130^2 = 202k faces: global path
+    data[2:30, 2:30, 2:30] = 9          # carved small label: per-label path
+    data[133:135, 133:135, 133:135] = 3  # tiny label
+    res = (16.0, 16.0, 40.0)
+    got = eng.mesh_chunk(data, resolution=res, reduction_factor=100,
+                         max_error=40.0)
+    want = oracle.mesh_chunk(data, resolution=res, reduction_factor=100,
+                             max_error=40.0)
+    _assert_meshsets_equal(got, want, "big-label global path")
+    full = oracle.mesh_chunk(data, resolution=res)
+    assert full[7][1].shape[0] > 65536      # really took the global path
+    assert got[7][1].shape[0] < full[7][1].shape[0] // 10
+
+
 def test_simplify_512_runs(eng):
     """BASELINE config 5: 512^3 with simplification_factor=100,
     max_error=40 through the quadric-collapse kernels; validity +
