@@ -832,9 +832,11 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     HIP_TRY(c, hipMemsetAsync((unsigned long long *)c->lh_misc.ptr + 8, 0,
                               48, s), 40);
   {
-    const char *clenv = getenv("MG_SIMP_CLLDS");  // "0" disables
-    auto *ksl = (clenv && clenv[0] == '0') ? k_simplify_label<false>
-                                           : k_simplify_label<true>;
+    // default OFF since sub-round groups: the +24 KB LDS CSR payload
+    // costs blocks/CU and the recompute phases it serves are now ~25%
+    const char *clenv = getenv("MG_SIMP_CLLDS");  // "1" enables
+    auto *ksl = (clenv && clenv[0] == '1') ? k_simplify_label<true>
+                                           : k_simplify_label<false>;
     hipLaunchKernelGGL(ksl, dim3((uint32_t)L), dim3(256), 0, s,
                        faces_g, (uint32_t *)c->simp_faces_alt.ptr,
                        (const uint32_t *)c->tri_off.ptr,

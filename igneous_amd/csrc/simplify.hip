@@ -445,6 +445,41 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
   return total;
 }
 
+// block-wide stable compaction of kept faces fa->fb (single pass over
+// the chunked keep flags after the partial-sum scan; the scatter rides
+// the same loop that would have written scan offsets)
+__device__ uint32_t blk_compact_faces(const uint32_t *__restrict__ fa,
+                                      uint32_t *__restrict__ fb,
+                                      const uint8_t *__restrict__ valid,
+                                      uint32_t n, uint32_t *s_sums) {
+  const uint32_t tid = threadIdx.x;
+  const uint32_t chunk = (n + 255) / 256;
+  const uint32_t lo = tid * chunk;
+  const uint32_t hi = lo + chunk < n ? lo + chunk : n;
+  uint32_t sum = 0;
+  for (uint32_t i = lo; i < hi; ++i) sum += valid[i];
+  s_sums[tid] = sum;
+  __syncthreads();
+  #pragma unroll
+  for (uint32_t st = 1; st < 256; st <<= 1) {
+    uint32_t x = (tid >= st) ? s_sums[tid - st] : 0;
+    __syncthreads();
+    s_sums[tid] += x;
+    __syncthreads();
+  }
+  uint32_t run = s_sums[tid] - sum;
+  if (tid == 255) s_sums[256] = s_sums[255];
+  __syncthreads();
+  for (uint32_t i = lo; i < hi; ++i)
+    if (valid[i]) {
+      fb[3*run] = fa[3*i]; fb[3*run+1] = fa[3*i+1]; fb[3*run+2] = fa[3*i+2];
+      ++run;
+    }
+  uint32_t total = s_sums[256];
+  __syncthreads();
+  return total;
+}
+
 template <bool CLLDS>
 __global__ __launch_bounds__(256) void k_simplify_label(
     uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
@@ -503,10 +538,14 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   // for A/B (MG_SIMP_CLLDS=0 disables; measured ~5% faster on, default on).
   constexpr uint32_t CAPF = CLLDS ? 4096u : 1u;
   __shared__ uint16_t s_cl[3 * CAPF];
+  // remap holds label-LOCAL canonical ids; LDS-resident when the label
+  // fits (rewrite's 3 gathers/face are the hot readers)
+  __shared__ uint32_t s_remap[CAPV];
   const bool lds_mode = (nv <= CAPV);
   uint32_t *dg = lds_mode ? s_deg : (deg + v0);
   unsigned long long *pick_l =
       lds_mode ? s_pick : (pick + v0);
+  uint32_t *rm = lds_mode ? s_remap : (remap + v0);
   // ping-pong face buffers: rewrite reads fa, compaction scatters into
   // fb, then the buffers swap — no copy-back pass. Parking at the end
   // reads whichever buffer is current.
@@ -691,7 +730,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     PHASE_MARK(4)  // edge picks
     // [7] matched-pair collapse (oracle step 3)
     if (tid == 0) s_collapses = 0;
-    for (uint32_t v = tid; v < nv; v += 256) remap[v0 + v] = v0 + v;
+    for (uint32_t v = tid; v < nv; v += 256) rm[v] = v;
     __syncthreads();
     for (uint32_t v = tid; v < nv; v += 256) {
       uint32_t u = v0 + v;
@@ -706,29 +745,20 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       verts[3ull*u+2] = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
       #pragma unroll
       for (int k = 0; k < 10; ++k) Q[10ull*u + k] += Q[10ull*w + k];
-      remap[w] = u;
+      rm[w - v0] = u - v0;
       atomicAdd(&s_collapses, 1u);
     }
     __syncthreads();
     if (s_collapses == 0) break;
     // [8] rewrite + stable compact (oracle step 4); keep flag in valid[]
-    // and widened into cl[] in the same pass
     for (uint32_t f = tid; f < nt; f += 256) {
-      uint32_t i0 = remap[fa[3*f]], i1 = remap[fa[3*f+1]],
-               i2 = remap[fa[3*f+2]];
+      uint32_t i0 = v0 + rm[fa[3*f] - v0], i1 = v0 + rm[fa[3*f+1] - v0],
+               i2 = v0 + rm[fa[3*f+2] - v0];
       fa[3*f] = i0; fa[3*f+1] = i1; fa[3*f+2] = i2;
-      uint32_t keep = (i0 != i1 && i1 != i2 && i0 != i2) ? 1u : 0u;
-      valid[f] = (uint8_t)keep;
-      cl[f] = keep;  // widen into cols scratch (3*nt >= nt slots)
+      valid[f] = (i0 != i1 && i1 != i2 && i0 != i2) ? 1 : 0;
     }
     __syncthreads();
-    uint32_t kept = blk_exscan(cl, cl + nt, nt, 0, s_sums);
-    for (uint32_t f = tid; f < nt; f += 256) {
-      if (!valid[f]) continue;
-      uint32_t o = cl[nt + f];
-      fb[3*o] = fa[3*f]; fb[3*o+1] = fa[3*f+1]; fb[3*o+2] = fa[3*f+2];
-    }
-    __syncthreads();
+    uint32_t kept = blk_compact_faces(fa, fb, valid, nt, s_sums);
     { uint32_t *t = fa; fa = fb; fb = t; }  // compacted faces now in fa
     if (tid == 0) s_nt = kept;
     __syncthreads();
