@@ -822,8 +822,8 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   // bigger labels go through the global-rounds machinery below.
   const uint32_t SIMP_BIG_CAP = 65536;
   // sub-rounds per quadric recompute (contract constant; the oracle
-  // reads the same env knob, default 4 on both sides)
-  uint32_t simp_subs = 4;
+  // reads the same env knob, default 6 on both sides)
+  uint32_t simp_subs = 6;
   {
     const char *e = getenv("MG_SIMP_SUBS");
     if (e && e[0]) { int v = atoi(e); simp_subs = v < 1 ? 1u : (uint32_t)v; }
@@ -835,9 +835,21 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     // default OFF since sub-round groups: the +24 KB LDS CSR payload
     // costs blocks/CU and the recompute phases it serves are now ~25%
     const char *clenv = getenv("MG_SIMP_CLLDS");  // "1" enables
-    auto *ksl = (clenv && clenv[0] == '1') ? k_simplify_label<true>
-                                           : k_simplify_label<false>;
-    hipLaunchKernelGGL(ksl, dim3((uint32_t)L), dim3(256), 0, s,
+    const bool use_cl = (clenv && clenv[0] == '1');
+    // block size: 256 default; MG_SIMP_BS in {128,256,512} for A/B
+    int bs = 256;
+    {
+      const char *e = getenv("MG_SIMP_BS");
+      if (e && e[0]) { int v = atoi(e); if (v==128||v==256||v==512) bs = v; }
+    }
+    decltype(&k_simplify_label<false, 256>) ksl;
+    if (bs == 128)
+      ksl = use_cl ? k_simplify_label<true,128> : k_simplify_label<false,128>;
+    else if (bs == 512)
+      ksl = use_cl ? k_simplify_label<true,512> : k_simplify_label<false,512>;
+    else
+      ksl = use_cl ? k_simplify_label<true,256> : k_simplify_label<false,256>;
+    hipLaunchKernelGGL(ksl, dim3((uint32_t)L), dim3(bs), 0, s,
                        faces_g, (uint32_t *)c->simp_faces_alt.ptr,
                        (const uint32_t *)c->tri_off.ptr,
                        (const uint32_t *)c->vbase.ptr,

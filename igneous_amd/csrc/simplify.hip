@@ -410,12 +410,13 @@ __global__ void k_gather_final(const uint32_t *__restrict__ park_faces,
 // matched collapse, stable compaction, same termination conditions).
 
 // block-wide exclusive scan of src[0..n) into dst (+base), returns total
+template <int BS>
 __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
                                uint32_t n, uint32_t base,
-                               uint32_t *s_sums /*257*/,
+                               uint32_t *s_sums /*BS+1*/,
                                uint32_t *dst2 = nullptr) {
   const uint32_t tid = threadIdx.x;
-  const uint32_t chunk = (n + 255) / 256;
+  const uint32_t chunk = (n + BS - 1) / BS;
   const uint32_t lo = tid * chunk;
   const uint32_t hi = lo + chunk < n ? lo + chunk : n;
   uint32_t sum = 0;
@@ -424,14 +425,14 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
   __syncthreads();
   // Hillis-Steele inclusive scan over the 256 partials (log steps)
   #pragma unroll
-  for (uint32_t st = 1; st < 256; st <<= 1) {
+  for (uint32_t st = 1; st < BS; st <<= 1) {
     uint32_t x = (tid >= st) ? s_sums[tid - st] : 0;
     __syncthreads();
     s_sums[tid] += x;
     __syncthreads();
   }
   uint32_t excl = s_sums[tid] - sum;      // exclusive from inclusive
-  if (tid == 255) s_sums[256] = s_sums[255];
+  if (tid == BS - 1) s_sums[BS] = s_sums[BS - 1];
   __syncthreads();
   uint32_t run = base + excl;
   for (uint32_t i = lo; i < hi; ++i) {
@@ -440,7 +441,7 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
     if (dst2) dst2[i] = run;  // optional cursor copy (src may alias
     run += t;                 // dst2: t was read first)
   }
-  uint32_t total = s_sums[256];
+  uint32_t total = s_sums[BS];
   __syncthreads();
   return total;
 }
@@ -448,12 +449,13 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
 // block-wide stable compaction of kept faces fa->fb (single pass over
 // the chunked keep flags after the partial-sum scan; the scatter rides
 // the same loop that would have written scan offsets)
+template <int BS>
 __device__ uint32_t blk_compact_faces(const uint32_t *__restrict__ fa,
                                       uint32_t *__restrict__ fb,
                                       const uint8_t *__restrict__ valid,
                                       uint32_t n, uint32_t *s_sums) {
   const uint32_t tid = threadIdx.x;
-  const uint32_t chunk = (n + 255) / 256;
+  const uint32_t chunk = (n + BS - 1) / BS;
   const uint32_t lo = tid * chunk;
   const uint32_t hi = lo + chunk < n ? lo + chunk : n;
   uint32_t sum = 0;
@@ -461,27 +463,27 @@ __device__ uint32_t blk_compact_faces(const uint32_t *__restrict__ fa,
   s_sums[tid] = sum;
   __syncthreads();
   #pragma unroll
-  for (uint32_t st = 1; st < 256; st <<= 1) {
+  for (uint32_t st = 1; st < BS; st <<= 1) {
     uint32_t x = (tid >= st) ? s_sums[tid - st] : 0;
     __syncthreads();
     s_sums[tid] += x;
     __syncthreads();
   }
   uint32_t run = s_sums[tid] - sum;
-  if (tid == 255) s_sums[256] = s_sums[255];
+  if (tid == BS - 1) s_sums[BS] = s_sums[BS - 1];
   __syncthreads();
   for (uint32_t i = lo; i < hi; ++i)
     if (valid[i]) {
       fb[3*run] = fa[3*i]; fb[3*run+1] = fa[3*i+1]; fb[3*run+2] = fa[3*i+2];
       ++run;
     }
-  uint32_t total = s_sums[256];
+  uint32_t total = s_sums[BS];
   __syncthreads();
   return total;
 }
 
-template <bool CLLDS>
-__global__ __launch_bounds__(256) void k_simplify_label(
+template <bool CLLDS, int BS>
+__global__ __launch_bounds__(BS) void k_simplify_label(
     uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
     uint32_t *__restrict__ faces_tmp,     // same slicing (scratch)
     const uint32_t *__restrict__ tri_off, // L+1 (original offsets)
@@ -509,7 +511,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   if (nt0 > big_cap) return;  // global-rounds path handles big labels
   if (!active[b]) {
     // already at/below target: final faces = original faces; park them
-    for (uint32_t i = threadIdx.x; i < 3 * nt0; i += 256)
+    for (uint32_t i = threadIdx.x; i < 3 * nt0; i += BS)
       park_faces[3ull * f0 + i] = faces_g[3ull * f0 + i];
     return;
   }
@@ -524,7 +526,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
   uint32_t *aoff = adj_off + v0;
   uint32_t *cl = cols + 3ull * f0;
 
-  __shared__ uint32_t s_sums[257];
+  __shared__ uint32_t s_sums[BS + 1];
   __shared__ uint32_t s_nt, s_collapses;
   // the two ATOMIC-hot per-vertex arrays live in LDS when the label fits:
   // neighboring faces' vertices share cache lines, so the global
@@ -571,9 +573,9 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     PHASE_MARK(0)  // loop head
 
     // [1+2] face planes fused with the CSR degree count (one face pass)
-    for (uint32_t v = tid; v < nv; v += 256) dg[v] = 0;
+    for (uint32_t v = tid; v < nv; v += BS) dg[v] = 0;
     __syncthreads();
-    for (uint32_t f = tid; f < nt; f += 256) {
+    for (uint32_t f = tid; f < nt; f += BS) {
       uint32_t i0 = fa[3*f], i1 = fa[3*f+1], i2 = fa[3*f+2];
       atomicAdd(&dg[i0 - v0], 1u);
       atomicAdd(&dg[i1 - v0], 1u);
@@ -596,15 +598,15 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     __syncthreads();
     PHASE_MARK(1)  // planes + degree count
     // [3] offsets; the scan also writes the fill cursors (dst2 = dg)
-    blk_exscan(dg, aoff, nv, 0, s_sums, dg);
+    blk_exscan<BS>(dg, aoff, nv, 0, s_sums, dg);
     if (clmode) {
-      for (uint32_t f = tid; f < nt; f += 256) {
+      for (uint32_t f = tid; f < nt; f += BS) {
         s_cl[atomicAdd(&dg[fa[3*f] - v0], 1u)] = (uint16_t)f;
         s_cl[atomicAdd(&dg[fa[3*f+1] - v0], 1u)] = (uint16_t)f;
         s_cl[atomicAdd(&dg[fa[3*f+2] - v0], 1u)] = (uint16_t)f;
       }
     } else {
-      for (uint32_t f = tid; f < nt; f += 256) {
+      for (uint32_t f = tid; f < nt; f += BS) {
         cl[atomicAdd(&dg[fa[3*f] - v0], 1u)] = f;
         cl[atomicAdd(&dg[fa[3*f+1] - v0], 1u)] = f;
         cl[atomicAdd(&dg[fa[3*f+2] - v0], 1u)] = f;
@@ -614,7 +616,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     PHASE_MARK(2)  // offsets scan + CSR fill
     // [5] per-vertex: sort incident faces ascending (insertion sort),
     // accumulate quadrics in that order (oracle step 1)
-    for (uint32_t v = tid; v < nv; v += 256) {
+    for (uint32_t v = tid; v < nv; v += BS) {
       uint32_t lo = aoff[v];
       uint32_t hi = dg[v];  // cursor ended at one-past-last
       uint32_t d = hi - lo;
@@ -697,11 +699,11 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     nt = s_nt;
     if (nt <= tgt) break;
     if (sub > 0) {
-      for (uint32_t v = tid; v < nv; v += 256) pick_l[v] = ~0ull;
+      for (uint32_t v = tid; v < nv; v += BS) pick_l[v] = ~0ull;
       __syncthreads();
     }
     // [6] picks (oracle step 2)
-    for (uint32_t f = tid; f < nt; f += 256) {
+    for (uint32_t f = tid; f < nt; f += BS) {
       uint32_t fc[3] = {fa[3*f], fa[3*f+1], fa[3*f+2]};
       #pragma unroll
       for (int e = 0; e < 3; ++e) {
@@ -730,9 +732,9 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     PHASE_MARK(4)  // edge picks
     // [7] matched-pair collapse (oracle step 3)
     if (tid == 0) s_collapses = 0;
-    for (uint32_t v = tid; v < nv; v += 256) rm[v] = v;
+    for (uint32_t v = tid; v < nv; v += BS) rm[v] = v;
     __syncthreads();
-    for (uint32_t v = tid; v < nv; v += 256) {
+    for (uint32_t v = tid; v < nv; v += BS) {
       uint32_t u = v0 + v;
       unsigned long long pu = pick_l[v];
       if (pu == ~0ull) continue;
@@ -751,14 +753,14 @@ __global__ __launch_bounds__(256) void k_simplify_label(
     __syncthreads();
     if (s_collapses == 0) break;
     // [8] rewrite + stable compact (oracle step 4); keep flag in valid[]
-    for (uint32_t f = tid; f < nt; f += 256) {
+    for (uint32_t f = tid; f < nt; f += BS) {
       uint32_t i0 = v0 + rm[fa[3*f] - v0], i1 = v0 + rm[fa[3*f+1] - v0],
                i2 = v0 + rm[fa[3*f+2] - v0];
       fa[3*f] = i0; fa[3*f+1] = i1; fa[3*f+2] = i2;
       valid[f] = (i0 != i1 && i1 != i2 && i0 != i2) ? 1 : 0;
     }
     __syncthreads();
-    uint32_t kept = blk_compact_faces(fa, fb, valid, nt, s_sums);
+    uint32_t kept = blk_compact_faces<BS>(fa, fb, valid, nt, s_sums);
     { uint32_t *t = fa; fa = fb; fb = t; }  // compacted faces now in fa
     if (tid == 0) s_nt = kept;
     __syncthreads();
@@ -774,7 +776,7 @@ __global__ __launch_bounds__(256) void k_simplify_label(
       if (ph[k]) atomicAdd(&prof[k], ph[k]);
   }
   // park the final faces at the label's original offset
-  for (uint32_t i = tid; i < 3 * s_nt; i += 256)
+  for (uint32_t i = tid; i < 3 * s_nt; i += BS)
     park_faces[3ull * f0 + i] = fa[i];
   if (tid == 0) {
     nt_cur[b] = s_nt;

@@ -64,8 +64,8 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
   const float max_cost = max_error * max_error;
 
   /* sub-rounds per quadric recompute; MG_SIMP_SUBS is a dev knob shared
-   * with the HIP engine (both default 4 -- part of the contract) */
-  uint32_t SUBS = 4;
+   * with the HIP engine (both default 6 -- part of the contract) */
+  uint32_t SUBS = 6;
   {
     const char *e = getenv("MG_SIMP_SUBS");
     if (e && e[0]) SUBS = (uint32_t)atoi(e);
