@@ -842,13 +842,45 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       const char *e = getenv("MG_SIMP_BS");
       if (e && e[0]) { int v = atoi(e); if (v==128||v==256||v==512) bs = v; }
     }
-    decltype(&k_simplify_label<false, 256>) ksl;
+    decltype(&k_simplify_label<false, 256, 2048>) ksl;
     if (bs == 128)
-      ksl = use_cl ? k_simplify_label<true,128> : k_simplify_label<false,128>;
+      ksl = use_cl ? k_simplify_label<true, 128, 2048>
+                   : k_simplify_label<false, 128, 2048>;
     else if (bs == 512)
-      ksl = use_cl ? k_simplify_label<true,512> : k_simplify_label<false,512>;
+      ksl = use_cl ? k_simplify_label<true, 512, 2048>
+                   : k_simplify_label<false, 512, 2048>;
     else
-      ksl = use_cl ? k_simplify_label<true,256> : k_simplify_label<false,256>;
+      ksl = use_cl ? k_simplify_label<true, 256, 2048>
+                   : k_simplify_label<false, 256, 2048>;
+    // single-wave small-LDS variant takes labels with nv <= small_cap
+    // (~8.3 KB LDS -> ~19 blocks/CU, intra-wave barriers); the default
+    // variant takes the rest. Measured ~1% WORSE in-box (the serialized
+    // second launch eats the occupancy gain) -> default off;
+    // MG_SIMP_SMALL=1 enables for experiments.
+    const char *sm = getenv("MG_SIMP_SMALL");
+    const uint32_t small_cap = (sm && sm[0] == '1') ? 512u : 0u;
+    if (small_cap) {
+      hipLaunchKernelGGL((k_simplify_label<false, 64, 512>),
+                         dim3((uint32_t)L), dim3(64), 0, s,
+                       faces_g, (uint32_t *)c->simp_faces_alt.ptr,
+                       (const uint32_t *)c->tri_off.ptr,
+                       (const uint32_t *)c->vbase.ptr,
+                       verts, (float *)c->simp_Q.ptr,
+                       (unsigned long long *)c->simp_pick.ptr,
+                       (uint32_t *)c->simp_remap.ptr,
+                       (uint32_t *)c->simp_deg.ptr,
+                       (uint32_t *)c->simp_adj.ptr,
+                       (uint32_t *)c->simp_pk.ptr,
+                       (SimpPlane *)c->simp_fq.ptr,
+                       (uint8_t *)c->simp_valid.ptr,
+                       nt_cur, target, active,
+                       (uint32_t *)c->simp_park.ptr,
+                       getenv("MG_SIMP_PROF")
+                           ? (unsigned long long *)c->lh_misc.ptr + 8
+                           : nullptr,
+                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, 0u, small_cap);
+      HIP_TRY(c, hipGetLastError(), 40);
+    }
     hipLaunchKernelGGL(ksl, dim3((uint32_t)L), dim3(bs), 0, s,
                        faces_g, (uint32_t *)c->simp_faces_alt.ptr,
                        (const uint32_t *)c->tri_off.ptr,
@@ -866,7 +898,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                        getenv("MG_SIMP_PROF")
                            ? (unsigned long long *)c->lh_misc.ptr + 8
                            : nullptr,
-                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs);
+                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, small_cap, 0xFFFFFFFFu);
     HIP_TRY(c, hipGetLastError(), 40);
     if (getenv("MG_SIMP_PROF")) {
       unsigned long long hp[6];

@@ -482,7 +482,7 @@ __device__ uint32_t blk_compact_faces(const uint32_t *__restrict__ fa,
   return total;
 }
 
-template <bool CLLDS, int BS>
+template <bool CLLDS, int BS, int CAPVT>
 __global__ __launch_bounds__(BS) void k_simplify_label(
     uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
     uint32_t *__restrict__ faces_tmp,     // same slicing (scratch)
@@ -503,20 +503,23 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
     uint32_t *__restrict__ park_faces,
     unsigned long long *__restrict__ prof,  // 6 phase counters or null
     float max_cost, uint32_t nlabels, uint32_t big_cap,
-    uint32_t subs) {
+    uint32_t subs, uint32_t nv_lo, uint32_t nv_hi) {
   const uint32_t b = blockIdx.x;
   if (b >= nlabels) return;
   const uint32_t f0 = tri_off[b];
   const uint32_t nt0 = tri_off[b + 1] - f0;
   if (nt0 > big_cap) return;  // global-rounds path handles big labels
+  const uint32_t v0 = vbase[b];
+  const uint32_t nv = vbase[b + 1] - v0;
+  // size-class dispatch: each launch variant owns an (nv_lo, nv_hi]
+  // vertex-count band (single-wave small-LDS variant for tiny labels)
+  if (nv <= nv_lo || nv > nv_hi) return;
   if (!active[b]) {
     // already at/below target: final faces = original faces; park them
     for (uint32_t i = threadIdx.x; i < 3 * nt0; i += BS)
       park_faces[3ull * f0 + i] = faces_g[3ull * f0 + i];
     return;
   }
-  const uint32_t v0 = vbase[b];
-  const uint32_t nv = vbase[b + 1] - v0;
   const uint32_t tgt = target[b];
   const uint32_t tid = threadIdx.x;
   uint32_t *faces = faces_g + 3ull * f0;
@@ -532,7 +535,7 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
   // neighboring faces' vertices share cache lines, so the global
   // atomicAdd/atomicMin streams serialize on hot L2 lines — LDS atomics
   // are bank-parallel. (~25 KB -> ~6 blocks/CU.)
-  constexpr uint32_t CAPV = 2048;
+  constexpr uint32_t CAPV = CAPVT;
   __shared__ uint32_t s_deg[CAPV];
   __shared__ unsigned long long s_pick[CAPV];
   // optional LDS-resident CSR payload (face ids, u16) for rounds whose
