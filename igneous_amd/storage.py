@@ -48,8 +48,13 @@ class CloudFiles:
             content = gzip.compress(content)
         elif compress not in (None, False, ""):
             raise ValueError(f"unsupported compression {compress!r}")
-        with open(path, "wb") as f:
+        # atomic publish: concurrent readers (multi-rank dispatch workers
+        # sharing one layer, e.g. the info/provenance JSON both ranks
+        # rewrite in create_meshing_tasks) must never see a torn file
+        tmp = f"{path}.tmp.{os.getpid()}"
+        with open(tmp, "wb") as f:
             f.write(content)
+        os.replace(tmp, path)
 
     def puts(self, items: Iterable[Tuple[str, bytes]],
              compress: Optional[str] = None, **kw) -> None:
