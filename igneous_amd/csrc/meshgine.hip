@@ -957,7 +957,8 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     T = act_total;
   }
 
-  for (int round = 0; round < 65536; ++round) {
+  bool simp_done = false;
+  for (int group = 0; group < 65536 && !simp_done; ++group) {
     // any label still active?
     HIP_TRY(c, hipMemsetAsync(d_any, 0, 4, s), 41);
     {
@@ -1003,13 +1004,21 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                          (const uint8_t *)c->simp_valid.ptr,
                          (float *)c->simp_Q.ptr, NP);
     }
+    // sub-rounds: reuse the group's quadrics with GH merging (the same
+    // group structure as the per-label kernel and the oracle). Measured
+    // on 512^3/200 big labels: subs>1 is ~5% SLOWER here (the T- and
+    // V-sized sub passes dwarf the recompute they skip, unlike the
+    // per-label kernel) -> the contract pins big labels at 1 sub/group.
+    const uint32_t subs_global = 1;
+    for (uint32_t sub = 0; sub < subs_global; ++sub) {
+    uint64_t nbs = (T + blk - 1) / blk;  // T shrinks between subs
     HIP_TRY(c, hipMemsetAsync(c->simp_pick.ptr, 0xFF, V * 8, s), 43);
-    hipLaunchKernelGGL(k_edge_pick, dim3((uint32_t)nbt), dim3(blk), 0, s,
+    hipLaunchKernelGGL(k_edge_pick, dim3((uint32_t)nbs), dim3(blk), 0, s,
                        faces_g, active, flab,
                        (const uint32_t *)c->vbase.ptr, verts,
                        (const float *)c->simp_Q.ptr,
                        (unsigned long long *)c->simp_pick.ptr, max_cost, T);
-    if (round == 0 && getenv("MG_DEBUG_SIMPLIFY")) {
+    if (group == 0 && sub == 0 && getenv("MG_DEBUG_SIMPLIFY")) {
       float hq[40]; unsigned long long hp[8]; uint32_t hf[12];
       (void)hipMemcpyAsync(hq, c->simp_Q.ptr, sizeof(hq), hipMemcpyDeviceToHost, s);
       (void)hipMemcpyAsync(hp, c->simp_pick.ptr, sizeof(hp), hipMemcpyDeviceToHost, s);
@@ -1031,9 +1040,10 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                          (uint32_t *)c->simp_remap.ptr, V);
       hipLaunchKernelGGL(k_collapse, dim3((uint32_t)nbv), dim3(blk), 0, s,
                          (const unsigned long long *)c->simp_pick.ptr, verts,
-                         (uint32_t *)c->simp_remap.ptr, V);
+                         (uint32_t *)c->simp_remap.ptr,
+                         (float *)c->simp_Q.ptr, V);
     }
-    hipLaunchKernelGGL(k_remap_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
+    hipLaunchKernelGGL(k_remap_faces, dim3((uint32_t)nbs), dim3(blk), 0, s,
                        faces_g, (const uint32_t *)c->simp_remap.ptr,
                        (uint32_t *)c->simp_keep.ptr, T);
     {
@@ -1058,7 +1068,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     HIP_TRY(c, hipMemcpyAsync(&kept, (uint32_t *)c->lh_misc.ptr + 4, 4,
                               hipMemcpyDeviceToHost, s), 44);
     HIP_TRY(c, hipMemsetAsync(nt_new, 0, L * 4, s), 44);
-    hipLaunchKernelGGL(k_compact_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
+    hipLaunchKernelGGL(k_compact_faces, dim3((uint32_t)nbs), dim3(blk), 0, s,
                        faces_g, flab, (const uint32_t *)c->simp_keep.ptr,
                        (const uint32_t *)c->simp_keep_scan.ptr,
                        (uint32_t *)c->simp_faces_alt.ptr,
@@ -1066,7 +1076,8 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     {
       uint32_t nbl = (uint32_t)((L + 255) / 256);
       hipLaunchKernelGGL(k_update_active, dim3(nbl), dim3(256), 0, s,
-                         nt_new, nt_cur, target, active, d_any, (uint32_t)L);
+                         nt_new, nt_cur, target, active, d_any, (uint32_t)L,
+                         sub == 0 ? 1u : 0u);
     }
     HIP_TRY(c, hipGetLastError(), 44);
     HIP_TRY(c, hipStreamSynchronize(s), 44);
@@ -1074,7 +1085,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     std::swap(c->simp_flab, c->simp_flab_alt);
     faces_g = (uint32_t *)c->faces.ptr;
     T = kept;
-    if (T == 0) break;
+    if (T == 0) { simp_done = true; break; }
 
     // park faces of labels that just went inactive: working set keeps
     // only active labels (late rounds touch only the active tail)
@@ -1121,6 +1132,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       T = act_total;
       if (T == 0) break;
     }
+    }  // sub loop
   }
 
   // any faces still in the working set (round-cap exit): force-park them

@@ -175,6 +175,7 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
 __global__ void k_collapse(const unsigned long long *__restrict__ pick,
                            float *__restrict__ verts,
                            uint32_t *__restrict__ remap,
+                           float *__restrict__ Q,
                            uint64_t nverts) {
   uint64_t u = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (u >= nverts) return;
@@ -187,6 +188,8 @@ __global__ void k_collapse(const unsigned long long *__restrict__ pick,
   verts[3*u]   = 0.5f*(verts[3*u]+verts[3*w]);
   verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
   verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+  #pragma unroll
+  for (int k = 0; k < 10; ++k) Q[10*u + k] += Q[10*(uint64_t)w + k];
   remap[w] = (uint32_t)u;
 }
 
@@ -230,14 +233,19 @@ __global__ void k_update_active(const uint32_t *__restrict__ nt_new,
                                 const uint32_t *__restrict__ target,
                                 uint8_t *__restrict__ active_lab,
                                 uint32_t *__restrict__ any_active,
-                                uint32_t nlabels) {
+                                uint32_t nlabels,
+                                uint32_t enforce_progress) {
+  // enforce_progress=1 on a group's FIRST sub-round: a label whose
+  // fresh-quadric sub collapses nothing is done (oracle termination).
+  // Later subs leave no-progress labels active — they recompute next
+  // group (their remaining subs this group are deterministic no-ops).
   uint32_t l = blockIdx.x * blockDim.x + threadIdx.x;
   if (l >= nlabels) return;
   if (!active_lab[l]) return;
   uint32_t nn = nt_new[l];
   bool progress = nn != nt_cur[l];
   nt_cur[l] = nn;
-  bool act = progress && nn > target[l];
+  bool act = nn > target[l] && (progress || !enforce_progress);
   active_lab[l] = act ? 1 : 0;
   if (act) atomicExch(any_active, 1u);
 }

@@ -26,7 +26,8 @@
  *    by up to OMC_SUBS pick/collapse/compact sub-rounds that reuse the
  *    merged quadrics (standard GH quadric merging; a full recompute
  *    resets the drift every group). Labels that START above 65536 faces
- *    use OMC_SUBS=1 (they run the engine's global-rounds path);
+ *    use 1 sub-round per group (they run the engine's global-rounds
+ *    path, where the sub passes dwarf the recompute they would skip);
  *  - iteration ends when ntris <= target or a group makes no progress
  *    (the first sub-round of a group collapses nothing).
  */
@@ -71,8 +72,7 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
     if (e && e[0]) SUBS = (uint32_t)atoi(e);
     if (SUBS < 1) SUBS = 1;
   }
-  if (nt > 65536) SUBS = 1; /* global-rounds path: plain handshake rounds */
-
+  if (nt > 65536) SUBS = 1; /* global-rounds path: 1 sub-round per group */
   uint32_t *remap = (uint32_t*)malloc(nv * sizeof(uint32_t));
   quad10 *Q = (quad10*)malloc(nv * sizeof(quad10));
   /* pick[v]: encoded best edge for vertex v */
