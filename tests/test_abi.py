@@ -51,3 +51,16 @@ def test_so_is_gfx950_only():
     if "gfx" in txt:
         assert "gfx950" in txt
         assert not re.search(r"gfx(?!950)\d+", txt), txt
+
+
+@pytest.mark.skipif(not _built(), reason="libmeshgine.so not built")
+def test_engine_fails_loudly_without_gpu():
+    """The product path NEVER falls back to a CPU implementation: on a
+    host with no usable HIP device, Engine construction raises (the
+    oracle is test infrastructure only — DESIGN.md §2)."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("GPU present: covered by -m gpu parity tests")
+    from igneous_amd.engine import Engine
+    with pytest.raises(RuntimeError, match="mg_init|HIP"):
+        Engine(0)
