@@ -25,6 +25,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--n", type=int, default=20)
     ap.add_argument("--seed", type=int, default=1000)
+    ap.add_argument("--maxdim", type=int, default=72)
     args = ap.parse_args()
 
     import oracle
@@ -34,7 +35,7 @@ def main():
     eng = Engine.get(0)
     rng = np.random.default_rng(args.seed)
     for case in range(args.n):
-        dims = tuple(int(d) for d in rng.integers(3, 72, size=3))
+        dims = tuple(int(d) for d in rng.integers(3, args.maxdim, size=3))
         dtype = np.uint64 if rng.integers(2) else np.uint32
         style = int(rng.integers(3))
         seed = int(rng.integers(2**31))
