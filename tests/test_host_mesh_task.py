@@ -308,3 +308,56 @@ def test_transfer_and_delete_mesh_files(tmp_path, oracle_mesher):
     assert cf_dst.get("mesh/1:0:0-64_0-64_0-64") is not None
     DeleteMeshFilesTask(cloudpath=dst, prefix="1:")
     assert list(cf_dst.list("mesh/")) == []
+
+
+def test_frag_path_unsharded_matches_reference(tmp_layer_path,
+                                                oracle_mesher, tmp_path):
+    """In the reference, frag_path is consumed ONLY by the sharded
+    MapBuffer uploader (mesh.py:385-387); the unsharded path
+    (_upload_individuals, mesh.py:399-417) always writes fragments to
+    layer_path. Mirror that: frag_path with sharded=False is inert."""
+    _make_box_layer(tmp_layer_path)
+    frag = f"file://{tmp_path}/frags"
+    MeshTask(shape=(64, 64, 64), offset=(0, 0, 0),
+             layer_path=tmp_layer_path, mip=0, frag_path=frag,
+             simplification_factor=0).execute()
+    assert CloudFiles(tmp_layer_path).get('mesh/1:0:0-64_0-64_0-64') \
+        is not None
+    assert CloudFiles(frag).get('mesh/1:0:0-64_0-64_0-64') is None
+
+
+def test_dry_run_writes_nothing(tmp_layer_path, oracle_mesher):
+    """dry_run computes but uploads nothing (the reference's dry_run path
+    at mesh.py:249-252 hits a latent NameError; ours returns the data —
+    DESIGN.md §7)."""
+    _make_box_layer(tmp_layer_path)
+    result = MeshTask(shape=(64, 64, 64), offset=(0, 0, 0),
+                      layer_path=tmp_layer_path, mip=0, dry_run=True,
+                      simplification_factor=0).execute()
+    assert result is not None
+    meshes, bboxes = result
+    assert 1 in meshes and 1 in bboxes
+    assert list(CloudFiles(tmp_layer_path).list('mesh/')) == []
+
+
+def test_fill_missing_through_task(tmp_layer_path, oracle_mesher):
+    """fill_missing zero-fills absent chunks instead of raising
+    (mesh.py:177-182 download kwargs)."""
+    data = np.zeros((64, 64, 32), dtype=np.uint32)
+    data[1:-1, 1:-1, 1:-1] = 1
+    PrecomputedVolume.from_numpy(
+        data, tmp_layer_path, resolution=(1, 1, 1),
+        chunk_size=(32, 32, 32), mesh_dir="mesh")
+    # delete one stored chunk file out from under the task
+    cf = CloudFiles(tmp_layer_path)
+    gone = "1_1_1/32-64_32-64_0-32"
+    assert cf.get(gone) is not None
+    cf.delete(gone)
+    with pytest.raises(FileNotFoundError):
+        MeshTask(shape=(64, 64, 32), offset=(0, 0, 0),
+                 layer_path=tmp_layer_path, mip=0,
+                 simplification_factor=0).execute()
+    MeshTask(shape=(64, 64, 32), offset=(0, 0, 0),
+             layer_path=tmp_layer_path, mip=0, fill_missing=True,
+             simplification_factor=0).execute()
+    assert cf.get('mesh/1:0:0-64_0-64_0-32') is not None
