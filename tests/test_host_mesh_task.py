@@ -361,3 +361,26 @@ def test_fill_missing_through_task(tmp_layer_path, oracle_mesher):
              layer_path=tmp_layer_path, mip=0, fill_missing=True,
              simplification_factor=0).execute()
     assert cf.get('mesh/1:0:0-64_0-64_0-32') is not None
+
+
+def test_manifest_prefix_task(tmp_layer_path, oracle_mesher):
+    """MeshManifestPrefixTask (reference mesh.py:672-724): only segids
+    under the given prefix get manifests."""
+    from igneous_amd import MeshManifestPrefixTask
+    data = np.zeros((64, 64, 64), dtype=np.uint32)
+    data[1:30, 1:30, 1:30] = 1
+    data[32:63, 32:63, 32:63] = 21
+    PrecomputedVolume.from_numpy(
+        data, tmp_layer_path, resolution=(1, 1, 1), chunk_size=(64, 64, 64),
+        mesh_dir="mesh")
+    MeshTask(shape=(64, 64, 64), offset=(0, 0, 0),
+             layer_path=tmp_layer_path, mip=0,
+             simplification_factor=0).execute()
+    cf = CloudFiles(tmp_layer_path)
+    MeshManifestPrefixTask(layer_path=tmp_layer_path, prefix="2")
+    assert cf.get_json('mesh/21:0') == {
+        "fragments": ["21:0:0-64_0-64_0-64"]}
+    assert cf.get_json('mesh/1:0') is None   # prefix "2" excludes segid 1
+    MeshManifestPrefixTask(layer_path=tmp_layer_path, prefix="1")
+    assert cf.get_json('mesh/1:0') == {
+        "fragments": ["1:0:0-64_0-64_0-64"]}
