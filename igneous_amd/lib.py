@@ -65,10 +65,18 @@ class Bbox:
             np.minimum(np.maximum(bbx.maxpt, bounds.minpt), bounds.maxpt),
         )
 
-    def to_filename(self) -> str:
-        # reference naming: "x0-x1_y0-y1_z0-z1" (mesh.py:409 via Bbox.to_filename)
+    def to_filename(self, precision=None) -> str:
+        # reference naming: "x0-x1_y0-y1_z0-z1" (mesh.py:409 via Bbox.to_filename).
+        # precision mirrors cloudvolume Bbox.to_filename(precision) as used by
+        # the spatial index (mesh.py:454,460): None → plain integer repr;
+        # an int → floats formatted to that many decimals.
+        if precision is None:
+            return "_".join(
+                f"{int(self.minpt[i])}-{int(self.maxpt[i])}" for i in range(3)
+            )
         return "_".join(
-            f"{int(self.minpt[i])}-{int(self.maxpt[i])}" for i in range(3)
+            f"{float(self.minpt[i]):.{int(precision)}f}-"
+            f"{float(self.maxpt[i]):.{int(precision)}f}" for i in range(3)
         )
 
     def contains_bbox(self, other: "Bbox") -> bool:

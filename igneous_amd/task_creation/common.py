@@ -53,7 +53,11 @@ class FinelyDividedTaskIterator:
             pt = self.to_coord(i)
             offset = pt * self.shape + self.bounds.minpt
             yield self.task(self.shape.clone(), offset.clone())
-        self.on_finish()
+        # Under multi-rank dispatch every rank drains the iterator
+        # (dispatch.shard_tasks), so side effects like provenance writes
+        # must run once: rank 0 only (lost-update race otherwise).
+        if int(os.environ.get("RANK", "0")) == 0:
+            self.on_finish()
 
     def to_coord(self, index: int) -> Vec:
         gx, gy, gz = np.ceil(self.bounds.size3() / self.shape).astype(int)

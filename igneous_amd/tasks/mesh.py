@@ -175,7 +175,9 @@ class MeshTask(RegisteredTask):
         mesher = _get_mesher()
         # copy=False where supported: this task consumes (shifts, encodes,
         # uploads) every mesh before its next engine call, so zero-copy
-        # views into the engine's staging buffers are safe
+        # views into the engine's staging buffers are safe — EXCEPT under
+        # dry_run, where the meshes are handed back to the caller and must
+        # own their storage (the reference returns owned arrays).
         try:
             raw = mesher(
                 data,
@@ -183,7 +185,7 @@ class MeshTask(RegisteredTask):
                 reduction_factor=int(opts['simplification_factor'] or 0),
                 max_error=float(opts['max_simplification_error']),
                 voxel_centered=True,
-                copy=False,
+                copy=bool(opts['dry_run']),
             )
         except TypeError:
             raw = mesher(
@@ -308,11 +310,16 @@ class MeshTask(RegisteredTask):
             )
 
     def _upload_spatial_index(self, bbox: Bbox, mesh_bboxes: dict):
+        # mirrors mesh.py:452-464: filename precision comes from the mesh
+        # info's spatial_index metadata (cloudvolume's
+        # vol.mesh.spatial_index.precision); absent → integer naming.
         cf = CloudFiles(self.layer_path)
+        mesh_info = cf.get_json(f"{self._mesh_dir}/info") or {}
+        precision = (mesh_info.get('spatial_index') or {}).get('precision', None)
         resolution = np.asarray(self._volume.resolution, dtype=np.int64)
         nm_bbox = Bbox(bbox.minpt * resolution, bbox.maxpt * resolution)
         cf.put_json(
-            f"{self._mesh_dir}/{nm_bbox.to_filename()}.spatial",
+            f"{self._mesh_dir}/{nm_bbox.to_filename(precision)}.spatial",
             {str(k): v for k, v in mesh_bboxes.items()},
             compress=self.options['compress'],
         )
