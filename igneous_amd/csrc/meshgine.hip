@@ -174,6 +174,29 @@ struct GridDims {
   int64_t nseg;           // total segments
 };
 
+// XCD-aware segment schedule: block b runs on XCD b%8 (observed CDNA4
+// dispatch), so give each XCD one CONTIGUOUS 1/8 slab of the segment
+// range — adjacent rows (which share a voxel row) then stream through
+// the SAME XCD's L2 instead of being split round-robin across all 8
+// (measured 2.1x row re-read from HBM with the naive linear stride).
+struct SegSched {
+  int64_t beg, end, step;
+};
+
+__device__ __forceinline__ SegSched xcd_seg_sched(int64_t nseg,
+                                                  int waves_per_blk,
+                                                  int wave_in_blk) {
+  const int64_t xcd = blockIdx.x & 7;
+  const int64_t j = blockIdx.x >> 3;
+  const int64_t nbx = (gridDim.x + 7) >> 3;  // blocks per XCD class
+  const int64_t slab = (nseg + 7) >> 3;
+  SegSched ss;
+  ss.beg = xcd * slab + j * waves_per_blk + wave_in_blk;
+  ss.end = std::min<int64_t>((xcd + 1) * slab, nseg);
+  ss.step = nbx * waves_per_blk;
+  return ss;
+}
+
 // [1] count: one wave per segment; build label hash; write per-segment count
 template <typename T>
 __global__ void k_count(const T *__restrict__ labels, GridDims g,
@@ -186,9 +209,8 @@ __global__ void k_count(const T *__restrict__ labels, GridDims g,
   const int wave_in_blk = threadIdx.x / WAVE;
   const int waves_per_blk = blockDim.x / WAVE;
   const int64_t sxy = g.sx * g.sy;
-  for (int64_t seg = (int64_t)blockIdx.x * waves_per_blk + wave_in_blk;
-       seg < g.nseg;
-       seg += (int64_t)gridDim.x * waves_per_blk) {
+  const SegSched ss = xcd_seg_sched(g.nseg, waves_per_blk, wave_in_blk);
+  for (int64_t seg = ss.beg; seg < ss.end; seg += ss.step) {
     const int64_t row = seg / g.nsegx;
     const int64_t segx = seg - row * g.nsegx;
     const int64_t cy = row % g.ncy;
@@ -224,23 +246,23 @@ __global__ void k_count(const T *__restrict__ labels, GridDims g,
   }
 }
 
-// [3] emit: recompute, wave prefix, write triangle records in canonical
-// order. Per corner we emit the 32-bit WELD SLOT (edge axis, voxel, side)
-// — a complete key: an edge midpoint is a vertex only for its two
-// endpoint labels, the label id lives per-triangle in tri_label, and the
-// doubled coordinates are recoverable from the slot.
-template <typename T>
-__global__ void k_emit(const T *__restrict__ labels, GridDims g,
-                       const uint32_t *__restrict__ segoff, LabelHash lh,
-                       uint32_t *__restrict__ tri_label,
-                       uint4 *__restrict__ tri_recs) {
-  // all case tables staged in LDS: global-memory byte gathers on these
-  // small tables were the emit kernel's dominant cost (L1 line replays)
-  __shared__ uint8_t s_cnt[256];
-  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
-  __shared__ uint32_t s_comb[12];  // eoff*3 + axis, folded slot math
+// Compact 8-B triangle record: {cell_lin, mask | (t<<8)}. The three weld
+// slots (edge axis, voxel, side) and the doubled vertex coordinates are
+// all recomputable from it via the case tables — consumers decode instead
+// of re-reading a fat 16-B record (the record stream dominated the MC
+// pipeline's HBM traffic at 16 B/tri).
+//
+// slot = ((cell_lin*3 + comb[e]) << 1) | side, where comb[e] folds the
+// edge's voxel offset and axis; side = is the label the edge's UPPER
+// endpoint, baked into MC_TRI_PACK at table-gen time. Collision-free: a
+// midpoint is a vertex only for its two endpoint labels; 6*nvox < 2^32
+// enforced on the host.
+
+// stage the decode tables into LDS (s_pack: 256*MC_MAX_TRIS u16,
+// s_comb: 12 u32). Call before __syncthreads().
+__device__ __forceinline__ void stage_decode_tables(
+    uint16_t *s_pack, uint32_t *s_comb, int64_t sx, int64_t sxy) {
   for (int k = threadIdx.x; k < 256; k += blockDim.x) {
-    s_cnt[k] = MC_TRI_COUNT[k];
     #pragma unroll
     for (int t = 0; t < MC_MAX_TRIS; ++t)
       s_pack[k * MC_MAX_TRIS + t] = MC_TRI_PACK[k][t];
@@ -248,22 +270,45 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
   if (threadIdx.x < 12) {
     int e = threadIdx.x;
     uint32_t eoff = (uint32_t)(MC_EDGE_DOFF[e][0] >> 1) +
-                    (uint32_t)(MC_EDGE_DOFF[e][1] >> 1) * (uint32_t)g.sx +
-                    (uint32_t)(MC_EDGE_DOFF[e][2] >> 1) *
-                        (uint32_t)(g.sx * g.sy);
+                    (uint32_t)(MC_EDGE_DOFF[e][1] >> 1) * (uint32_t)sx +
+                    (uint32_t)(MC_EDGE_DOFF[e][2] >> 1) * (uint32_t)sxy;
     uint32_t axis = (MC_EDGE_DOFF[e][0] & 1)
                         ? 0u
                         : ((MC_EDGE_DOFF[e][1] & 1) ? 1u : 2u);
     s_comb[e] = eoff * 3u + axis;
   }
+}
+
+__device__ __forceinline__ void decode_rec_slots(
+    uint2 rec, const uint16_t *s_pack, const uint32_t *s_comb,
+    uint32_t s[3]) {
+  const uint32_t cl3 = rec.x * 3u;
+  const uint32_t pk =
+      s_pack[(rec.y & 255u) * MC_MAX_TRIS + (rec.y >> 8)];
+  s[0] = ((cl3 + s_comb[pk & 15]) << 1) | ((pk >> 4) & 1);
+  s[1] = ((cl3 + s_comb[(pk >> 5) & 15]) << 1) | ((pk >> 9) & 1);
+  s[2] = ((cl3 + s_comb[(pk >> 10) & 15]) << 1) | ((pk >> 14) & 1);
+}
+
+// [3] emit: recompute counts, wave prefix, write one 8-B record + the
+// 4-B label id per triangle in canonical global order.
+template <typename T>
+__global__ void k_emit(const T *__restrict__ labels, GridDims g,
+                       const uint32_t *__restrict__ segoff, LabelHash lh,
+                       uint32_t *__restrict__ tri_label,
+                       uint2 *__restrict__ tri_recs) {
+  // count table staged in LDS: global-memory byte gathers on the small
+  // case tables were the emit kernel's dominant cost (L1 line replays)
+  __shared__ uint8_t s_cnt[256];
+  for (int k = threadIdx.x; k < 256; k += blockDim.x)
+    s_cnt[k] = MC_TRI_COUNT[k];
   __syncthreads();
   const int lane = threadIdx.x & (WAVE - 1);
   const int wave_in_blk = threadIdx.x / WAVE;
   const int waves_per_blk = blockDim.x / WAVE;
   const int64_t sxy = g.sx * g.sy;
-  for (int64_t seg = (int64_t)blockIdx.x * waves_per_blk + wave_in_blk;
-       seg < g.nseg;
-       seg += (int64_t)gridDim.x * waves_per_blk) {
+  const SegSched ss = xcd_seg_sched(g.nseg, waves_per_blk, wave_in_blk);
+  for (int64_t seg = ss.beg; seg < ss.end; seg += ss.step) {
     const int64_t row = seg / g.nsegx;
     const int64_t segx = seg - row * g.nsegx;
     const int64_t cy = row % g.ncy;
@@ -281,6 +326,7 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
     uint32_t base = segoff[seg] + incl - cnt;
     if (active) {
       uint32_t pos = base;
+      const uint32_t cell_lin = (uint32_t)((cz * g.sy + cy) * g.sx + cx);
       #pragma unroll
       for (int i = 0; i < 8; ++i) {
         T L = c[i];
@@ -291,41 +337,17 @@ __global__ void k_emit(const T *__restrict__ labels, GridDims g,
         unsigned mask = 0;
         #pragma unroll
         for (int j = 0; j < 8; ++j) mask |= (c[j] == L) ? (1u << j) : 0u;
-        uint32_t nt = MC_TRI_COUNT[mask];
+        uint32_t nt = s_cnt[mask];
         if (!nt) continue;
         uint32_t lid = label_lookup(lh, (uint64_t)L);
-        // slot = ((cell_lin + eoff)*3 + axis) << 1 | side
-        //      = ((cell_lin*3 + s_comb[e]) << 1) | side
-        // (voxel-interleaved layout; side = is L the edge's UPPER
-        // endpoint label, baked into MC_TRI_PACK at table-gen time.
-        // Collision-free: a midpoint is a vertex only for its two
-        // endpoint labels. 6*nvox < 2^32 enforced on the host.)
-        const uint32_t cl3 =
-            (uint32_t)((cz * g.sy + cy) * g.sx + cx) * 3u;
         for (uint32_t t = 0; t < nt; ++t) {
-          uint32_t pk = s_pack[mask * MC_MAX_TRIS + t];
           tri_label[pos] = lid;
-          uint4 rec;
-          rec.x = ((cl3 + s_comb[pk & 15]) << 1) | ((pk >> 4) & 1);
-          rec.y = ((cl3 + s_comb[(pk >> 5) & 15]) << 1) | ((pk >> 9) & 1);
-          rec.z = ((cl3 + s_comb[(pk >> 10) & 15]) << 1) | ((pk >> 14) & 1);
-          rec.w = lid;
-          tri_recs[pos] = rec;  // one 16-B store per triangle
+          tri_recs[pos] = make_uint2(cell_lin, mask | (t << 8));
           ++pos;
         }
       }
     }
   }
-}
-
-// [4b] gather triangle records into label-partitioned order (16 B each)
-__global__ void k_gather_recs(const uint4 *__restrict__ tri_recs,
-                              const uint32_t *__restrict__ order,
-                              uint4 *__restrict__ recs_sorted,
-                              uint64_t ntris) {
-  uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= ntris) return;
-  recs_sorted[i] = tri_recs[order[i]];
 }
 
 // [4c] per-label triangle ranges (labels sorted, every id present)
@@ -365,10 +387,11 @@ __global__ void k_label_ranges(const uint32_t *__restrict__ lab_sorted,
 // atomic count roughly in half.
 template <int BLK, int TPT>
 __global__ __launch_bounds__(BLK) void k_weld_insert(
-    const uint4 *__restrict__ tri_recs,   // emit order
+    const uint2 *__restrict__ tri_recs,   // emit order (8-B records)
     const uint32_t *__restrict__ order,   // label partition permutation
-    uint4 *__restrict__ recs_sorted,      // gathered here (fused pass)
+    uint2 *__restrict__ recs_sorted,      // gathered here (fused pass)
     uint32_t *__restrict__ wminp,
+    int64_t sx, int64_t sxy,
     uint64_t ntris) {
   // LDS table sized for ~BLK*TPT*3 corners at ~0.4 load; wider windows
   // dedup more of a vertex's ~6 corner occurrences before the global
@@ -377,6 +400,9 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
   constexpr int LG = __builtin_ctz(WI_LDS_SLOTS);
   __shared__ uint32_t lkey[WI_LDS_SLOTS];
   __shared__ uint32_t lval[WI_LDS_SLOTS];
+  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
+  __shared__ uint32_t s_comb[12];
+  stage_decode_tables(s_pack, s_comb, sx, sxy);
   for (int k = threadIdx.x; k < WI_LDS_SLOTS; k += BLK) {
     lkey[k] = 0;
     lval[k] = 0;
@@ -387,9 +413,10 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
   for (int rep = 0; rep < TPT; ++rep) {
     uint64_t t = span0 + (uint64_t)rep * BLK + threadIdx.x;
     if (t >= ntris) break;
-    uint4 rec = tri_recs[order[t]];
+    uint2 rec = tri_recs[order[t]];
     recs_sorted[t] = rec;
-    const uint32_t s[3] = {rec.x, rec.y, rec.z};
+    uint32_t s[3];
+    decode_rec_slots(rec, s_pack, s_comb, s);
     #pragma unroll
     for (int v = 0; v < 3; ++v) {
       uint32_t slot = s[v];
@@ -421,10 +448,15 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
 // consecutive corners and ballots the flags into one u64 word. Vertex
 // ids then come from a word-granular scan + popcount instead of a full
 // per-element scan array.
-__global__ void k_weld_flag_bits(const uint4 *__restrict__ recs_sorted,
+__global__ void k_weld_flag_bits(const uint2 *__restrict__ recs_sorted,
                                  const uint32_t *__restrict__ wminp,
                                  unsigned long long *__restrict__ bits,
+                                 int64_t sx, int64_t sxy,
                                  uint64_t ncorners, uint64_t nwords) {
+  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
+  __shared__ uint32_t s_comb[12];
+  stage_decode_tables(s_pack, s_comb, sx, sxy);
+  __syncthreads();
   uint64_t gid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   uint64_t word = gid / 64;
   if (word >= nwords) return;
@@ -432,9 +464,9 @@ __global__ void k_weld_flag_bits(const uint4 *__restrict__ recs_sorted,
   uint64_t i = word * 64 + lane;
   bool flag = false;
   if (i < ncorners) {
-    uint4 rec = recs_sorted[i / 3];
-    uint32_t slot = (i % 3 == 0) ? rec.x : ((i % 3 == 1) ? rec.y : rec.z);
-    flag = (wminp[slot] == ~(uint32_t)i);
+    uint32_t s[3];
+    decode_rec_slots(recs_sorted[i / 3], s_pack, s_comb, s);
+    flag = (wminp[s[i % 3]] == ~(uint32_t)i);
   }
   unsigned long long m = __ballot(flag);
   if (lane == 0) bits[word] = m;
@@ -462,8 +494,8 @@ __global__ void k_total_verts(const uint32_t *__restrict__ wscan,
 }
 
 // [5d] first occurrences: record vertex id in the table, write the vertex
-// (doubled coordinates decoded from the slot: voxel, axis, +1 on axis)
-__global__ void k_weld_verts(const uint4 *__restrict__ recs_sorted,
+// (doubled coordinates = 2*cell + edge offset, decoded from the record)
+__global__ void k_weld_verts(const uint2 *__restrict__ recs_sorted,
                              const uint32_t *__restrict__ wscan,
                              const unsigned long long *__restrict__ bits,
                              const uint32_t *__restrict__ wminp,
@@ -472,27 +504,38 @@ __global__ void k_weld_verts(const uint4 *__restrict__ recs_sorted,
                              uint32_t usx, uint32_t usxy,
                              float rx, float ry, float rz, float shift,
                              uint64_t ntris) {
+  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
+  __shared__ uint32_t s_comb[12];
+  __shared__ uint8_t s_doff[12][3];
+  stage_decode_tables(s_pack, s_comb, usx, usxy);
+  if (threadIdx.x < 12) {
+    #pragma unroll
+    for (int k = 0; k < 3; ++k)
+      s_doff[threadIdx.x][k] = MC_EDGE_DOFF[threadIdx.x][k];
+  }
+  __syncthreads();
   uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= ntris) return;
-  uint4 rec = recs_sorted[t];
-  const uint32_t s[3] = {rec.x, rec.y, rec.z};
+  uint2 rec = recs_sorted[t];
+  const uint32_t cl3 = rec.x * 3u;
+  const uint32_t pk =
+      s_pack[(rec.y & 255u) * MC_MAX_TRIS + (rec.y >> 8)];
+  uint32_t cz = rec.x / usxy;
+  uint32_t rem = rec.x - cz * usxy;
+  uint32_t cy = rem / usx;
+  uint32_t cx = rem - cy * usx;
   #pragma unroll
   for (int v = 0; v < 3; ++v) {
+    uint32_t nib = (pk >> (5 * v)) & 31u;  // 4-bit edge + 1-bit side
+    uint32_t e = nib & 15u;
+    uint32_t slot = ((cl3 + s_comb[e]) << 1) | (nib >> 4);
     uint32_t i = (uint32_t)(3 * t + v);
-    uint32_t slot = s[v];
     if (wminp[slot] != ~i) continue;
     uint32_t vid = vtx_id_of(wscan, bits, i);
     wvtx[slot] = vid;
-    uint32_t eslot = slot >> 1;
-    uint32_t axis = eslot % 3u;
-    uint32_t lin = eslot / 3u;
-    uint32_t vz = lin / usxy;
-    uint32_t rem = lin - vz * usxy;
-    uint32_t vy = rem / usx;
-    uint32_t vx = rem - vy * usx;
-    float dx = (float)(2 * vx + (axis == 0));
-    float dy = (float)(2 * vy + (axis == 1));
-    float dz = (float)(2 * vz + (axis == 2));
+    float dx = (float)(2 * cx + s_doff[e][0]);
+    float dy = (float)(2 * cy + s_doff[e][1]);
+    float dz = (float)(2 * cz + s_doff[e][2]);
     verts[3ull * vid + 0] = (0.5f * dx + shift) * rx;
     verts[3ull * vid + 1] = (0.5f * dy + shift) * ry;
     verts[3ull * vid + 2] = (0.5f * dz + shift) * rz;
@@ -511,20 +554,27 @@ __global__ void k_vbase(const uint32_t *__restrict__ tri_off,
   else vbase[l] = vtx_id_of(wscan, bits, 3ull * tri_off[l]);
 }
 
-// [5f] faces: per-label local vertex indices (the record carries its
-// label id in .w)
-__global__ void k_faces(const uint4 *__restrict__ recs_sorted,
+// [5f] faces: per-label local vertex indices (label id from the sorted
+// label array — the partition sort's key output)
+__global__ void k_faces(const uint2 *__restrict__ recs_sorted,
+                        const uint32_t *__restrict__ lab_sorted,
                         const uint32_t *__restrict__ wvtx,
                         const uint32_t *__restrict__ vbase,
                         uint32_t *__restrict__ faces,
+                        int64_t sx, int64_t sxy,
                         uint64_t ntris) {
+  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
+  __shared__ uint32_t s_comb[12];
+  stage_decode_tables(s_pack, s_comb, sx, sxy);
+  __syncthreads();
   uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= ntris) return;
-  uint4 rec = recs_sorted[t];
-  uint32_t base = vbase[rec.w];
-  faces[3 * t + 0] = wvtx[rec.x] - base;
-  faces[3 * t + 1] = wvtx[rec.y] - base;
-  faces[3 * t + 2] = wvtx[rec.z] - base;
+  uint32_t s[3];
+  decode_rec_slots(recs_sorted[t], s_pack, s_comb, s);
+  uint32_t base = vbase[lab_sorted[t]];
+  faces[3 * t + 0] = wvtx[s[0]] - base;
+  faces[3 * t + 1] = wvtx[s[1]] - base;
+  faces[3 * t + 2] = wvtx[s[2]] - base;
 }
 
 __global__ void k_iota(uint32_t *p, uint64_t n) {
@@ -576,7 +626,7 @@ struct mg_ctx {
       simp_Q, simp_pick, simp_remap, simp_flab, simp_flab_alt,
       simp_faces_alt, simp_verts_alt, simp_vbase_alt, simp_meta, simp_ref,
       simp_keep, simp_keep_scan, simp_park, simp_first, simp_troff_final,
-      simp_deg, simp_adj;
+      simp_deg, simp_adj, simp_rh;
   uint64_t lh_slots = 1ull << 20;
   HostBuf h_verts, h_faces;  // pinned output staging, reused across calls
   hipEvent_t ev[16] = {};
@@ -682,7 +732,8 @@ void mg_destroy(mg_ctx *c) {
                   &c->simp_faces_alt, &c->simp_verts_alt, &c->simp_vbase_alt,
                   &c->simp_meta, &c->simp_ref, &c->simp_keep,
                   &c->simp_keep_scan, &c->simp_park, &c->simp_first,
-                  &c->simp_troff_final, &c->simp_deg, &c->simp_adj}) {
+                  &c->simp_troff_final, &c->simp_deg, &c->simp_adj,
+                  &c->simp_rh}) {
     if (b->ptr) (void)hipFree(b->ptr);
   }
   if (c->h_verts.ptr) (void)hipHostFree(c->h_verts.ptr);
@@ -759,6 +810,7 @@ static double ev_ms(mg_ctx *c, int a, int b) {
 // oracle/simplify.c round for round. Updates faces/verts/vbase/tri_off
 // in the ctx; p_T/p_V become the post-simplification totals.
 static int run_simplify(mg_ctx *c, uint32_t nlabels,
+                        const uint32_t *lab_sorted,
                         uint32_t reduction_factor, float max_error,
                         uint64_t *p_T, uint64_t *p_V) {
   hipStream_t s = c->stream;
@@ -807,7 +859,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   {
     uint64_t nbt = (T + blk - 1) / blk;
     hipLaunchKernelGGL(k_globalize_faces, dim3((uint32_t)nbt), dim3(blk), 0,
-                       s, faces_g, (const uint4 *)c->keys_sorted.ptr,
+                       s, faces_g, lab_sorted,
                        (const uint32_t *)c->vbase.ptr,
                        (uint32_t *)c->simp_flab.ptr, T);
     uint32_t nbl = (uint32_t)((L + 255) / 256);
@@ -831,6 +883,13 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   if (getenv("MG_SIMP_PROF"))
     HIP_TRY(c, hipMemsetAsync((unsigned long long *)c->lh_misc.ptr + 8, 0,
                               48, s), 40);
+  // round-count histogram (MG_SIMP_ROUNDHIST=1): groups/subs per label
+  uint32_t *d_rh = nullptr;
+  if (getenv("MG_SIMP_ROUNDHIST")) {
+    if (ensure(c, c->simp_rh, 160 * 4)) return 40;
+    d_rh = (uint32_t *)c->simp_rh.ptr;
+    HIP_TRY(c, hipMemsetAsync(d_rh, 0, 160 * 4, s), 40);
+  }
   {
     // default OFF since sub-round groups: the +24 KB LDS CSR payload
     // costs blocks/CU and the recompute phases it serves are now ~25%
@@ -878,6 +937,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                        getenv("MG_SIMP_PROF")
                            ? (unsigned long long *)c->lh_misc.ptr + 8
                            : nullptr,
+                       d_rh,
                        max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, 0u, small_cap);
       HIP_TRY(c, hipGetLastError(), 40);
     }
@@ -898,6 +958,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                        getenv("MG_SIMP_PROF")
                            ? (unsigned long long *)c->lh_misc.ptr + 8
                            : nullptr,
+                       d_rh,
                        max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, small_cap, 0xFFFFFFFFu);
     HIP_TRY(c, hipGetLastError(), 40);
     if (getenv("MG_SIMP_PROF")) {
@@ -910,6 +971,21 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       fprintf(stderr, "[mg simp phases, summed tid0 cycles]");
       for (int k = 0; k < 6; ++k)
         fprintf(stderr, " %s=%llu", nm[k], hp[k]);
+      fprintf(stderr, "\n");
+    }
+    if (d_rh) {
+      uint32_t h[160];
+      HIP_TRY(c, hipMemcpyAsync(h, d_rh, 160 * 4, hipMemcpyDeviceToHost, s),
+              40);
+      HIP_TRY(c, hipStreamSynchronize(s), 40);
+      fprintf(stderr, "[mg simp rounds] labels=%u sum_groups=%u "
+              "sum_subs=%u sum_nt0=%u\n", h[130], h[128], h[129], h[131]);
+      fprintf(stderr, "[mg simp group hist]");
+      for (int k = 0; k < 64; ++k)
+        if (h[k]) fprintf(stderr, " %d:%u", k, h[k]);
+      fprintf(stderr, "\n[mg simp subs hist]");
+      for (int k = 0; k < 64; ++k)
+        if (h[64 + k]) fprintf(stderr, " %d:%u", k, h[64 + k]);
       fprintf(stderr, "\n");
     }
   }
@@ -1382,24 +1458,25 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
 
   // [3] emit
   if (ensure(c, c->tri_label, T * 4)) return 18;
-  if (ensure(c, c->tri_keys, T * 16)) return 18;
+  if (ensure(c, c->tri_keys, T * 8)) return 18;
   {
     int blk = 256;
     int waves_per_blk = blk / WAVE;
     int64_t nb = std::min<int64_t>((g.nseg + waves_per_blk - 1) / waves_per_blk,
                                    8192);
+    nb = (nb + 7) & ~7ll;  // multiple of 8: XCD slab schedule coverage
     if (dtype == MG_U64)
       hipLaunchKernelGGL(k_emit<uint64_t>, dim3((uint32_t)nb), dim3(blk), 0, s,
                          (const uint64_t *)c->labels.ptr, g,
                          (const uint32_t *)c->segoff.ptr, lh,
                          (uint32_t *)c->tri_label.ptr,
-                         (uint4 *)c->tri_keys.ptr);
+                         (uint2 *)c->tri_keys.ptr);
     else
       hipLaunchKernelGGL(k_emit<uint32_t>, dim3((uint32_t)nb), dim3(blk), 0, s,
                          (const uint32_t *)c->labels.ptr, g,
                          (const uint32_t *)c->segoff.ptr, lh,
                          (uint32_t *)c->tri_label.ptr,
-                         (uint4 *)c->tri_keys.ptr);
+                         (uint2 *)c->tri_keys.ptr);
   }
   HIP_TRY(c, hipGetLastError(), 18);
   HIP_TRY(c, hipEventRecord(c->ev[4], s), 18);
@@ -1408,8 +1485,9 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   if (ensure(c, c->order, T * 4)) return 19;
   if (ensure(c, c->order_alt, T * 4)) return 19;
   if (ensure(c, c->tri_label_alt, T * 4)) return 19;
-  if (ensure(c, c->keys_sorted, T * 16)) return 19;
+  if (ensure(c, c->keys_sorted, T * 8)) return 19;
   uint32_t *order_sorted = nullptr;
+  uint32_t *lab_sorted = nullptr;  // sort key output: label id per tri
   {
     int blk = 256;
     uint64_t nb = (T + blk - 1) / blk;
@@ -1431,8 +1509,9 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     e = rocprim::radix_sort_pairs(
         c->sort_tmp.ptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
     if (e != hipSuccess) { SET_ERR(c, "radix_sort failed"); return 19; }
-    // (the 16-B record gather is fused into k_weld_insert below)
+    // (the record gather is fused into k_weld_insert below)
     order_sorted = d_vals.current();
+    lab_sorted = d_keys.current();
     // label ranges
     if (ensure(c, c->tri_off, ((uint64_t)nlabels + 1) * 4)) return 19;
     hipLaunchKernelGGL(k_label_ranges, dim3((uint32_t)nb), dim3(blk), 0, s,
@@ -1457,29 +1536,30 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0, wslots * 4, s), 20);
   uint32_t *wminp = (uint32_t *)c->wh_keys.ptr;
   uint32_t *wvtx = (uint32_t *)c->wh_vtx.ptr;
-  const uint4 *recs_sorted = (const uint4 *)c->keys_sorted.ptr;
+  const uint2 *recs_sorted = (const uint2 *)c->keys_sorted.ptr;
   {
     int blk = 256;
     uint64_t nbt = (T + blk - 1) / blk;
     int wi_cfg = 1;  // 0: 1024x1, 1: 1024x2, 2: 1024x4
     if (const char *e = getenv("MG_WELD_INSERT_CFG")) wi_cfg = atoi(e);
-    uint4 *rs_mut = (uint4 *)c->keys_sorted.ptr;
-    const uint4 *tr = (const uint4 *)c->tri_keys.ptr;
+    uint2 *rs_mut = (uint2 *)c->keys_sorted.ptr;
+    const uint2 *tr = (const uint2 *)c->tri_keys.ptr;
+    const int64_t sxy64 = g.sx * g.sy;
     if (wi_cfg == 0) {
       uint64_t nb2 = (T + 1023) / 1024;
       hipLaunchKernelGGL((k_weld_insert<1024, 1>), dim3((uint32_t)nb2),
                          dim3(1024), 0, s, tr, order_sorted, rs_mut, wminp,
-                         T);
+                         g.sx, sxy64, T);
     } else if (wi_cfg == 1) {
       uint64_t nb2 = (T + 2047) / 2048;
       hipLaunchKernelGGL((k_weld_insert<1024, 2>), dim3((uint32_t)nb2),
                          dim3(1024), 0, s, tr, order_sorted, rs_mut, wminp,
-                         T);
+                         g.sx, sxy64, T);
     } else {
       uint64_t nb2 = (T + 4095) / 4096;
       hipLaunchKernelGGL((k_weld_insert<1024, 4>), dim3((uint32_t)nb2),
                          dim3(1024), 0, s, tr, order_sorted, rs_mut, wminp,
-                         T);
+                         g.sx, sxy64, T);
     }
     // first-occurrence flags as a bit array + word-granular scan
     const uint64_t nwords = (NC + 63) / 64;
@@ -1490,7 +1570,8 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     {
       uint64_t nbw = (nwords * 64 + blk - 1) / blk;
       hipLaunchKernelGGL(k_weld_flag_bits, dim3((uint32_t)nbw), dim3(blk),
-                         0, s, recs_sorted, wminp, bits, NC, nwords);
+                         0, s, recs_sorted, wminp, bits, g.sx, sxy64,
+                         NC, nwords);
     }
     auto it = rocprim::make_transform_iterator(bits, PopcWord{});
     size_t tmp_bytes = 0;
@@ -1536,9 +1617,9 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                        wscan, bits,
                        (uint32_t *)c->vbase.ptr, nlabels, total_verts);
     hipLaunchKernelGGL(k_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
-                       recs_sorted, wvtx,
+                       recs_sorted, lab_sorted, wvtx,
                        (const uint32_t *)c->vbase.ptr,
-                       (uint32_t *)c->faces.ptr, T);
+                       (uint32_t *)c->faces.ptr, g.sx, g.sx * g.sy, T);
   }
   HIP_TRY(c, hipGetLastError(), 22);
   HIP_TRY(c, hipEventRecord(c->ev[6], s), 22);
@@ -1546,8 +1627,8 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   // [6] per-label quadric simplification (mesh.py:376-381 semantics)
   uint64_t Tcur = T, Vcur = total_verts;
   if (reduction_factor > 1 && T > 0) {
-    int rc = run_simplify(c, nlabels, reduction_factor, max_error,
-                          &Tcur, &Vcur);
+    int rc = run_simplify(c, nlabels, lab_sorted, reduction_factor,
+                          max_error, &Tcur, &Vcur);
     if (rc) return rc;
   }
   HIP_TRY(c, hipEventRecord(c->ev[8], s), 22);
@@ -1641,6 +1722,7 @@ static int run_count_emit(mg_ctx *c, const T *d_labels, const GridDims &g,
   int waves_per_blk = blk / WAVE;
   int64_t nb = std::min<int64_t>(
       (g.nseg + waves_per_blk - 1) / waves_per_blk, 8192);
+  nb = (nb + 7) & ~7ll;  // multiple of 8: XCD slab schedule coverage
   hipLaunchKernelGGL(k_count<T>, dim3((uint32_t)nb), dim3(blk), 0, s,
                      d_labels, g, d_segcnt, lh);
   HIP_TRY(c, hipGetLastError(), 30);

@@ -268,13 +268,13 @@ __global__ void k_init_simplify(const uint32_t *__restrict__ tri_off,
 
 // faces local -> global vertex ids (vbase offset per label)
 __global__ void k_globalize_faces(uint32_t *__restrict__ faces,
-                                  const uint4 *__restrict__ recs_sorted,
+                                  const uint32_t *__restrict__ lab_sorted,
                                   const uint32_t *__restrict__ vbase,
                                   uint32_t *__restrict__ flab,
                                   uint64_t ntris) {
   uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= ntris) return;
-  uint32_t lab = recs_sorted[t].w;
+  uint32_t lab = lab_sorted[t];
   flab[t] = lab;
   uint32_t base = vbase[lab];
   faces[3*t] += base; faces[3*t+1] += base; faces[3*t+2] += base;
@@ -510,6 +510,7 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
     uint8_t *__restrict__ active,
     uint32_t *__restrict__ park_faces,
     unsigned long long *__restrict__ prof,  // 6 phase counters or null
+    uint32_t *__restrict__ roundhist,  // 160 u32 or null (see host)
     float max_cost, uint32_t nlabels, uint32_t big_cap,
     uint32_t subs, uint32_t nv_lo, uint32_t nv_hi) {
   const uint32_t b = blockIdx.x;
@@ -576,9 +577,11 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
     t_last = now;                                              \
   }
 
+  uint32_t n_groups = 0, n_subs = 0;
   for (int round = 0; round < 65536; ++round) {
     uint32_t nt = s_nt;
     if (nt <= tgt) break;
+    ++n_groups;
     const uint32_t nt_group = nt;  // group-progress watermark
     const bool clmode = CLLDS && (nt <= CAPF);
     PHASE_MARK(0)  // loop head
@@ -709,6 +712,7 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
     for (uint32_t sub = 0; sub < subs; ++sub) {
     nt = s_nt;
     if (nt <= tgt) break;
+    ++n_subs;
     if (sub > 0) {
       for (uint32_t v = tid; v < nv; v += BS) pick_l[v] = ~0ull;
       __syncthreads();
@@ -792,5 +796,16 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
   if (tid == 0) {
     nt_cur[b] = s_nt;
     active[b] = 0;  // done: global rounds skip this label
+    if (roundhist) {
+      // layout: [0..63] group-count hist, [64..127] sub-count hist
+      // (capped), [128] sum groups, [129] sum subs, [130] labels seen,
+      // [131] sum nt0 (faces entering), [132..159] spare
+      atomicAdd(&roundhist[n_groups < 64 ? n_groups : 63], 1u);
+      atomicAdd(&roundhist[64 + (n_subs < 64 ? n_subs : 63)], 1u);
+      atomicAdd(&roundhist[128], n_groups);
+      atomicAdd(&roundhist[129], n_subs);
+      atomicAdd(&roundhist[130], 1u);
+      atomicAdd(&roundhist[131], nt0);
+    }
   }
 }
