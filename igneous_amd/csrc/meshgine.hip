@@ -279,6 +279,15 @@ __device__ __forceinline__ void stage_decode_tables(
   }
 }
 
+// decode one corner's weld slot (v in 0..2)
+__device__ __forceinline__ uint32_t decode_rec_slot1(
+    uint2 rec, const uint16_t *s_pack, const uint32_t *s_comb, uint32_t v) {
+  const uint32_t pk =
+      s_pack[(rec.y & 255u) * MC_MAX_TRIS + (rec.y >> 8)];
+  const uint32_t nib = (pk >> (5u * v)) & 31u;
+  return ((rec.x * 3u + s_comb[nib & 15u]) << 1) | (nib >> 4);
+}
+
 __device__ __forceinline__ void decode_rec_slots(
     uint2 rec, const uint16_t *s_pack, const uint32_t *s_comb,
     uint32_t s[3]) {
@@ -464,9 +473,9 @@ __global__ void k_weld_flag_bits(const uint2 *__restrict__ recs_sorted,
   uint64_t i = word * 64 + lane;
   bool flag = false;
   if (i < ncorners) {
-    uint32_t s[3];
-    decode_rec_slots(recs_sorted[i / 3], s_pack, s_comb, s);
-    flag = (wminp[s[i % 3]] == ~(uint32_t)i);
+    uint32_t slot = decode_rec_slot1(recs_sorted[i / 3], s_pack, s_comb,
+                                     (uint32_t)(i % 3));
+    flag = (wminp[slot] == ~(uint32_t)i);
   }
   unsigned long long m = __ballot(flag);
   if (lane == 0) bits[word] = m;
@@ -841,7 +850,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   if (ensure(c, c->simp_pk_alt, 3 * T * 4)) return 40;
   if (ensure(c, c->simp_pv, 3 * T * 4)) return 40;
   if (ensure(c, c->simp_pv_alt, 3 * T * 4)) return 40;
-  if (ensure(c, c->simp_Q, V * 40)) return 40;
+  if (ensure(c, c->simp_Q, V * 48)) return 40;  // 12-float rows
   if (ensure(c, c->simp_pick, V * 8)) return 40;
   if (ensure(c, c->simp_remap, V * 4)) return 40;
   if (ensure(c, c->simp_keep, T * 4)) return 40;

@@ -110,8 +110,10 @@ __global__ void k_accum_quadrics(const uint32_t *__restrict__ pk,
     SimpPlane p = fq[f];
     sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
   }
-  #pragma unroll
-  for (int k = 0; k < 10; ++k) Q[10ull*v + k] = q[k];
+  float4 *qo = (float4 *)(Q + 12ull*v);
+  qo[0] = make_float4(q[0], q[1], q[2], q[3]);
+  qo[1] = make_float4(q[4], q[5], q[6], q[7]);
+  qo[2] = make_float4(q[8], q[9], 0.0f, 0.0f);
 }
 
 // [S4] per-vertex cheapest incident edge (cost-bits<<32 | peer, min)
@@ -134,8 +136,11 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
   float cq[3][10], cp[3][3];
   #pragma unroll
   for (int ci = 0; ci < 3; ++ci) {
-    #pragma unroll
-    for (int k = 0; k < 10; ++k) cq[ci][k] = Q[10ull*fc[ci] + k];
+    const float4 *qp = (const float4 *)(Q + 12ull*fc[ci]);
+    float4 a = qp[0], b = qp[1], cc = qp[2];
+    cq[ci][0] = a.x; cq[ci][1] = a.y; cq[ci][2] = a.z; cq[ci][3] = a.w;
+    cq[ci][4] = b.x; cq[ci][5] = b.y; cq[ci][6] = b.z; cq[ci][7] = b.w;
+    cq[ci][8] = cc.x; cq[ci][9] = cc.y;
     #pragma unroll
     for (int k = 0; k < 3; ++k) cp[ci][k] = verts[3ull*fc[ci] + k];
   }
@@ -189,7 +194,7 @@ __global__ void k_collapse(const unsigned long long *__restrict__ pick,
   verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
   verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
   #pragma unroll
-  for (int k = 0; k < 10; ++k) Q[10*u + k] += Q[10*(uint64_t)w + k];
+  for (int k = 0; k < 10; ++k) Q[12*u + k] += Q[12*(uint64_t)w + k];
   remap[w] = (uint32_t)u;
 }
 
@@ -417,11 +422,44 @@ __global__ void k_gather_final(const uint32_t *__restrict__ park_faces,
 // the sorted CSR, jittered pick encoding on label-local ids, mutual-pick
 // matched collapse, stable compaction, same termination conditions).
 
+// wave-shuffle prefix machinery shared by the block-wide helpers below:
+// each thread sums its contiguous chunk, a 64-lane shuffle scan orders
+// the per-thread partials within the wave (no LDS), ONE barrier shares
+// the NW wave totals, and every thread derives its global offset from
+// the (tiny) wave-total array. 2 barriers per helper call instead of the
+// 16+ of the old Hillis-Steele block scan — the round loop's barrier
+// count was the per-label kernel's dominant WAIT source.
+template <int BS>
+__device__ __forceinline__ uint32_t blk_prefix(uint32_t sum,
+                                               uint32_t *s_wsum /*NW*/,
+                                               uint32_t *p_total) {
+  constexpr int NW = BS / 64;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  uint32_t incl = sum;
+  #pragma unroll
+  for (int d = 1; d < 64; d <<= 1) {
+    uint32_t y = __shfl_up(incl, d, 64);
+    if (lane >= d) incl += y;
+  }
+  if (lane == 63) s_wsum[wid] = incl;
+  __syncthreads();
+  uint32_t wbase = 0, total = 0;
+  #pragma unroll
+  for (int w = 0; w < NW; ++w) {
+    uint32_t x = s_wsum[w];
+    if (w < wid) wbase += x;
+    total += x;
+  }
+  *p_total = total;
+  return wbase + incl - sum;  // exclusive prefix of this thread's sum
+}
+
 // block-wide exclusive scan of src[0..n) into dst (+base), returns total
 template <int BS>
 __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
                                uint32_t n, uint32_t base,
-                               uint32_t *s_sums /*BS+1*/,
+                               uint32_t *s_sums /*>= BS/64*/,
                                uint32_t *dst2 = nullptr) {
   const uint32_t tid = threadIdx.x;
   const uint32_t chunk = (n + BS - 1) / BS;
@@ -429,19 +467,8 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
   const uint32_t hi = lo + chunk < n ? lo + chunk : n;
   uint32_t sum = 0;
   for (uint32_t i = lo; i < hi; ++i) sum += src[i];
-  s_sums[tid] = sum;
-  __syncthreads();
-  // Hillis-Steele inclusive scan over the 256 partials (log steps)
-  #pragma unroll
-  for (uint32_t st = 1; st < BS; st <<= 1) {
-    uint32_t x = (tid >= st) ? s_sums[tid - st] : 0;
-    __syncthreads();
-    s_sums[tid] += x;
-    __syncthreads();
-  }
-  uint32_t excl = s_sums[tid] - sum;      // exclusive from inclusive
-  if (tid == BS - 1) s_sums[BS] = s_sums[BS - 1];
-  __syncthreads();
+  uint32_t total;
+  uint32_t excl = blk_prefix<BS>(sum, s_sums, &total);
   uint32_t run = base + excl;
   for (uint32_t i = lo; i < hi; ++i) {
     uint32_t t = src[i];
@@ -449,44 +476,46 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
     if (dst2) dst2[i] = run;  // optional cursor copy (src may alias
     run += t;                 // dst2: t was read first)
   }
-  uint32_t total = s_sums[BS];
-  __syncthreads();
+  __syncthreads();  // s_sums reusable after this
   return total;
 }
 
-// block-wide stable compaction of kept faces fa->fb (single pass over
-// the chunked keep flags after the partial-sum scan; the scatter rides
-// the same loop that would have written scan offsets)
+// fused face rewrite + stable compaction: remap corners through the
+// label-local rm[] (LDS), keep-flag in a register bitmask (no valid[]
+// array traffic), scan via blk_prefix, scatter kept faces fa->fb.
+// Identical output order to the old rewrite + compact pair.
 template <int BS>
-__device__ uint32_t blk_compact_faces(const uint32_t *__restrict__ fa,
-                                      uint32_t *__restrict__ fb,
-                                      const uint8_t *__restrict__ valid,
-                                      uint32_t n, uint32_t *s_sums) {
+__device__ uint32_t blk_rewrite_compact(uint32_t *__restrict__ fa,
+                                        uint32_t *__restrict__ fb,
+                                        const uint32_t *__restrict__ rm,
+                                        uint32_t v0,
+                                        uint32_t n, uint32_t *s_sums) {
   const uint32_t tid = threadIdx.x;
   const uint32_t chunk = (n + BS - 1) / BS;
   const uint32_t lo = tid * chunk;
   const uint32_t hi = lo + chunk < n ? lo + chunk : n;
+  // chunk <= big_cap/BS; bitmask register array sized for big_cap=65536
+  constexpr uint32_t MAXW = (65536u / BS + 63) / 64 + 1;
+  unsigned long long bm[MAXW] = {};
   uint32_t sum = 0;
-  for (uint32_t i = lo; i < hi; ++i) sum += valid[i];
-  s_sums[tid] = sum;
-  __syncthreads();
-  #pragma unroll
-  for (uint32_t st = 1; st < BS; st <<= 1) {
-    uint32_t x = (tid >= st) ? s_sums[tid - st] : 0;
-    __syncthreads();
-    s_sums[tid] += x;
-    __syncthreads();
+  for (uint32_t i = lo; i < hi; ++i) {
+    uint32_t i0 = v0 + rm[fa[3*i] - v0];
+    uint32_t i1 = v0 + rm[fa[3*i+1] - v0];
+    uint32_t i2 = v0 + rm[fa[3*i+2] - v0];
+    fa[3*i] = i0; fa[3*i+1] = i1; fa[3*i+2] = i2;
+    if (i0 != i1 && i1 != i2 && i0 != i2) {
+      bm[(i - lo) >> 6] |= 1ull << ((i - lo) & 63);
+      ++sum;
+    }
   }
-  uint32_t run = s_sums[tid] - sum;
-  if (tid == BS - 1) s_sums[BS] = s_sums[BS - 1];
-  __syncthreads();
+  uint32_t total;
+  uint32_t run = blk_prefix<BS>(sum, s_sums, &total);
   for (uint32_t i = lo; i < hi; ++i)
-    if (valid[i]) {
+    if (bm[(i - lo) >> 6] & (1ull << ((i - lo) & 63))) {
       fb[3*run] = fa[3*i]; fb[3*run+1] = fa[3*i+1]; fb[3*run+2] = fa[3*i+2];
       ++run;
     }
-  uint32_t total = s_sums[BS];
-  __syncthreads();
+  __syncthreads();  // s_sums reusable after this
   return total;
 }
 
@@ -534,11 +563,10 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
   uint32_t *faces = faces_g + 3ull * f0;
   uint32_t *ftmp = faces_tmp + 3ull * f0;
   SimpPlane *pl = fq + f0;
-  uint8_t *valid = fvalid + f0;
   uint32_t *aoff = adj_off + v0;
   uint32_t *cl = cols + 3ull * f0;
 
-  __shared__ uint32_t s_sums[BS + 1];
+  __shared__ uint32_t s_sums[BS / 64];  // wave totals for blk_prefix
   __shared__ uint32_t s_nt, s_collapses;
   // the two ATOMIC-hot per-vertex arrays live in LDS when the label fits:
   // neighboring faces' vertices share cache lines, so the global
@@ -700,8 +728,12 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
           sq_add_plane(q, p.nx, p.ny, p.nz, p.d, 1.0f);
         }
       }
-      #pragma unroll
-      for (int k = 0; k < 10; ++k) Q[10ull*(v0+v) + k] = q[k];
+      {  // 12-float row (2 pad zeros), 3 dwordx4 stores
+        float4 *qo = (float4 *)(Q + 12ull*(v0+v));
+        qo[0] = make_float4(q[0], q[1], q[2], q[3]);
+        qo[1] = make_float4(q[4], q[5], q[6], q[7]);
+        qo[2] = make_float4(q[8], q[9], 0.0f, 0.0f);
+      }
       pick_l[v] = ~0ull;  // fused pick reset (same-thread slot)
     }
     __syncthreads();
@@ -717,7 +749,8 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
       for (uint32_t v = tid; v < nv; v += BS) pick_l[v] = ~0ull;
       __syncthreads();
     }
-    // [6] picks (oracle step 2)
+    // [6] picks (oracle step 2). Q rows are 12 floats (48 B, 16-B
+    // aligned) so the two quadric gathers are 3 dwordx4 loads each.
     for (uint32_t f = tid; f < nt; f += BS) {
       uint32_t fc[3] = {fa[3*f], fa[3*f+1], fa[3*f+2]};
       #pragma unroll
@@ -728,10 +761,14 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
         float mx = 0.5f*(verts[3ull*u]+verts[3ull*w]);
         float my = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
         float mz = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
-        float S[10];
-        #pragma unroll
-        for (int k = 0; k < 10; ++k)
-          S[k] = Q[10ull*u + k] + Q[10ull*w + k];
+        const float4 *qu = (const float4 *)(Q + 12ull*u);
+        const float4 *qw = (const float4 *)(Q + 12ull*w);
+        float4 su0 = qu[0], su1 = qu[1], su2 = qu[2];
+        float4 sw0 = qw[0], sw1 = qw[1], sw2 = qw[2];
+        float S[10] = {su0.x + sw0.x, su0.y + sw0.y, su0.z + sw0.z,
+                       su0.w + sw0.w, su1.x + sw1.x, su1.y + sw1.y,
+                       su1.z + sw1.z, su1.w + sw1.w, su2.x + sw2.x,
+                       su2.y + sw2.y};
         float cost = sq_eval(S, mx, my, mz);
         if (cost < 0.0f) cost = 0.0f;
         if (cost > max_cost) continue;
@@ -761,21 +798,14 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
       verts[3ull*u+1] = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
       verts[3ull*u+2] = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
       #pragma unroll
-      for (int k = 0; k < 10; ++k) Q[10ull*u + k] += Q[10ull*w + k];
+      for (int k = 0; k < 10; ++k) Q[12ull*u + k] += Q[12ull*w + k];
       rm[w - v0] = u - v0;
       atomicAdd(&s_collapses, 1u);
     }
     __syncthreads();
     if (s_collapses == 0) break;
-    // [8] rewrite + stable compact (oracle step 4); keep flag in valid[]
-    for (uint32_t f = tid; f < nt; f += BS) {
-      uint32_t i0 = v0 + rm[fa[3*f] - v0], i1 = v0 + rm[fa[3*f+1] - v0],
-               i2 = v0 + rm[fa[3*f+2] - v0];
-      fa[3*f] = i0; fa[3*f+1] = i1; fa[3*f+2] = i2;
-      valid[f] = (i0 != i1 && i1 != i2 && i0 != i2) ? 1 : 0;
-    }
-    __syncthreads();
-    uint32_t kept = blk_compact_faces<BS>(fa, fb, valid, nt, s_sums);
+    // [8] fused rewrite + stable compact (oracle step 4)
+    uint32_t kept = blk_rewrite_compact<BS>(fa, fb, rm, v0, nt, s_sums);
     { uint32_t *t = fa; fa = fb; fb = t; }  // compacted faces now in fa
     if (tid == 0) s_nt = kept;
     __syncthreads();
