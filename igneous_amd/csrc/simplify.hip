@@ -665,11 +665,45 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
       float q[10];
       #pragma unroll
       for (int k = 0; k < 10; ++k) q[k] = 0.0f;
-      if (d <= 16) {
+      if (d <= 8) {
+        // dominant tier: interior vertices have degree ~6 — an 8-wide
+        // bitonic network + 8 plane gathers costs ~1/3 of the 16-wide
+        // path it replaces. Sorted order is unique (face ids distinct),
+        // so this matches the oracle's insertion sort exactly.
+        uint32_t fl[8];
+        #pragma unroll
+        for (int k = 0; k < 8; ++k)
+          fl[k] = (k < (int)d)
+                      ? (clmode ? (uint32_t)s_cl[lo + k] : cl[lo + k])
+                      : 0xFFFFFFFFu;
+        #pragma unroll
+        for (int ksz = 2; ksz <= 8; ksz <<= 1) {
+          #pragma unroll
+          for (int j = ksz >> 1; j > 0; j >>= 1) {
+            #pragma unroll
+            for (int i = 0; i < 8; ++i) {
+              int l = i ^ j;
+              if (l > i) {
+                bool up = ((i & ksz) == 0);
+                uint32_t a = fl[i], b2 = fl[l];
+                bool sw = up ? (a > b2) : (a < b2);
+                fl[i] = sw ? b2 : a;
+                fl[l] = sw ? a : b2;
+              }
+            }
+          }
+        }
+        SimpPlane ps[8];
+        #pragma unroll
+        for (int k = 0; k < 8; ++k)
+          ps[k] = pl[k < (int)d ? fl[k] : 0u];
+        #pragma unroll
+        for (int k = 0; k < 8; ++k)
+          if (k < (int)d)
+            sq_add_plane(q, ps[k].nx, ps[k].ny, ps[k].nz, ps[k].d, 1.0f);
+      } else if (d <= 16) {
         // register path: one batched load of the face list, bitonic
         // sort network (compile-time indices, no global RMW chains).
-        // Sorted order is unique (face ids distinct), so this matches
-        // the oracle's insertion sort exactly.
         uint32_t fl[16];
         #pragma unroll
         for (int k = 0; k < 16; ++k)
@@ -750,25 +784,39 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
       __syncthreads();
     }
     // [6] picks (oracle step 2). Q rows are 12 floats (48 B, 16-B
-    // aligned) so the two quadric gathers are 3 dwordx4 loads each.
+    // aligned): stage each corner's quadric ONCE per face (3 dwordx4
+    // loads) and build every edge's sum from registers — the edge loop
+    // re-loading both quadrics per edge doubled the phase's traffic.
+    // f32 order preserved via selects on compile-time indices, exactly
+    // like the global k_edge_pick.
     for (uint32_t f = tid; f < nt; f += BS) {
       uint32_t fc[3] = {fa[3*f], fa[3*f+1], fa[3*f+2]};
+      float cq[3][10], cp[3][3];
+      #pragma unroll
+      for (int ci = 0; ci < 3; ++ci) {
+        const float4 *qp = (const float4 *)(Q + 12ull*fc[ci]);
+        float4 a4 = qp[0], b4 = qp[1], c4 = qp[2];
+        cq[ci][0] = a4.x; cq[ci][1] = a4.y; cq[ci][2] = a4.z;
+        cq[ci][3] = a4.w; cq[ci][4] = b4.x; cq[ci][5] = b4.y;
+        cq[ci][6] = b4.z; cq[ci][7] = b4.w; cq[ci][8] = c4.x;
+        cq[ci][9] = c4.y;
+        #pragma unroll
+        for (int k = 0; k < 3; ++k) cp[ci][k] = verts[3ull*fc[ci] + k];
+      }
       #pragma unroll
       for (int e = 0; e < 3; ++e) {
-        uint32_t a = fc[e], bb = fc[(e+1)%3];
+        const int ea = e, eb = (e + 1) % 3;  // compile-time after unroll
+        uint32_t a = fc[ea], bb = fc[eb];
         if (a == bb) continue;
-        uint32_t u = a < bb ? a : bb, w = a < bb ? bb : a;
-        float mx = 0.5f*(verts[3ull*u]+verts[3ull*w]);
-        float my = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
-        float mz = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
-        const float4 *qu = (const float4 *)(Q + 12ull*u);
-        const float4 *qw = (const float4 *)(Q + 12ull*w);
-        float4 su0 = qu[0], su1 = qu[1], su2 = qu[2];
-        float4 sw0 = qw[0], sw1 = qw[1], sw2 = qw[2];
-        float S[10] = {su0.x + sw0.x, su0.y + sw0.y, su0.z + sw0.z,
-                       su0.w + sw0.w, su1.x + sw1.x, su1.y + sw1.y,
-                       su1.z + sw1.z, su1.w + sw1.w, su2.x + sw2.x,
-                       su2.y + sw2.y};
+        bool fwd = a < bb;
+        uint32_t u = fwd ? a : bb, w = fwd ? bb : a;
+        float mx = 0.5f*((fwd ? cp[ea][0] : cp[eb][0]) + (fwd ? cp[eb][0] : cp[ea][0]));
+        float my = 0.5f*((fwd ? cp[ea][1] : cp[eb][1]) + (fwd ? cp[eb][1] : cp[ea][1]));
+        float mz = 0.5f*((fwd ? cp[ea][2] : cp[eb][2]) + (fwd ? cp[eb][2] : cp[ea][2]));
+        float S[10];
+        #pragma unroll
+        for (int k = 0; k < 10; ++k)
+          S[k] = (fwd ? cq[ea][k] : cq[eb][k]) + (fwd ? cq[eb][k] : cq[ea][k]);
         float cost = sq_eval(S, mx, my, mz);
         if (cost < 0.0f) cost = 0.0f;
         if (cost > max_cost) continue;
