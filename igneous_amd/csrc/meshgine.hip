@@ -635,7 +635,7 @@ struct mg_ctx {
       simp_Q, simp_pick, simp_remap, simp_flab, simp_flab_alt,
       simp_faces_alt, simp_verts_alt, simp_vbase_alt, simp_meta, simp_ref,
       simp_keep, simp_keep_scan, simp_park, simp_first, simp_troff_final,
-      simp_deg, simp_adj, simp_rh;
+      simp_deg, simp_adj, simp_rh, simp_sched;
   uint64_t lh_slots = 1ull << 20;
   HostBuf h_verts, h_faces;  // pinned output staging, reused across calls
   hipEvent_t ev[16] = {};
@@ -742,7 +742,7 @@ void mg_destroy(mg_ctx *c) {
                   &c->simp_meta, &c->simp_ref, &c->simp_keep,
                   &c->simp_keep_scan, &c->simp_park, &c->simp_first,
                   &c->simp_troff_final, &c->simp_deg, &c->simp_adj,
-                  &c->simp_rh}) {
+                  &c->simp_rh, &c->simp_sched}) {
     if (b->ptr) (void)hipFree(b->ptr);
   }
   if (c->h_verts.ptr) (void)hipHostFree(c->h_verts.ptr);
@@ -899,6 +899,31 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     d_rh = (uint32_t *)c->simp_rh.ptr;
     HIP_TRY(c, hipMemsetAsync(d_rh, 0, 160 * 4, s), 40);
   }
+  // biggest-label-first dispatch order (MG_SIMP_SCHED=0 disables):
+  // per-label runtime ~ nt0, so sort label ids by descending face count
+  // — a straggler dispatched late otherwise extends the launch by its
+  // whole serial runtime
+  uint32_t *d_sched = nullptr;
+  {
+    const char *e = getenv("MG_SIMP_SCHED");
+    if (!(e && e[0] == '0')) {
+      if (ensure(c, c->simp_sched, 3 * L * 4)) return 40;
+      uint32_t *iota = (uint32_t *)c->simp_sched.ptr;
+      uint32_t *keys_out = iota + L;
+      d_sched = keys_out + L;
+      uint32_t nbl = (uint32_t)((L + 255) / 256);
+      hipLaunchKernelGGL(k_iota, dim3(nbl), dim3(256), 0, s, iota, L);
+      size_t tmp = 0;
+      hipError_t err = rocprim::radix_sort_pairs_desc(
+          nullptr, tmp, nt_cur, keys_out, iota, d_sched, L, 0u, 32u, s);
+      if (err != hipSuccess) { SET_ERR(c, "sched sort size"); return 40; }
+      if (ensure(c, c->sort_tmp, tmp)) return 40;
+      err = rocprim::radix_sort_pairs_desc(
+          c->sort_tmp.ptr, tmp, nt_cur, keys_out, iota, d_sched, L, 0u,
+          32u, s);
+      if (err != hipSuccess) { SET_ERR(c, "sched sort"); return 40; }
+    }
+  }
   {
     // default OFF since sub-round groups: the +24 KB LDS CSR payload
     // costs blocks/CU and the recompute phases it serves are now ~25%
@@ -947,7 +972,8 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                            ? (unsigned long long *)c->lh_misc.ptr + 8
                            : nullptr,
                        d_rh,
-                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, 0u, small_cap);
+                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, 0u, small_cap,
+                       d_sched);
       HIP_TRY(c, hipGetLastError(), 40);
     }
     hipLaunchKernelGGL(ksl, dim3((uint32_t)L), dim3(bs), 0, s,
@@ -968,7 +994,8 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                            ? (unsigned long long *)c->lh_misc.ptr + 8
                            : nullptr,
                        d_rh,
-                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, small_cap, 0xFFFFFFFFu);
+                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, small_cap, 0xFFFFFFFFu,
+                       d_sched);
     HIP_TRY(c, hipGetLastError(), 40);
     if (getenv("MG_SIMP_PROF")) {
       unsigned long long hp[6];
@@ -988,7 +1015,12 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
               40);
       HIP_TRY(c, hipStreamSynchronize(s), 40);
       fprintf(stderr, "[mg simp rounds] labels=%u sum_groups=%u "
-              "sum_subs=%u sum_nt0=%u\n", h[130], h[128], h[129], h[131]);
+              "sum_subs=%u sum_nt0=%u max_cycles=%u (nt0=%u)\n",
+              h[130], h[128], h[129], h[131], h[132], h[133]);
+      fprintf(stderr, "[mg simp log2-cycle hist]");
+      for (int k = 0; k < 26; ++k)
+        if (h[134 + k]) fprintf(stderr, " %d:%u", k, h[134 + k]);
+      fprintf(stderr, "\n");
       fprintf(stderr, "[mg simp group hist]");
       for (int k = 0; k < 64; ++k)
         if (h[k]) fprintf(stderr, " %d:%u", k, h[k]);

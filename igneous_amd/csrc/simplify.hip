@@ -541,9 +541,15 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
     unsigned long long *__restrict__ prof,  // 6 phase counters or null
     uint32_t *__restrict__ roundhist,  // 160 u32 or null (see host)
     float max_cost, uint32_t nlabels, uint32_t big_cap,
-    uint32_t subs, uint32_t nv_lo, uint32_t nv_hi) {
-  const uint32_t b = blockIdx.x;
-  if (b >= nlabels) return;
+    uint32_t subs, uint32_t nv_lo, uint32_t nv_hi,
+    const uint32_t *__restrict__ sched /*block->label, largest first*/) {
+  if (blockIdx.x >= nlabels) return;
+  // biggest-label-first dispatch: per-label serial time scales with nt0,
+  // and a 65k-face label dispatched late extends the whole launch by its
+  // full runtime — schedule stragglers first so they overlap the swarm
+  const uint32_t b = sched ? sched[blockIdx.x] : blockIdx.x;
+  unsigned long long t_start =
+      roundhist ? __builtin_amdgcn_s_memtime() : 0;
   const uint32_t f0 = tri_off[b];
   const uint32_t nt0 = tri_off[b + 1] - f0;
   if (nt0 > big_cap) return;  // global-rounds path handles big labels
@@ -877,7 +883,14 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
     if (roundhist) {
       // layout: [0..63] group-count hist, [64..127] sub-count hist
       // (capped), [128] sum groups, [129] sum subs, [130] labels seen,
-      // [131] sum nt0 (faces entering), [132..159] spare
+      // [131] sum nt0 (faces entering), [132] max label cycles,
+      // [133] nt0 of (a) max-cycle label, [134..159] log2 cycle hist
+      unsigned long long cyc = __builtin_amdgcn_s_memtime() - t_start;
+      uint32_t ck = (uint32_t)(cyc > 0xFFFFFFFFull ? 0xFFFFFFFFull : cyc);
+      uint32_t prev = atomicMax(&roundhist[132], ck);
+      if (ck > prev) roundhist[133] = nt0;  // racy but indicative
+      int lg = 32 - __clz(ck | 1);          // 1..32
+      atomicAdd(&roundhist[134 + (lg < 26 ? lg : 25)], 1u);
       atomicAdd(&roundhist[n_groups < 64 ? n_groups : 63], 1u);
       atomicAdd(&roundhist[64 + (n_subs < 64 ? n_subs : 63)], 1u);
       atomicAdd(&roundhist[128], n_groups);
