@@ -621,6 +621,7 @@ struct HostBuf {
 struct mg_ctx {
   int device = 0;
   hipStream_t stream = nullptr;
+  hipStream_t stream2 = nullptr;  // concurrent simplify band
   std::mutex lock;
   std::string err;
   mg_stats stats = {};
@@ -721,6 +722,11 @@ mg_ctx *mg_init(int device_id) {
     delete c;
     return nullptr;
   }
+  if (hipStreamCreateWithFlags(&c->stream2, hipStreamNonBlocking) !=
+      hipSuccess) {
+    delete c;
+    return nullptr;
+  }
   for (auto &e : c->ev)
     if (hipEventCreate(&e) != hipSuccess) { delete c; return nullptr; }
   return c;
@@ -748,6 +754,7 @@ void mg_destroy(mg_ctx *c) {
   if (c->h_verts.ptr) (void)hipHostFree(c->h_verts.ptr);
   if (c->h_faces.ptr) (void)hipHostFree(c->h_faces.ptr);
   for (auto &e : c->ev) if (e) (void)hipEventDestroy(e);
+  if (c->stream2) (void)hipStreamDestroy(c->stream2);
   if (c->stream) (void)hipStreamDestroy(c->stream);
   delete c;
 }
@@ -895,9 +902,9 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   // round-count histogram (MG_SIMP_ROUNDHIST=1): groups/subs per label
   uint32_t *d_rh = nullptr;
   if (getenv("MG_SIMP_ROUNDHIST")) {
-    if (ensure(c, c->simp_rh, 160 * 4)) return 40;
+    if (ensure(c, c->simp_rh, 176 * 4)) return 40;
     d_rh = (uint32_t *)c->simp_rh.ptr;
-    HIP_TRY(c, hipMemsetAsync(d_rh, 0, 160 * 4, s), 40);
+    HIP_TRY(c, hipMemsetAsync(d_rh, 0, 176 * 4, s), 40);
   }
   // biggest-label-first dispatch order (MG_SIMP_SCHED=1 enables;
   // measured 8% WORSE on the 512^3/50k config: the front-loaded big
@@ -953,50 +960,52 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     // MG_SIMP_SMALL=1 enables for experiments.
     const char *sm = getenv("MG_SIMP_SMALL");
     const uint32_t small_cap = (sm && sm[0] == '1') ? 512u : 0u;
-    if (small_cap) {
-      hipLaunchKernelGGL((k_simplify_label<false, 64, 512>),
-                         dim3((uint32_t)L), dim3(64), 0, s,
-                       faces_g, (uint32_t *)c->simp_faces_alt.ptr,
-                       (const uint32_t *)c->tri_off.ptr,
-                       (const uint32_t *)c->vbase.ptr,
-                       verts, (float *)c->simp_Q.ptr,
-                       (unsigned long long *)c->simp_pick.ptr,
-                       (uint32_t *)c->simp_remap.ptr,
-                       (uint32_t *)c->simp_deg.ptr,
-                       (uint32_t *)c->simp_adj.ptr,
-                       (uint32_t *)c->simp_pk.ptr,
-                       (SimpPlane *)c->simp_fq.ptr,
-                       (uint8_t *)c->simp_valid.ptr,
-                       nt_cur, target, active,
-                       (uint32_t *)c->simp_park.ptr,
-                       getenv("MG_SIMP_PROF")
-                           ? (unsigned long long *)c->lh_misc.ptr + 8
-                           : nullptr,
-                       d_rh,
-                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, 0u, small_cap,
-                       d_sched);
-      HIP_TRY(c, hipGetLastError(), 40);
+    auto launch_band = [&](decltype(&k_simplify_label<false, 256, 2048>) fn,
+                           hipStream_t st, int bsz,
+                           uint32_t nv_lo, uint32_t nv_hi) {
+      hipLaunchKernelGGL(fn, dim3((uint32_t)L), dim3(bsz), 0, st,
+                         faces_g, (uint32_t *)c->simp_faces_alt.ptr,
+                         (const uint32_t *)c->tri_off.ptr,
+                         (const uint32_t *)c->vbase.ptr,
+                         verts, (float *)c->simp_Q.ptr,
+                         (unsigned long long *)c->simp_pick.ptr,
+                         (uint32_t *)c->simp_remap.ptr,
+                         (uint32_t *)c->simp_deg.ptr,
+                         (uint32_t *)c->simp_adj.ptr,
+                         (uint32_t *)c->simp_pk.ptr,
+                         (SimpPlane *)c->simp_fq.ptr,
+                         (uint8_t *)c->simp_valid.ptr,
+                         nt_cur, target, active,
+                         (uint32_t *)c->simp_park.ptr,
+                         getenv("MG_SIMP_PROF")
+                             ? (unsigned long long *)c->lh_misc.ptr + 8
+                             : nullptr,
+                         d_rh,
+                         max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs,
+                         nv_lo, nv_hi, d_sched);
+    };
+    if (small_cap)
+      launch_band(k_simplify_label<false, 64, 512>, s, 64, 0u, small_cap);
+    // nv-banded dispatch (MG_SIMP_BANDS=0 disables): labels with
+    // 2048 < nv <= 4096 get their own CAPV=4096 instantiation (64 KB
+    // LDS, 2 blocks/CU) on a SECOND stream, concurrent with the main
+    // band — previously they fell into the global-array mode whose
+    // hot per-vertex atomics serialize on L2 lines
+    const char *bandsenv = getenv("MG_SIMP_BANDS");
+    const bool bands = !(bandsenv && bandsenv[0] == '0') && bs == 256 &&
+                       !use_cl;
+    if (bands) {
+      launch_band(ksl, s, bs, small_cap, 2048u);
+      HIP_TRY(c, hipEventRecord(c->ev[9], s), 40);
+      HIP_TRY(c, hipStreamWaitEvent(c->stream2, c->ev[9], 0), 40);
+      launch_band(k_simplify_label<false, 256, 4096>, c->stream2, 256,
+                  2048u, 4096u);
+      launch_band(ksl, c->stream2, bs, 4096u, 0xFFFFFFFFu);
+      HIP_TRY(c, hipEventRecord(c->ev[10], c->stream2), 40);
+      HIP_TRY(c, hipStreamWaitEvent(s, c->ev[10], 0), 40);
+    } else {
+      launch_band(ksl, s, bs, small_cap, 0xFFFFFFFFu);
     }
-    hipLaunchKernelGGL(ksl, dim3((uint32_t)L), dim3(bs), 0, s,
-                       faces_g, (uint32_t *)c->simp_faces_alt.ptr,
-                       (const uint32_t *)c->tri_off.ptr,
-                       (const uint32_t *)c->vbase.ptr,
-                       verts, (float *)c->simp_Q.ptr,
-                       (unsigned long long *)c->simp_pick.ptr,
-                       (uint32_t *)c->simp_remap.ptr,
-                       (uint32_t *)c->simp_deg.ptr,
-                       (uint32_t *)c->simp_adj.ptr,
-                       (uint32_t *)c->simp_pk.ptr,
-                       (SimpPlane *)c->simp_fq.ptr,
-                       (uint8_t *)c->simp_valid.ptr,
-                       nt_cur, target, active,
-                       (uint32_t *)c->simp_park.ptr,
-                       getenv("MG_SIMP_PROF")
-                           ? (unsigned long long *)c->lh_misc.ptr + 8
-                           : nullptr,
-                       d_rh,
-                       max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs, small_cap, 0xFFFFFFFFu,
-                       d_sched);
     HIP_TRY(c, hipGetLastError(), 40);
     if (getenv("MG_SIMP_PROF")) {
       unsigned long long hp[6];
@@ -1011,13 +1020,19 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       fprintf(stderr, "\n");
     }
     if (d_rh) {
-      uint32_t h[160];
-      HIP_TRY(c, hipMemcpyAsync(h, d_rh, 160 * 4, hipMemcpyDeviceToHost, s),
+      uint32_t h[176];
+      HIP_TRY(c, hipMemcpyAsync(h, d_rh, 176 * 4, hipMemcpyDeviceToHost, s),
               40);
       HIP_TRY(c, hipStreamSynchronize(s), 40);
       fprintf(stderr, "[mg simp rounds] labels=%u sum_groups=%u "
               "sum_subs=%u sum_nt0=%u max_cycles=%u (nt0=%u)\n",
               h[130], h[128], h[129], h[131], h[132], h[133]);
+      {
+        unsigned long long lds_c, glob_c;
+        memcpy(&lds_c, &h[160], 8); memcpy(&glob_c, &h[162], 8);
+        fprintf(stderr, "[mg simp mode split] lds: n=%u cyc=%llu | "
+                "global: n=%u cyc=%llu\n", h[164], lds_c, h[165], glob_c);
+      }
       fprintf(stderr, "[mg simp log2-cycle hist]");
       for (int k = 0; k < 26; ++k)
         if (h[134 + k]) fprintf(stderr, " %d:%u", k, h[134 + k]);

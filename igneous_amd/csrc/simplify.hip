@@ -519,6 +519,9 @@ __device__ uint32_t blk_rewrite_compact(uint32_t *__restrict__ fa,
   return total;
 }
 
+// (an amdgpu_waves_per_eu(4) floor was measured here: it fits 127 VGPRs
+// with 52 B/lane scratch, and the spill traffic LOSES 7% net — the
+// natural 148-VGPR / 3-waves-per-SIMD allocation stands)
 template <bool CLLDS, int BS, int CAPVT>
 __global__ __launch_bounds__(BS) void k_simplify_label(
     uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
@@ -733,18 +736,23 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
             }
           }
         }
-        // gather all planes first (independent loads), then the ordered
-        // f32 accumulation from registers
-        SimpPlane ps[16];
+        // gather planes in two 8-batches (8 independent loads in flight
+        // each) — staging all 16 at once cost 64 VGPRs and pushed the
+        // kernel past the 128-VGPR / 4-waves-per-SIMD occupancy step.
+        // Accumulation order (ascending fl) is unchanged.
         #pragma unroll
-        for (int k = 0; k < 16; ++k)
-          // pad with face 0 (nt >= 1 inside the loop); fl[k] is the
-          // 0xFFFFFFFF sort sentinel beyond d and must not be indexed
-          ps[k] = pl[k < (int)d ? fl[k] : 0u];
-        #pragma unroll
-        for (int k = 0; k < 16; ++k)
-          if (k < (int)d)
-            sq_add_plane(q, ps[k].nx, ps[k].ny, ps[k].nz, ps[k].d, 1.0f);
+        for (int h = 0; h < 2; ++h) {
+          SimpPlane ps[8];
+          #pragma unroll
+          for (int k = 0; k < 8; ++k)
+            // pad with face 0 (nt >= 1 inside the loop); fl[k] is the
+            // 0xFFFFFFFF sort sentinel beyond d, must not be indexed
+            ps[k] = pl[8*h + k < (int)d ? fl[8*h + k] : 0u];
+          #pragma unroll
+          for (int k = 0; k < 8; ++k)
+            if (8*h + k < (int)d)
+              sq_add_plane(q, ps[k].nx, ps[k].ny, ps[k].nz, ps[k].d, 1.0f);
+        }
       } else if (clmode) {
         for (uint32_t i = lo + 1; i < hi; ++i) {
           uint16_t x = s_cl[i];
@@ -891,6 +899,11 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
       if (ck > prev) roundhist[133] = nt0;  // racy but indicative
       int lg = 32 - __clz(ck | 1);          // 1..32
       atomicAdd(&roundhist[134 + (lg < 26 ? lg : 25)], 1u);
+      // cycle totals split by LDS vs global-array mode ([160..163] u64x2,
+      // [164] lds count, [165] global count — buffer is 176 u32)
+      atomicAdd((unsigned long long *)&roundhist[lds_mode ? 160 : 162],
+                cyc);
+      atomicAdd(&roundhist[164 + (lds_mode ? 0 : 1)], 1u);
       atomicAdd(&roundhist[n_groups < 64 ? n_groups : 63], 1u);
       atomicAdd(&roundhist[64 + (n_subs < 64 ? n_subs : 63)], 1u);
       atomicAdd(&roundhist[128], n_groups);
