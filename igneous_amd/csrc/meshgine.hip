@@ -992,8 +992,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     // band — previously they fell into the global-array mode whose
     // hot per-vertex atomics serialize on L2 lines
     const char *bandsenv = getenv("MG_SIMP_BANDS");
-    const bool bands = !(bandsenv && bandsenv[0] == '0') && bs == 256 &&
-                       !use_cl;
+    const bool bands = !(bandsenv && bandsenv[0] == '0') && !use_cl;
     if (bands) {
       launch_band(ksl, s, bs, small_cap, 2048u);
       HIP_TRY(c, hipEventRecord(c->ev[9], s), 40);
