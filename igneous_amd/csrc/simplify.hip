@@ -482,10 +482,13 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
 
 // fused face rewrite + stable compaction: remap corners through the
 // label-local rm[] (LDS), keep-flag in a register bitmask (no valid[]
-// array traffic), scan via blk_prefix, scatter kept faces fa->fb.
-// Identical output order to the old rewrite + compact pair.
+// array traffic), scan via blk_prefix, scatter kept remapped faces into
+// fb. fa is never written: the second pass re-reads fa and re-applies
+// the (cheap, LDS) rm gathers instead of paying a 12 B/face global
+// write + re-read. Identical output order to the old rewrite + compact
+// pair.
 template <int BS>
-__device__ uint32_t blk_rewrite_compact(uint32_t *__restrict__ fa,
+__device__ uint32_t blk_rewrite_compact(const uint32_t *__restrict__ fa,
                                         uint32_t *__restrict__ fb,
                                         const uint32_t *__restrict__ rm,
                                         uint32_t v0,
@@ -499,10 +502,9 @@ __device__ uint32_t blk_rewrite_compact(uint32_t *__restrict__ fa,
   unsigned long long bm[MAXW] = {};
   uint32_t sum = 0;
   for (uint32_t i = lo; i < hi; ++i) {
-    uint32_t i0 = v0 + rm[fa[3*i] - v0];
-    uint32_t i1 = v0 + rm[fa[3*i+1] - v0];
-    uint32_t i2 = v0 + rm[fa[3*i+2] - v0];
-    fa[3*i] = i0; fa[3*i+1] = i1; fa[3*i+2] = i2;
+    uint32_t i0 = rm[fa[3*i] - v0];
+    uint32_t i1 = rm[fa[3*i+1] - v0];
+    uint32_t i2 = rm[fa[3*i+2] - v0];
     if (i0 != i1 && i1 != i2 && i0 != i2) {
       bm[(i - lo) >> 6] |= 1ull << ((i - lo) & 63);
       ++sum;
@@ -512,7 +514,9 @@ __device__ uint32_t blk_rewrite_compact(uint32_t *__restrict__ fa,
   uint32_t run = blk_prefix<BS>(sum, s_sums, &total);
   for (uint32_t i = lo; i < hi; ++i)
     if (bm[(i - lo) >> 6] & (1ull << ((i - lo) & 63))) {
-      fb[3*run] = fa[3*i]; fb[3*run+1] = fa[3*i+1]; fb[3*run+2] = fa[3*i+2];
+      fb[3*run] = v0 + rm[fa[3*i] - v0];
+      fb[3*run+1] = v0 + rm[fa[3*i+1] - v0];
+      fb[3*run+2] = v0 + rm[fa[3*i+2] - v0];
       ++run;
     }
   __syncthreads();  // s_sums reusable after this
@@ -522,6 +526,13 @@ __device__ uint32_t blk_rewrite_compact(uint32_t *__restrict__ fa,
 // (an amdgpu_waves_per_eu(4) floor was measured here: it fits 127 VGPRs
 // with 52 B/lane scratch, and the spill traffic LOSES 7% net — the
 // natural 148-VGPR / 3-waves-per-SIMD allocation stands)
+// The 16-wide bitonic sort tier is the kernel's VGPR peak: with it the
+// allocation is 148 VGPRs (3 waves/SIMD), without it 116 (4 waves/SIMD)
+// — degree>8 vertices are rare enough that the insertion-sort fallback
+// plus the extra wave slot wins (A/B'd on the 512^3/50k config).
+#ifndef SIMP_D16_TIER
+#define SIMP_D16_TIER 0
+#endif
 template <bool CLLDS, int BS, int CAPVT>
 __global__ __launch_bounds__(BS) void k_simplify_label(
     uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
@@ -710,7 +721,7 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
         for (int k = 0; k < 8; ++k)
           if (k < (int)d)
             sq_add_plane(q, ps[k].nx, ps[k].ny, ps[k].nz, ps[k].d, 1.0f);
-      } else if (d <= 16) {
+      } else if (SIMP_D16_TIER && d <= 16) {
         // register path: one batched load of the face list, bitonic
         // sort network (compile-time indices, no global RMW chains).
         uint32_t fl[16];
