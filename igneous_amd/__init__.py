@@ -12,8 +12,12 @@ from .queue import LocalTaskQueue, RegisteredTask
 from .tasks import (
     MeshTask, MeshManifestPrefixTask, MeshManifestFilesystemTask,
     TransferMeshFilesTask, DeleteMeshFilesTask,
+    MultiResShardedMeshMergeTask, MultiResUnshardedMeshMergeTask,
 )
-from .task_creation import create_meshing_tasks
+from .task_creation import (
+    create_meshing_tasks, create_sharded_multires_mesh_tasks,
+    create_unsharded_multires_mesh_tasks,
+)
 from .volume import PrecomputedVolume
 from .lib import Bbox, Vec
 

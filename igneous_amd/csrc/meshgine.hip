@@ -1770,6 +1770,80 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   return 0;
 }
 
+// ---------------------------------------------------------------------------
+// Standalone mesh simplification (multires LOD chain; see meshgine.h).
+// Reuses the per-label simplify machinery with ONE label spanning the
+// whole mesh.
+
+extern "C" int mg_simplify_mesh(mg_ctx *c,
+                                const float *verts, uint32_t nverts,
+                                const uint32_t *faces, uint32_t ntris,
+                                uint32_t reduction_factor, float max_error,
+                                const float **out_verts,
+                                uint32_t *out_nverts,
+                                const uint32_t **out_faces,
+                                uint32_t *out_ntris) {
+  if (!c) { SET_ERR(c, "null ctx"); return 1; }
+  std::lock_guard<std::mutex> g(c->lock);
+  c->err.clear();
+  if (!verts || !faces || !out_verts || !out_faces) {
+    SET_ERR(c, "null argument");
+    return 1;
+  }
+  if (ntris > (1u << 31) / 3) {
+    SET_ERR(c, "mesh too large (%u tris)", ntris);
+    return 2;
+  }
+  HIP_TRY(c, hipSetDevice(c->device), 4);
+  hipStream_t s = c->stream;
+  uint64_t T = ntris, V = nverts;
+
+  if (reduction_factor > 1 && T > 0 && V > 0) {
+    if (ensure(c, c->verts, V * 12 + 12)) return 10;
+    if (ensure(c, c->faces, 3 * T * 4 + 4)) return 10;
+    if (ensure(c, c->tri_off, 8)) return 10;
+    if (ensure(c, c->vbase, 8)) return 10;
+    if (ensure(c, c->tri_label, T * 4 + 4)) return 10;
+    if (ensure(c, c->vtx_scan, (std::max<uint64_t>(V + 1, 3 * T)) * 4))
+      return 10;
+    HIP_TRY(c, hipMemcpyAsync(c->verts.ptr, verts, V * 12,
+                              hipMemcpyHostToDevice, s), 11);
+    HIP_TRY(c, hipMemcpyAsync(c->faces.ptr, faces, 3 * T * 4,
+                              hipMemcpyHostToDevice, s), 11);
+    uint32_t off_h[2] = {0, (uint32_t)T};
+    uint32_t vb_h[2] = {0, (uint32_t)V};
+    HIP_TRY(c, hipMemcpyAsync(c->tri_off.ptr, off_h, 8,
+                              hipMemcpyHostToDevice, s), 11);
+    HIP_TRY(c, hipMemcpyAsync(c->vbase.ptr, vb_h, 8,
+                              hipMemcpyHostToDevice, s), 11);
+    HIP_TRY(c, hipMemsetAsync(c->tri_label.ptr, 0, T * 4, s), 11);
+    if (ensure(c, c->lh_misc, 256)) return 11;
+    int rc = run_simplify(c, 1, (const uint32_t *)c->tri_label.ptr,
+                          reduction_factor, max_error, &T, &V);
+    if (rc) return rc;
+    if (ensure_host(c, c->h_verts, V * 12 + 12)) return 12;
+    if (ensure_host(c, c->h_faces, 3 * T * 4 + 4)) return 12;
+    if (V > 0)
+      HIP_TRY(c, hipMemcpyAsync(c->h_verts.ptr, c->verts.ptr, V * 12,
+                                hipMemcpyDeviceToHost, s), 12);
+    if (T > 0)
+      HIP_TRY(c, hipMemcpyAsync(c->h_faces.ptr, c->faces.ptr, 3 * T * 4,
+                                hipMemcpyDeviceToHost, s), 12);
+    HIP_TRY(c, hipStreamSynchronize(s), 12);
+  } else {
+    // no-op request: hand back a staged copy (uniform ownership)
+    if (ensure_host(c, c->h_verts, V * 12 + 12)) return 12;
+    if (ensure_host(c, c->h_faces, 3 * T * 4 + 4)) return 12;
+    memcpy(c->h_verts.ptr, verts, V * 12);
+    memcpy(c->h_faces.ptr, faces, 3 * T * 4);
+  }
+  *out_verts = (const float *)c->h_verts.ptr;
+  *out_nverts = (uint32_t)V;
+  *out_faces = (const uint32_t *)c->h_faces.ptr;
+  *out_ntris = (uint32_t)T;
+  return 0;
+}
+
 template <typename T>
 static int run_count_emit(mg_ctx *c, const T *d_labels, const GridDims &g,
                           LabelHash lh, uint32_t *d_segcnt,

@@ -114,6 +114,17 @@ def load_library() -> ctypes.CDLL:
         lib.mg_last_error.argtypes = [ctypes.c_void_p]
         lib.mg_device_count.restype = ctypes.c_int
         lib.mg_version.restype = ctypes.c_char_p
+        lib.mg_simplify_mesh.restype = ctypes.c_int
+        lib.mg_simplify_mesh.argtypes = [
+            ctypes.c_void_p,
+            ctypes.c_void_p, ctypes.c_uint32,
+            ctypes.c_void_p, ctypes.c_uint32,
+            ctypes.c_uint32, ctypes.c_float,
+            ctypes.POINTER(ctypes.POINTER(ctypes.c_float)),
+            ctypes.POINTER(ctypes.c_uint32),
+            ctypes.POINTER(ctypes.POINTER(ctypes.c_uint32)),
+            ctypes.POINTER(ctypes.c_uint32),
+        ]
         _lib = lib
         return lib
 
@@ -232,6 +243,33 @@ class Engine:
                 self.lib.mg_meshset_free(out)
         return result
 
+    def simplify_mesh(self, verts: np.ndarray, faces: np.ndarray,
+                      reduction_factor: int,
+                      max_error: float = 1e30):
+        """Standalone quadric simplification of one mesh on the GPU
+        (mg_simplify_mesh; the multires LOD chain's simplify_fqmr
+        replacement, multires.py:342). Returns owned (verts, faces)."""
+        v = np.ascontiguousarray(verts, dtype=np.float32)
+        f = np.ascontiguousarray(faces, dtype=np.uint32)
+        ov = ctypes.POINTER(ctypes.c_float)()
+        of = ctypes.POINTER(ctypes.c_uint32)()
+        onv = ctypes.c_uint32()
+        ont = ctypes.c_uint32()
+        rc = self.lib.mg_simplify_mesh(
+            self.ctx,
+            v.ctypes.data_as(ctypes.c_void_p), v.shape[0],
+            f.ctypes.data_as(ctypes.c_void_p), f.shape[0],
+            int(reduction_factor), float(max_error),
+            ctypes.byref(ov), ctypes.byref(onv),
+            ctypes.byref(of), ctypes.byref(ont))
+        if rc != 0:
+            raise RuntimeError(
+                f"mg_simplify_mesh failed rc={rc}: "
+                f"{self.lib.mg_last_error(self.ctx).decode()}")
+        out_v = np.ctypeslib.as_array(ov, shape=(onv.value, 3)).copy()
+        out_f = np.ctypeslib.as_array(of, shape=(ont.value, 3)).copy()
+        return out_v, out_f
+
     def stats(self) -> dict:
         s = MgStats()
         rc = self.lib.mg_get_stats(self.ctx, ctypes.byref(s))
@@ -251,3 +289,21 @@ def mesh_chunk(labels: np.ndarray, resolution=(1.0, 1.0, 1.0),
     return Engine.get(device_id).mesh_chunk(
         labels, resolution, reduction_factor, max_error, voxel_centered,
         copy=copy)
+
+
+def simplify_mesh(mesh, target_count: int, max_error: float = 1e30,
+                  device_id: Optional[int] = None):
+    """Module-level LOD simplifier: Mesh -> Mesh at ~target_count faces
+    (reduction_factor = ntris // target; <=1 returns the mesh as-is)."""
+    from .meshes import Mesh
+    nt = int(mesh.faces.shape[0])
+    target = max(int(target_count), 1)
+    if nt <= target:
+        return Mesh(mesh.vertices.copy(), mesh.faces.copy(), id=mesh.id)
+    rf = max(nt // target, 2)
+    if device_id is None:
+        device_id = int(os.environ.get("MESHGINE_DEVICE",
+                                       os.environ.get("LOCAL_RANK", "0")))
+    v, f = Engine.get(device_id).simplify_mesh(
+        mesh.vertices, mesh.faces, rf, max_error)
+    return Mesh(v, f, id=mesh.id)

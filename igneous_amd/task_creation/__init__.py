@@ -1,2 +1,8 @@
-from .mesh import create_meshing_tasks, create_mesh_manifest_tasks
+from .mesh import (
+    create_meshing_tasks,
+    create_mesh_manifest_tasks,
+    create_sharded_multires_mesh_tasks,
+    create_unsharded_multires_mesh_tasks,
+    configure_multires_info,
+)
 from .common import FinelyDividedTaskIterator, num_tasks

@@ -16,7 +16,10 @@ from ..tasks.mesh import MeshTask
 from ..volume import PrecomputedVolume
 from .common import FinelyDividedTaskIterator, operator_contact
 
-__all__ = ["create_meshing_tasks", "create_mesh_manifest_tasks"]
+__all__ = ["create_meshing_tasks", "create_mesh_manifest_tasks",
+           "create_sharded_multires_mesh_tasks",
+           "create_unsharded_multires_mesh_tasks",
+           "configure_multires_info"]
 
 
 def create_meshing_tasks(
@@ -140,3 +143,164 @@ def create_mesh_manifest_tasks(layer_path: str, magnitude: int = 3,
     assert int(magnitude) == magnitude and magnitude > 0
     return [partial(MeshManifestFilesystemTask,
                     layer_path=layer_path, mesh_dir=mesh_dir)]
+
+
+# ---------------------------------------------------------------------------
+# Multires / sharded task creation (reference task_creation/mesh.py:
+# configure_multires_info :437-479, create_unsharded_multires_mesh_tasks
+# :481-530, create_sharded_multires_mesh_tasks :706-813).
+
+import copy as _copy
+
+import numpy as _np
+
+from ..formats import sharding as _sharding
+
+
+def configure_multires_info(cloudpath: str,
+                            vertex_quantization_bits: int,
+                            mesh_dir: Optional[str]):
+    """Write the neuroglancer_multilod_draco mesh info
+    (task_creation/mesh.py:437-479)."""
+    assert vertex_quantization_bits in (10, 16), vertex_quantization_bits
+    vol = PrecomputedVolume(cloudpath)
+    mesh_dir = mesh_dir or vol.info.get("mesh", None)
+    if "mesh" not in vol.info:
+        vol.info["mesh"] = mesh_dir
+        vol.commit_info()
+
+    cf = CloudFiles(cloudpath)
+    info_filename = f"{mesh_dir}/info"
+    mesh_info = cf.get_json(info_filename) or {}
+    mip = int(mesh_info.get("mip", 0))
+    res = PrecomputedVolume(cloudpath, mip=mip).resolution
+    new_mesh_info = _copy.deepcopy(mesh_info)
+    new_mesh_info['@type'] = "neuroglancer_multilod_draco"
+    new_mesh_info['vertex_quantization_bits'] = vertex_quantization_bits
+    new_mesh_info['transform'] = [
+        int(res[0]), 0, 0, 0,
+        0, int(res[1]), 0, 0,
+        0, 0, int(res[2]), 0,
+    ]
+    new_mesh_info['lod_scale_multiplier'] = 1.0
+    if new_mesh_info != mesh_info:
+        cf.put_json(info_filename, new_mesh_info,
+                    cache_control="no-cache")
+    return new_mesh_info
+
+
+def create_unsharded_multires_mesh_tasks(
+        cloudpath: str, num_lod: int = 0,
+        magnitude: int = 3, mesh_dir: Optional[str] = None,
+        vertex_quantization_bits: int = 16,
+        min_chunk_size=(256, 256, 256)):
+    """One prefix-task list over existing unsharded fragments
+    (task_creation/mesh.py:481-530; file:// -> single full-prefix task)."""
+    from functools import partial
+    from ..tasks.multires import MultiResUnshardedMeshMergeTask
+
+    configure_multires_info(cloudpath, vertex_quantization_bits, mesh_dir)
+    return [partial(MultiResUnshardedMeshMergeTask,
+                    cloudpath, prefix="",
+                    mesh_dir=mesh_dir, num_lod=num_lod,
+                    min_chunk_size=min_chunk_size)]
+
+
+def create_sharded_multires_mesh_tasks(
+        cloudpath: str,
+        shard_index_bytes: int = 2 ** 13,
+        minishard_index_bytes: int = 2 ** 15,
+        min_shards: int = 1,
+        num_lod: int = 0,
+        draco_compression_level: int = 7,
+        vertex_quantization_bits: int = 16,
+        minishard_index_encoding: str = "gzip",
+        mesh_dir: Optional[str] = None,
+        spatial_index_db: Optional[str] = None,
+        frag_path: Optional[str] = None,
+        cache: Optional[bool] = False,
+        min_chunk_size=(256, 256, 256),
+        max_labels_per_shard: Optional[int] = None,
+        progress: bool = True):
+    """Mirror of task_creation/mesh.py:706-813: write the multilod info
+    + sharding spec, assign labels to shards from the spatial index,
+    persist {shard}.labels, return one MultiResShardedMeshMergeTask per
+    shard."""
+    from functools import partial
+    from ..spatial_index import SpatialIndex
+    from ..tasks.multires import MultiResShardedMeshMergeTask
+
+    mesh_info = configure_multires_info(
+        cloudpath, vertex_quantization_bits, mesh_dir)
+    mdir = mesh_dir or PrecomputedVolume(cloudpath).info.get("mesh")
+
+    all_labels = SpatialIndex(cloudpath, mdir).query()
+
+    if max_labels_per_shard is not None:
+        assert max_labels_per_shard >= 1
+        min_shards = max(
+            int(_np.ceil(len(all_labels) / max_labels_per_shard)),
+            min_shards)
+
+    (shard_bits, minishard_bits, preshift_bits) = \
+        _sharding.compute_shard_params_for_hashed(
+            num_labels=len(all_labels),
+            shard_index_bytes=int(shard_index_bytes),
+            minishard_index_bytes=int(minishard_index_bytes),
+            min_shards=min_shards)
+
+    spec = _sharding.ShardingSpecification(
+        type='neuroglancer_uint64_sharded_v1',
+        preshift_bits=preshift_bits,
+        hash='murmurhash3_x86_128',
+        minishard_bits=minishard_bits,
+        shard_bits=shard_bits,
+        minishard_index_encoding=minishard_index_encoding,
+        data_encoding="raw",  # draco encoded meshes
+    )
+
+    cf = CloudFiles(cloudpath)
+    mesh_info['sharding'] = spec.to_dict()
+    cf.put_json(f"{mdir}/info", mesh_info, cache_control="no-cache")
+
+    shard_labels = _sharding.assign_labels_to_shards(
+        _np.asarray(all_labels, dtype=_np.uint64),
+        preshift_bits, shard_bits, minishard_bits)
+    cf_mesh = CloudFiles(f"{cloudpath.rstrip('/')}/{mdir}")
+    for shardno, labels in shard_labels.items():
+        cf_mesh.put_json(str(shardno) + '.labels', labels,
+                         compress="gzip", cache_control="no-cache")
+
+    vol = PrecomputedVolume(cloudpath)
+    vol.provenance.processing.append({
+        'method': {
+            'task': 'MultiResShardedMeshMergeTask',
+            'cloudpath': cloudpath,
+            'mip': int(mesh_info.get('mip', 0)),
+            'num_lod': num_lod,
+            'vertex_quantization_bits': vertex_quantization_bits,
+            'preshift_bits': preshift_bits,
+            'minishard_bits': minishard_bits,
+            'shard_bits': shard_bits,
+            'mesh_dir': mdir,
+            'frag_path': frag_path,
+            'draco_compression_level': draco_compression_level,
+            'min_chunk_size': list(min_chunk_size),
+        },
+        'by': operator_contact(),
+        'date': strftime('%Y-%m-%d %H:%M %Z'),
+    })
+    vol.commit_provenance()
+
+    return [
+        partial(MultiResShardedMeshMergeTask,
+                cloudpath, shard_no,
+                num_lod=num_lod,
+                mesh_dir=mdir,
+                frag_path=frag_path,
+                cache=cache,
+                spatial_index_db=spatial_index_db,
+                draco_compression_level=draco_compression_level,
+                min_chunk_size=min_chunk_size)
+        for shard_no in shard_labels.keys()
+    ]

@@ -6,3 +6,7 @@ from .mesh import (
     DeleteMeshFilesTask,
     set_mesher,
 )
+from .multires import (
+    MultiResShardedMeshMergeTask,
+    MultiResUnshardedMeshMergeTask,
+)

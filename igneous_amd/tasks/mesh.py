@@ -23,8 +23,11 @@ Deliberate deviations (documented in DESIGN.md):
   - dry_run returns (meshes, bounding_boxes) computed; the reference's
     dry_run references bounding_boxes before assignment (mesh.py:249-252,
     a latent NameError).
-  - draco encoding, sharded output, fill_holes>0 and dust_global raise
-    NotImplementedError (out of the hot-path scope).
+  - fill_holes>0 and dust_global raise NotImplementedError (out of the
+    hot-path scope). Sharded output uses the in-repo
+    MapBuffer restatement and draco encoding the in-repo Draco
+    restatement (formats/, byte-level external parity unpinned offline
+    — DESIGN.md §7).
 """
 from __future__ import annotations
 
@@ -124,18 +127,10 @@ class MeshTask(RegisteredTask):
     # ------------------------------------------------------------------
     def execute(self):
         opts = self.options
-        if opts['encoding'] == 'draco':
-            raise NotImplementedError(
-                "draco encoding is outside the meshgine hot-path scope "
-                "(SURVEY §8f row 4); use encoding='precomputed'")
         if opts['fill_holes'] > 0:
             raise NotImplementedError(
                 "fill_holes>0 requires fastmorph/crackle (out of scope, "
                 "SURVEY §2 'fastmorph, crackle' row)")
-        if opts['sharded']:
-            raise NotImplementedError(
-                "sharded (MapBuffer) fragment output is a SURVEY §8f 'next' "
-                "row, not yet implemented")
 
         self._volume = PrecomputedVolume(
             self.layer_path, opts['mip'], bounded=False,
@@ -150,6 +145,18 @@ class MeshTask(RegisteredTask):
         data_bounds.maxpt += opts['high_padding']
 
         self._mesh_dir = self.get_mesh_dir()
+
+        if opts['encoding'] == 'draco':
+            from .draco import draco_encoding_settings
+            self.draco_encoding_settings = draco_encoding_settings(
+                shape=(self.shape + opts['low_padding']
+                       + opts['high_padding']),
+                offset=self.offset,
+                resolution=self._volume.resolution,
+                compression_level=opts['draco_compression_level'],
+                create_metadata=opts['draco_create_metadata'],
+                uses_new_draco_bin_size=False,
+            )
 
         data = self._volume.download(data_bounds)
 
@@ -211,7 +218,10 @@ class MeshTask(RegisteredTask):
         if opts['dry_run']:
             return (meshes, bounding_boxes)
 
-        self._upload_individuals(binaries, opts['generate_manifests'])
+        if opts['sharded']:
+            self._upload_batch(binaries, self._bounds)
+        else:
+            self._upload_individuals(binaries, opts['generate_manifests'])
 
         if opts['spatial_index']:
             self._upload_spatial_index(self._bounds, bounding_boxes)
@@ -286,17 +296,47 @@ class MeshTask(RegisteredTask):
         # flat [minx,miny,minz,maxx,maxy,maxz] like Bbox.to_list() (mesh.py:257)
         mesh_bounds = (np.amin(mesh.vertices, axis=0).tolist()
                        + np.amax(mesh.vertices, axis=0).tolist())
+        if self.options['encoding'] == 'draco':
+            # mesh.py:442-446 via our draco restatement (formats/draco.py)
+            from ..formats import draco as draco_fmt
+            binary = draco_fmt.encode(
+                mesh.vertices, mesh.faces, **self.draco_encoding_settings)
+            return binary, mesh_bounds
         return mesh.to_precomputed(), mesh_bounds
+
+    def _upload_batch(self, mesh_binaries, bbox: Bbox):
+        """Sharded fragment output (reference mesh.py:385-397): all the
+        chunk's meshes in ONE MapBuffer file
+        "{mesh_dir}/{bbox.to_filename()}.frags", consumed per label by
+        the sharded multires merge (multires.py:425). The reference
+        compresses values with brotli ("br"); offline that downgrades to
+        gzip inside our MapBuffer restatement (formats/mapbuffer.py,
+        DESIGN.md §7)."""
+        from ..formats.mapbuffer import MapBuffer
+        frag_path = self.options['frag_path'] or self.layer_path
+        cf = CloudFiles(frag_path)
+        mbuf = MapBuffer(
+            {int(k): v for k, v in mesh_binaries.items()}, compress="br")
+        cf.put(
+            f"{self._mesh_dir}/{bbox.to_filename()}.frags",
+            mbuf.tobytes(),
+            compress=None,
+            content_type="application/x.mapbuffer",
+            cache_control=False,
+        )
 
     def _upload_individuals(self, mesh_binaries, generate_manifests):
         cf = CloudFiles(self.layer_path)
+        content_type = ("model/x.draco"
+                        if self.options["encoding"] == "draco"
+                        else "model/mesh")  # mesh.py:403-406
         cf.puts(
             ((f"{self._mesh_dir}/{segid}:{self.options['lod']}:"
               f"{self._bounds.to_filename()}", binary)
              for segid, binary in mesh_binaries.items()),
             compress=self._encoding_to_compression_dict[self.options['encoding']],
             cache_control=self.options['cache_control'],
-            content_type="model/mesh",
+            content_type=content_type,
         )
         if generate_manifests:
             cf.put_jsons(

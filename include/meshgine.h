@@ -143,6 +143,25 @@ int mg_mesh_chunk(mg_ctx *ctx, const void *labels,
 
 void mg_meshset_free(mg_meshset *ms);
 
+/* Quadric edge-collapse simplification of ONE standalone mesh on the
+ * GPU — the simplifier reused at merge granularity: replaces
+ * zmesh.simplify_fqmr at the multires LOD chain call site
+ * (/root/reference/igneous/tasks/mesh/multires.py:342 via
+ * generate_lods). Same deterministic contract as the per-label chunk
+ * simplifier (oracle/simplify.c omc_simplify_mesh is the checker).
+ *   verts/faces      host arrays (float32 V*3 / uint32 T*3)
+ *   reduction_factor target = ntris/reduction_factor (<=1: no-op)
+ *   max_error        cost bound in the verts' own units
+ *   out_*            ctx-owned pinned staging, valid until the ctx's
+ *                    next mg_mesh_chunk/mg_simplify_mesh call
+ * Returns 0 on success. */
+int mg_simplify_mesh(mg_ctx *ctx,
+                     const float *verts, uint32_t nverts,
+                     const uint32_t *faces, uint32_t ntris,
+                     uint32_t reduction_factor, float max_error,
+                     const float **out_verts, uint32_t *out_nverts,
+                     const uint32_t **out_faces, uint32_t *out_ntris);
+
 /* Stats of the last mg_mesh_chunk on this ctx. Returns 0 on success. */
 int mg_get_stats(mg_ctx *ctx, mg_stats *out);
 

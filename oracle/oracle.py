@@ -63,6 +63,11 @@ def _get_lib():
             ctypes.POINTER(ctypes.POINTER(_MeshSet)),
         ]
         _lib.omc_meshset_free.argtypes = [ctypes.POINTER(_MeshSet)]
+        _lib.omc_simplify_mesh.argtypes = [
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32),
+            ctypes.c_void_p, ctypes.POINTER(ctypes.c_uint32),
+            ctypes.c_uint32, ctypes.c_float,
+        ]
     return _lib
 
 
@@ -103,3 +108,20 @@ def mesh_chunk(labels: np.ndarray, resolution=(1.0, 1.0, 1.0),
     finally:
         lib.omc_meshset_free(out)
     return result
+
+
+def simplify_mesh(verts: np.ndarray, faces: np.ndarray,
+                  reduction_factor: int, max_error: float = 1e30):
+    """Standalone quadric edge-collapse on one mesh (the same
+    omc_simplify_mesh the per-label path runs; checker for the engine's
+    mg_simplify_mesh used by the multires LOD chain)."""
+    lib = _get_lib()
+    v = np.ascontiguousarray(verts, dtype=np.float32).copy()
+    f = np.ascontiguousarray(faces, dtype=np.uint32).copy()
+    nv = ctypes.c_uint32(v.shape[0])
+    nt = ctypes.c_uint32(f.shape[0])
+    lib.omc_simplify_mesh(
+        v.ctypes.data_as(ctypes.c_void_p), ctypes.byref(nv),
+        f.ctypes.data_as(ctypes.c_void_p), ctypes.byref(nt),
+        int(reduction_factor), float(max_error))
+    return v[:nv.value].copy(), f[:nt.value].copy()

@@ -44,3 +44,26 @@ def oracle_mesher():
     mesh_mod.set_mesher(fn)
     yield fn
     mesh_mod.set_mesher(None)
+
+
+@pytest.fixture
+def oracle_simplifier():
+    """Inject the CPU oracle's quadric simplifier as the multires LOD
+    simplifier (checker role; the product path is mg_simplify_mesh)."""
+    import oracle as _oracle
+    from igneous_amd.meshes import Mesh
+    from igneous_amd.tasks import multires as multires_mod
+
+    def fn(mesh, target_count):
+        nt = int(mesh.faces.shape[0])
+        target = max(int(target_count), 1)
+        if nt <= target:
+            return Mesh(mesh.vertices.copy(), mesh.faces.copy(),
+                        id=mesh.id)
+        rf = max(nt // target, 2)
+        v, f = _oracle.simplify_mesh(mesh.vertices, mesh.faces, rf, 1e30)
+        return Mesh(v, f, id=mesh.id)
+
+    multires_mod.set_simplifier(fn)
+    yield fn
+    multires_mod.set_simplifier(None)
