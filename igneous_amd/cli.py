@@ -55,7 +55,8 @@ def mesh():
 @click.option("--compress", default="gzip", help="gzip or none")
 @click.option("--spatial-index/--no-spatial-index", default=True)
 @click.option("--sharded", is_flag=True, default=False,
-              help="(not supported in round 1: raises)")
+              help="write MapBuffer .frags fragment files for the "
+                   "sharded multires merge instead of individual meshes")
 @click.option("--fill-missing", is_flag=True, default=False)
 @click.option("--closed-edge/--open-edge", default=True,
               help="close meshes at dataset boundaries")
@@ -85,6 +86,58 @@ def forge(path, queue, mip, shape, simplify, max_error, dust_threshold,
     rank, world = rank_world()
     n = execute_tasks(tasks)
     click.echo(f"rank {rank}/{world}: executed {n} tasks")
+
+
+@mesh.command()
+@click.argument("path")
+@click.option("--dir", "mesh_dir", default=None,
+              help="mesh subdirectory (overrides info)")
+@click.option("--magnitude", default=3)
+def merge(path, mesh_dir, magnitude):
+    """Stage 2: merge mesh fragment manifests (reference mesh merge,
+    cli.py:1082-1103)."""
+    from .task_creation import create_mesh_manifest_tasks
+    tasks = create_mesh_manifest_tasks(path, magnitude=magnitude,
+                                       mesh_dir=mesh_dir)
+    n = execute_tasks(tasks)
+    click.echo(f"executed {n} manifest tasks")
+
+
+@mesh.command(name="merge-sharded")
+@click.argument("path")
+@click.option("--nlod", "num_lod", default=0,
+              help="number of additional levels of detail")
+@click.option("--vqb", "vertex_quantization_bits", default=16,
+              type=click.Choice(["10", "16"]))
+@click.option("--shard-index-bytes", default=2 ** 13)
+@click.option("--minishard-index-bytes", default=2 ** 15)
+@click.option("--min-shards", default=1)
+@click.option("--min-chunk-size", default="256,256,256", callback=_vec3)
+@click.option("--dir", "mesh_dir", default=None)
+@click.option("--frag-path", default=None)
+@click.option("--draco-compression-level", default=7)
+def merge_sharded(path, num_lod, vertex_quantization_bits,
+                  shard_index_bytes, minishard_index_bytes, min_shards,
+                  min_chunk_size, mesh_dir, frag_path,
+                  draco_compression_level):
+    """Stage 2 (sharded): merge .frags fragments into multires
+    neuroglancer shard files (reference mesh merge-sharded,
+    cli.py:1105-1161)."""
+    from .task_creation import create_sharded_multires_mesh_tasks
+    tasks = create_sharded_multires_mesh_tasks(
+        path,
+        shard_index_bytes=shard_index_bytes,
+        minishard_index_bytes=minishard_index_bytes,
+        min_shards=min_shards,
+        num_lod=num_lod,
+        draco_compression_level=draco_compression_level,
+        vertex_quantization_bits=int(vertex_quantization_bits),
+        mesh_dir=mesh_dir,
+        frag_path=frag_path,
+        min_chunk_size=min_chunk_size,
+    )
+    n = execute_tasks(tasks)
+    click.echo(f"executed {n} sharded multires merge tasks")
 
 
 @main.command()
