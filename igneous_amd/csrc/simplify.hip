@@ -39,6 +39,38 @@ __device__ __forceinline__ float sq_eval(const float *q, float x, float y,
        + q[9];
 }
 
+// Canonical collapse placement — textually identical arithmetic to
+// oracle/simplify.c quad_place (the contract expression): GH optimal
+// point via Cramer on the summed quadric, midpoint fallback when the
+// 3x3 system is near-singular. Writes the placement, returns its cost.
+__device__ __forceinline__ float sq_place(const float *S, float mx,
+                                          float my, float mz,
+                                          float *px, float *py,
+                                          float *pz) {
+  float a00 = S[0], a01 = S[1], a02 = S[2], b0 = S[3];
+  float a11 = S[4], a12 = S[5], b1 = S[6];
+  float a22 = S[7], b2 = S[8];
+  float m00 = a11*a22 - a12*a12;
+  float m01 = a02*a12 - a01*a22;
+  float m02 = a01*a12 - a02*a11;
+  float m11 = a00*a22 - a02*a02;
+  float m12 = a01*a02 - a00*a12;
+  float m22 = a00*a11 - a01*a01;
+  float det = a00*m00 + a01*m01 + a02*m02;
+  float tr = a00 + a11 + a22;
+  float x = mx, y = my, z = mz;
+  if (fabsf(det) > 1e-6f * tr * tr * tr) {
+    float inv = 1.0f / det;
+    x = -(m00*b0 + m01*b1 + m02*b2) * inv;
+    y = -(m01*b0 + m11*b1 + m12*b2) * inv;
+    z = -(m02*b0 + m12*b1 + m22*b2) * inv;
+  }
+  *px = x; *py = y; *pz = z;
+  float cost = sq_eval(S, x, y, z);
+  if (cost < 0.0f) cost = 0.0f;
+  return cost;
+}
+
 // [S1] per-face plane (recomputed each round; verts move)
 __global__ void k_face_planes(const uint32_t *__restrict__ faces_g,
                               const float *__restrict__ verts,
@@ -161,8 +193,8 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
     #pragma unroll
     for (int k = 0; k < 10; ++k)
       S[k] = (fwd ? cq[ea][k] : cq[eb][k]) + (fwd ? cq[eb][k] : cq[ea][k]);
-    float cost = sq_eval(S, mx, my, mz);
-    if (cost < 0.0f) cost = 0.0f;
+    float px, py, pz;
+    float cost = sq_place(S, mx, my, mz, &px, &py, &pz);
     if (cost > max_cost) continue;
     uint32_t cb = __float_as_uint(cost);
     // per-edge tie jitter — identical to oracle/simplify.c, which works
@@ -190,9 +222,18 @@ __global__ void k_collapse(const unsigned long long *__restrict__ pick,
   if (w <= u) return;
   unsigned long long pw = pick[w];
   if (pw == ~0ull || (uint32_t)pw != u) return;
-  verts[3*u]   = 0.5f*(verts[3*u]+verts[3*w]);
-  verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
-  verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+  {
+    float mx = 0.5f*(verts[3*u]+verts[3*w]);
+    float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+    float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+    float S[10];
+    #pragma unroll
+    for (int k = 0; k < 10; ++k)
+      S[k] = Q[12*u + k] + Q[12*(uint64_t)w + k];
+    float px, py, pz;
+    (void)sq_place(S, mx, my, mz, &px, &py, &pz);
+    verts[3*u] = px; verts[3*u+1] = py; verts[3*u+2] = pz;
+  }
   #pragma unroll
   for (int k = 0; k < 10; ++k) Q[12*u + k] += Q[12*(uint64_t)w + k];
   remap[w] = (uint32_t)u;
@@ -842,8 +883,8 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
         #pragma unroll
         for (int k = 0; k < 10; ++k)
           S[k] = (fwd ? cq[ea][k] : cq[eb][k]) + (fwd ? cq[eb][k] : cq[ea][k]);
-        float cost = sq_eval(S, mx, my, mz);
-        if (cost < 0.0f) cost = 0.0f;
+        float px, py, pz;
+        float cost = sq_place(S, mx, my, mz, &px, &py, &pz);
         if (cost > max_cost) continue;
         uint32_t cb = __float_as_uint(cost);
         uint32_t ul = u - v0, wl = w - v0;  // label-local ids (oracle)
@@ -867,9 +908,18 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
       if (w <= u) continue;
       unsigned long long pw = pick_l[w - v0];
       if (pw == ~0ull || (uint32_t)pw != u) continue;
-      verts[3ull*u]   = 0.5f*(verts[3ull*u]+verts[3ull*w]);
-      verts[3ull*u+1] = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
-      verts[3ull*u+2] = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
+      {
+        float mx = 0.5f*(verts[3ull*u]+verts[3ull*w]);
+        float my = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
+        float mz = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
+        float S[10];
+        #pragma unroll
+        for (int k = 0; k < 10; ++k)
+          S[k] = Q[12ull*u + k] + Q[12ull*w + k];
+        float px, py, pz;
+        (void)sq_place(S, mx, my, mz, &px, &py, &pz);
+        verts[3ull*u] = px; verts[3ull*u+1] = py; verts[3ull*u+2] = pz;
+      }
       #pragma unroll
       for (int k = 0; k < 10; ++k) Q[12ull*u + k] += Q[12ull*w + k];
       rm[w - v0] = u - v0;

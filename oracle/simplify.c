@@ -14,12 +14,14 @@
  *    the face's UNIT plane quadric (w=1; f32 arithmetic, plain summation in
  *    ascending face index order) — unit weight keeps the cost in nm^2 so it
  *    compares directly against max_error^2;
- *  - candidate per edge (u,v), u<v: placement = midpoint (0.5*(pu+pv)),
- *    cost = (Qu+Qv)(m) in nm^2;
+ *  - candidate per edge (u,v), u<v: placement = the Garland-Heckbert
+ *    OPTIMAL point (Cramer solve of the summed quadric's normal
+ *    equations, quad_place below), falling back to the midpoint when
+ *    the 3x3 system is near-singular; cost = (Qu+Qv)(placement) in nm^2;
  *  - a round: every vertex picks its cheapest incident edge (ties -> the
  *    smaller peer index); an edge collapses iff both endpoints picked it
  *    AND cost <= max_error^2;
- *  - collapse moves both endpoints to the midpoint, remaps v->u (u<v),
+ *  - collapse moves both endpoints to the placement, remaps v->u (u<v),
  *    ADDS the loser's quadric into the winner (Q_u += Q_w, f32), and
  *    drops degenerate faces;
  *  - rounds are grouped: one full quadric recompute (step 1) is followed
@@ -53,6 +55,39 @@ static float quad_eval(const quad10 *Q, float x, float y, float z) {
        + Q->q[4]*y*y + 2.0f*Q->q[5]*y*z + 2.0f*Q->q[6]*y
        + Q->q[7]*z*z + 2.0f*Q->q[8]*z
        + Q->q[9];
+}
+
+/* Canonical collapse placement (contract; the HIP kernels mirror this
+ * arithmetic VERBATIM — f32, fixed expression order, -ffp-contract=off):
+ * the Garland-Heckbert optimal point solving A x = -b by Cramer's rule
+ * on the summed quadric S (A = upper-left 3x3, b = (q3,q6,q8)), falling
+ * back to the edge midpoint when A is near-singular (flat regions,
+ * where the midpoint is already cost-0). Writes the placement, returns
+ * its cost. */
+static float quad_place(const quad10 *S, float mx, float my, float mz,
+                        float *px, float *py, float *pz) {
+  float a00 = S->q[0], a01 = S->q[1], a02 = S->q[2], b0 = S->q[3];
+  float a11 = S->q[4], a12 = S->q[5], b1 = S->q[6];
+  float a22 = S->q[7], b2 = S->q[8];
+  float m00 = a11*a22 - a12*a12;
+  float m01 = a02*a12 - a01*a22;
+  float m02 = a01*a12 - a02*a11;
+  float m11 = a00*a22 - a02*a02;
+  float m12 = a01*a02 - a00*a12;
+  float m22 = a00*a11 - a01*a01;
+  float det = a00*m00 + a01*m01 + a02*m02;
+  float tr = a00 + a11 + a22;
+  float x = mx, y = my, z = mz;
+  if (fabsf(det) > 1e-6f * tr * tr * tr) {
+    float inv = 1.0f / det;
+    x = -(m00*b0 + m01*b1 + m02*b2) * inv;
+    y = -(m01*b0 + m11*b1 + m12*b2) * inv;
+    z = -(m02*b0 + m12*b1 + m22*b2) * inv;
+  }
+  *px = x; *py = y; *pz = z;
+  float cost = quad_eval(S, x, y, z);
+  if (cost < 0.0f) cost = 0.0f;
+  return cost;
 }
 
 void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
@@ -116,8 +151,8 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
         float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
         quad10 S;
         for (int k = 0; k < 10; k++) S.q[k] = Q[u].q[k] + Q[w].q[k];
-        float cost = quad_eval(&S, mx, my, mz);
-        if (cost < 0.0f) cost = 0.0f;
+        float px, py, pz;
+        float cost = quad_place(&S, mx, my, mz, &px, &py, &pz);
         if (cost > max_cost) continue;
         uint32_t cb; memcpy(&cb, &cost, 4);
         /* deterministic per-edge jitter on the 3 low cost bits: breaks
@@ -143,9 +178,16 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
       if (w <= u) continue;               /* handle each pair from its min end */
       if (pick[w] == UINT64_MAX || (uint32_t)pick[w] != u) continue; /* not matched */
       if (remap[u] != u || remap[w] != w) continue; /* already touched */
-      verts[3*u]   = 0.5f*(verts[3*u]+verts[3*w]);
-      verts[3*u+1] = 0.5f*(verts[3*u+1]+verts[3*w+1]);
-      verts[3*u+2] = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+      {
+        float mx = 0.5f*(verts[3*u]+verts[3*w]);
+        float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+        float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+        quad10 S;
+        for (int k = 0; k < 10; k++) S.q[k] = Q[u].q[k] + Q[w].q[k];
+        float px, py, pz;
+        (void)quad_place(&S, mx, my, mz, &px, &py, &pz);
+        verts[3*u] = px; verts[3*u+1] = py; verts[3*u+2] = pz;
+      }
       for (int k = 0; k < 10; k++) Q[u].q[k] += Q[w].q[k];
       remap[w] = u;
       collapses++;
