@@ -71,6 +71,34 @@ __device__ __forceinline__ float sq_place(const float *S, float mx,
   return cost;
 }
 
+// cost-only form for the pick phase (identical arithmetic; the
+// placement registers don't outlive the call — keeps the per-label
+// kernel at <=128 VGPRs / 4 waves per SIMD)
+__device__ __forceinline__ float sq_place_cost(const float *S, float mx,
+                                               float my, float mz) {
+  float a00 = S[0], a01 = S[1], a02 = S[2], b0 = S[3];
+  float a11 = S[4], a12 = S[5], b1 = S[6];
+  float a22 = S[7], b2 = S[8];
+  float m00 = a11*a22 - a12*a12;
+  float m01 = a02*a12 - a01*a22;
+  float m02 = a01*a12 - a02*a11;
+  float m11 = a00*a22 - a02*a02;
+  float m12 = a01*a02 - a00*a12;
+  float m22 = a00*a11 - a01*a01;
+  float det = a00*m00 + a01*m01 + a02*m02;
+  float tr = a00 + a11 + a22;
+  float x = mx, y = my, z = mz;
+  if (fabsf(det) > 1e-6f * tr * tr * tr) {
+    float inv = 1.0f / det;
+    x = -(m00*b0 + m01*b1 + m02*b2) * inv;
+    y = -(m01*b0 + m11*b1 + m12*b2) * inv;
+    z = -(m02*b0 + m12*b1 + m22*b2) * inv;
+  }
+  float cost = sq_eval(S, x, y, z);
+  if (cost < 0.0f) cost = 0.0f;
+  return cost;
+}
+
 // [S1] per-face plane (recomputed each round; verts move)
 __global__ void k_face_planes(const uint32_t *__restrict__ faces_g,
                               const float *__restrict__ verts,
@@ -193,8 +221,7 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
     #pragma unroll
     for (int k = 0; k < 10; ++k)
       S[k] = (fwd ? cq[ea][k] : cq[eb][k]) + (fwd ? cq[eb][k] : cq[ea][k]);
-    float px, py, pz;
-    float cost = sq_place(S, mx, my, mz, &px, &py, &pz);
+    float cost = sq_place_cost(S, mx, my, mz);
     if (cost > max_cost) continue;
     uint32_t cb = __float_as_uint(cost);
     // per-edge tie jitter — identical to oracle/simplify.c, which works
@@ -575,7 +602,8 @@ __device__ uint32_t blk_rewrite_compact(const uint32_t *__restrict__ fa,
 #define SIMP_D16_TIER 0
 #endif
 template <bool CLLDS, int BS, int CAPVT>
-__global__ __launch_bounds__(BS) void k_simplify_label(
+__global__ __launch_bounds__(BS)
+__attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
     uint32_t *__restrict__ faces_tmp,     // same slicing (scratch)
     const uint32_t *__restrict__ tri_off, // L+1 (original offsets)
@@ -883,8 +911,7 @@ __global__ __launch_bounds__(BS) void k_simplify_label(
         #pragma unroll
         for (int k = 0; k < 10; ++k)
           S[k] = (fwd ? cq[ea][k] : cq[eb][k]) + (fwd ? cq[eb][k] : cq[ea][k]);
-        float px, py, pz;
-        float cost = sq_place(S, mx, my, mz, &px, &py, &pz);
+        float cost = sq_place_cost(S, mx, my, mz);
         if (cost > max_cost) continue;
         uint32_t cb = __float_as_uint(cost);
         uint32_t ul = u - v0, wl = w - v0;  // label-local ids (oracle)
