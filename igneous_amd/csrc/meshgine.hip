@@ -119,6 +119,45 @@ __device__ __forceinline__ uint32_t label_lookup(LabelHash h, uint64_t label) {
   return 0;
 }
 
+// ---------------------------------------------------------------------------
+// device dust removal (MeshTask dust_threshold, reference mesh.py:313-323
+// via fastremap.unique + mask): three volume passes — build the label
+// hash, histogram voxels per label, zero voxels of labels below the
+// threshold. Replaces a multi-second host numpy unique/mask at 512^3.
+
+template <typename T>
+__global__ void k_dust_build(const T *__restrict__ labels, uint64_t nvox,
+                             LabelHash lh) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < nvox; i += (uint64_t)gridDim.x * blockDim.x) {
+    T L = labels[i];
+    if (L != 0) label_insert(lh, (uint64_t)L);
+  }
+}
+
+template <typename T>
+__global__ void k_dust_count(const T *__restrict__ labels, uint64_t nvox,
+                             LabelHash lh, uint32_t *__restrict__ counts) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < nvox; i += (uint64_t)gridDim.x * blockDim.x) {
+    T L = labels[i];
+    if (L != 0) atomicAdd(&counts[label_lookup(lh, (uint64_t)L)], 1u);
+  }
+}
+
+template <typename T>
+__global__ void k_dust_zero(T *__restrict__ labels, uint64_t nvox,
+                            LabelHash lh,
+                            const uint32_t *__restrict__ counts,
+                            uint64_t thr) {
+  for (uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < nvox; i += (uint64_t)gridDim.x * blockDim.x) {
+    T L = labels[i];
+    if (L != 0 && (uint64_t)counts[label_lookup(lh, (uint64_t)L)] < thr)
+      labels[i] = 0;
+  }
+}
+
 // invert: label id -> label value (fill after count)
 __global__ void k_label_values(LabelHash h, uint64_t *values, uint64_t nslots) {
   uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -781,13 +820,15 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                            int sx, int sy, int sz, int dtype,
                            float rx, float ry, float rz,
                            uint32_t reduction_factor, float max_error,
-                           int voxel_centered, uint32_t flags_,
+                           int voxel_centered, uint64_t dust_threshold,
+                           uint32_t flags_,
                            mg_meshset **out);
 
 int mg_mesh_chunk(mg_ctx *c, const void *labels, int sx, int sy, int sz,
                   int dtype, float rx, float ry, float rz,
                   uint32_t reduction_factor, float max_error,
-                  int voxel_centered, uint32_t flags, mg_meshset **out) {
+                  int voxel_centered, uint64_t dust_threshold,
+                  uint32_t flags, mg_meshset **out) {
   if (!c) { SET_ERR(c, "null ctx"); return 1; }
   std::lock_guard<std::mutex> g(c->lock);
   c->err.clear();
@@ -803,7 +844,7 @@ int mg_mesh_chunk(mg_ctx *c, const void *labels, int sx, int sy, int sz,
   HIP_TRY(c, hipSetDevice(c->device), 4);
   return mesh_chunk_impl(c, labels, sx, sy, sz, dtype, rx, ry, rz,
                          reduction_factor, max_error, voxel_centered,
-                         flags, out);
+                         dust_threshold, flags, out);
 }
 
 }  // extern "C"
@@ -1396,7 +1437,8 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                            int sx, int sy, int sz, int dtype,
                            float rx, float ry, float rz,
                            uint32_t reduction_factor, float max_error,
-                           int voxel_centered, uint32_t flags_,
+                           int voxel_centered, uint64_t dust_threshold,
+                           uint32_t flags_,
                            mg_meshset **out) {
   const size_t esize = (dtype == MG_U64) ? 8 : 4;
   const uint64_t nvox = (uint64_t)sx * sy * sz;
@@ -1432,6 +1474,65 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                               hipMemcpyHostToDevice, s), 11);
   }
   HIP_TRY(c, hipEventRecord(c->ev[1], s), 11);
+
+  // device dust removal (three volume passes; see k_dust_* above)
+  if (dust_threshold > 0) {
+    const int dblk = 256;
+    const uint32_t dnb = 4096;
+    for (;;) {
+      if (ensure(c, c->lh_keys, c->lh_slots * 8)) return 16;
+      if (ensure(c, c->lh_vals, c->lh_slots * 4)) return 16;
+      if (ensure(c, c->lh_misc, 256)) return 16;
+      HIP_TRY(c, hipMemsetAsync(c->lh_keys.ptr, 0, c->lh_slots * 8, s), 16);
+      HIP_TRY(c, hipMemsetAsync(c->lh_misc.ptr, 0, 256, s), 16);
+      LabelHash dlh;
+      dlh.keys = (uint64_t *)c->lh_keys.ptr;
+      dlh.vals = (uint32_t *)c->lh_vals.ptr;
+      dlh.counter = (uint32_t *)c->lh_misc.ptr;
+      dlh.overflow = (uint32_t *)c->lh_misc.ptr + 1;
+      dlh.nslots = c->lh_slots;
+      if (dtype == MG_U64)
+        hipLaunchKernelGGL(k_dust_build<uint64_t>, dim3(dnb), dim3(dblk),
+                           0, s, (const uint64_t *)c->labels.ptr, nvox, dlh);
+      else
+        hipLaunchKernelGGL(k_dust_build<uint32_t>, dim3(dnb), dim3(dblk),
+                           0, s, (const uint32_t *)c->labels.ptr, nvox, dlh);
+      uint32_t misc[2] = {0, 0};
+      HIP_TRY(c, hipMemcpyAsync(misc, c->lh_misc.ptr, 8,
+                                hipMemcpyDeviceToHost, s), 16);
+      HIP_TRY(c, hipStreamSynchronize(s), 16);
+      if (misc[1]) {  // hash overflow: grow and retry
+        if (c->lh_slots >= (1ull << 27)) {
+          SET_ERR(c, "label hash overflow (dust) at %llu slots",
+                  (unsigned long long)c->lh_slots);
+          return 16;
+        }
+        c->lh_slots <<= 2;
+        continue;
+      }
+      uint32_t ndl = misc[0];
+      if (ndl == 0) break;
+      if (ensure(c, c->order, (uint64_t)ndl * 4)) return 16;
+      HIP_TRY(c, hipMemsetAsync(c->order.ptr, 0, (uint64_t)ndl * 4, s), 16);
+      if (dtype == MG_U64) {
+        hipLaunchKernelGGL(k_dust_count<uint64_t>, dim3(dnb), dim3(dblk),
+                           0, s, (const uint64_t *)c->labels.ptr, nvox,
+                           dlh, (uint32_t *)c->order.ptr);
+        hipLaunchKernelGGL(k_dust_zero<uint64_t>, dim3(dnb), dim3(dblk),
+                           0, s, (uint64_t *)c->labels.ptr, nvox, dlh,
+                           (const uint32_t *)c->order.ptr, dust_threshold);
+      } else {
+        hipLaunchKernelGGL(k_dust_count<uint32_t>, dim3(dnb), dim3(dblk),
+                           0, s, (const uint32_t *)c->labels.ptr, nvox,
+                           dlh, (uint32_t *)c->order.ptr);
+        hipLaunchKernelGGL(k_dust_zero<uint32_t>, dim3(dnb), dim3(dblk),
+                           0, s, (uint32_t *)c->labels.ptr, nvox, dlh,
+                           (const uint32_t *)c->order.ptr, dust_threshold);
+      }
+      HIP_TRY(c, hipGetLastError(), 16);
+      break;
+    }
+  }
 
   // label hash (grow-and-retry on overflow)
   uint64_t total_tris = 0;

@@ -104,7 +104,8 @@ def load_library() -> ctypes.CDLL:
             ctypes.c_void_p, ctypes.c_void_p,
             ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_int,
             ctypes.c_float, ctypes.c_float, ctypes.c_float,
-            ctypes.c_uint32, ctypes.c_float, ctypes.c_int, ctypes.c_uint32,
+            ctypes.c_uint32, ctypes.c_float, ctypes.c_int,
+            ctypes.c_uint64, ctypes.c_uint32,
             ctypes.POINTER(ctypes.POINTER(_MgMeshSet)),
         ]
         lib.mg_meshset_free.argtypes = [ctypes.POINTER(_MgMeshSet)]
@@ -169,6 +170,7 @@ class Engine:
                    voxel_centered: bool = True,
                    device_only: bool = False,
                    skip_h2d: bool = False,
+                   dust_threshold: int = 0,
                    copy: bool = True) -> dict:
         """GPU counterpart of oracle.mesh_chunk: F-order (sx,sy,sz)
         uint32/uint64 labels -> {label: (verts (V,3) f32 nm, faces (F,3) u32)},
@@ -195,7 +197,8 @@ class Engine:
                 sx, sy, sz, dtype,
                 float(resolution[0]), float(resolution[1]), float(resolution[2]),
                 int(reduction_factor), float(max_error),
-                int(bool(voxel_centered)), flags, ctypes.byref(out))
+                int(bool(voxel_centered)), int(dust_threshold), flags,
+                ctypes.byref(out))
             if rc != 0:
                 err = self.lib.mg_last_error(self.ctx)
                 raise RuntimeError(
@@ -281,6 +284,7 @@ class Engine:
 def mesh_chunk(labels: np.ndarray, resolution=(1.0, 1.0, 1.0),
                reduction_factor: int = 0, max_error: float = 40.0,
                voxel_centered: bool = True, device_id: Optional[int] = None,
+               dust_threshold: int = 0,
                copy: bool = True) -> dict:
     """Module-level product mesher (the default MeshTask path)."""
     if device_id is None:
@@ -288,7 +292,12 @@ def mesh_chunk(labels: np.ndarray, resolution=(1.0, 1.0, 1.0),
                                        os.environ.get("LOCAL_RANK", "0")))
     return Engine.get(device_id).mesh_chunk(
         labels, resolution, reduction_factor, max_error, voxel_centered,
-        copy=copy)
+        dust_threshold=dust_threshold, copy=copy)
+
+
+# MeshTask checks this to delegate dust_threshold preprocessing to the
+# device (three HIP volume passes) instead of a host numpy unique/mask
+mesh_chunk.handles_dust = True
 
 
 def simplify_mesh(mesh, target_count: int, max_error: float = 1e30,

@@ -169,7 +169,18 @@ class MeshTask(RegisteredTask):
         if opts['closed_dataset_edges']:
             data, left_offset = self._handle_dataset_boundary(data, data_bounds)
 
-        data = self._remove_dust(data, opts['dust_threshold'], opts['dust_global'])
+        # dust removal: delegated to the engine's device passes when the
+        # mesher advertises support (three HIP volume passes instead of
+        # a multi-second host unique/mask at 512^3); host fallback keeps
+        # the reference semantics for injected CPU meshers
+        mesher_probe = _get_mesher()
+        device_dust = 0
+        if (opts['dust_threshold'] and not opts['dust_global']
+                and getattr(mesher_probe, 'handles_dust', False)):
+            device_dust = int(opts['dust_threshold'])
+        else:
+            data = self._remove_dust(data, opts['dust_threshold'],
+                                     opts['dust_global'])
         data = self._remap(data)
 
         if opts['object_ids']:
@@ -185,6 +196,9 @@ class MeshTask(RegisteredTask):
         # views into the engine's staging buffers are safe — EXCEPT under
         # dry_run, where the meshes are handed back to the caller and must
         # own their storage (the reference returns owned arrays).
+        extra = {}
+        if device_dust:
+            extra['dust_threshold'] = device_dust
         try:
             raw = mesher(
                 data,
@@ -193,6 +207,7 @@ class MeshTask(RegisteredTask):
                 max_error=float(opts['max_simplification_error']),
                 voxel_centered=True,
                 copy=bool(opts['dry_run']),
+                **extra,
             )
         except TypeError:
             raw = mesher(
@@ -201,6 +216,7 @@ class MeshTask(RegisteredTask):
                 reduction_factor=int(opts['simplification_factor'] or 0),
                 max_error=float(opts['max_simplification_error']),
                 voxel_centered=True,
+                **extra,
             )
         del data
         meshes = {

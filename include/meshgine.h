@@ -131,6 +131,11 @@ void     mg_destroy(mg_ctx *ctx);
  *   max_error     max simplification error in nm (mesh.py 'max_error')
  *   voxel_centered nonzero: voxel centers at integer coordinates
  *                  (mesh.py:380 voxel_centered=True)
+ *   dust_threshold 0 = off; else labels with fewer voxels than this are
+ *                  zeroed ON DEVICE before meshing (the reference's
+ *                  dust_threshold preprocessing, mesh.py:313-323 via
+ *                  fastremap — here three HIP volume passes instead of
+ *                  a multi-second host unique/mask)
  *   flags         MG_FLAG_*
  *   out           receives the meshset (caller frees)
  * Returns 0 on success. */
@@ -138,7 +143,8 @@ int mg_mesh_chunk(mg_ctx *ctx, const void *labels,
                   int sx, int sy, int sz, int dtype,
                   float rx, float ry, float rz,
                   uint32_t reduction_factor, float max_error,
-                  int voxel_centered, uint32_t flags,
+                  int voxel_centered, uint64_t dust_threshold,
+                  uint32_t flags,
                   mg_meshset **out);
 
 void mg_meshset_free(mg_meshset *ms);

@@ -1,0 +1,49 @@
+#!/usr/bin/env python3
+"""Probe the current host for any way to obtain the real zmesh wheel
+(the reference's meshing kernel) so parity could be re-anchored against
+it. Records the attempt — VERDICT r01 asks for this probe on every GPU
+lease. Writes gpurun_out/zmesh_probe.json when run under gpurun."""
+import glob
+import json
+import os
+import subprocess
+import sys
+import time
+
+
+def probe() -> dict:
+    out = {"when": time.strftime("%Y-%m-%d %H:%M:%S"),
+           "host": os.uname().nodename}
+    try:
+        import zmesh  # noqa: F401
+        out["importable"] = True
+        out["version"] = getattr(zmesh, "__version__", "?")
+        return out
+    except ImportError as e:
+        out["importable"] = False
+        out["import_error"] = str(e)
+    wheels = []
+    for root in ("/opt", "/usr/share", "/root", "/tmp", "/var/cache"):
+        wheels += glob.glob(os.path.join(root, "**", "zmesh*.whl"),
+                            recursive=True)
+    out["local_wheels"] = wheels
+    try:
+        r = subprocess.run(
+            [sys.executable, "-m", "pip", "install", "--no-input",
+             "--disable-pip-version-check", "zmesh"],
+            capture_output=True, text=True, timeout=60)
+        out["pip_rc"] = r.returncode
+        out["pip_tail"] = (r.stdout + r.stderr)[-400:]
+    except Exception as e:
+        out["pip_rc"] = -1
+        out["pip_tail"] = str(e)
+    return out
+
+
+if __name__ == "__main__":
+    result = probe()
+    os.makedirs("gpurun_out", exist_ok=True)
+    path = os.path.join("gpurun_out", "zmesh_probe.json")
+    with open(path, "w") as f:
+        json.dump(result, f, indent=1)
+    print(json.dumps(result))
