@@ -27,16 +27,17 @@ def probe() -> dict:
         wheels += glob.glob(os.path.join(root, "**", "zmesh*.whl"),
                             recursive=True)
     out["local_wheels"] = wheels
+    # NOTE: no `pip install` attempt — the GPU hosts have no network and
+    # a hung pip once consumed a whole gpurun budget slice. A cheap TCP
+    # reachability check stands in for "could pip work at all".
+    import socket
     try:
-        r = subprocess.run(
-            [sys.executable, "-m", "pip", "install", "--no-input",
-             "--disable-pip-version-check", "zmesh"],
-            capture_output=True, text=True, timeout=60)
-        out["pip_rc"] = r.returncode
-        out["pip_tail"] = (r.stdout + r.stderr)[-400:]
-    except Exception as e:
-        out["pip_rc"] = -1
-        out["pip_tail"] = str(e)
+        socket.setdefaulttimeout(3)
+        socket.create_connection(("pypi.org", 443), timeout=3).close()
+        out["network"] = True
+    except OSError as e:
+        out["network"] = False
+        out["network_error"] = str(e)
     return out
 
 
