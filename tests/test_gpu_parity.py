@@ -432,28 +432,36 @@ def test_topology_properties_at_512(eng):
     got = eng.mesh_chunk(data, resolution=res)
     lim = np.array([511 * r for r in res], dtype=np.float32)
     checked = 0
+    manifold = 0
     for lab in sorted(got.keys()):
         v, f = got[lab]
         if len(v) == 0 or v.min() <= 0.0 or np.any(v.max(axis=0) >= lim):
             continue  # touches the chunk boundary: open there by design
-        # directed edge multiset: each (a,b) exactly once, with (b,a)
+        # watertightness as directed-edge BALANCE: the multiset of
+        # directed edges equals its reversal (regions that touch
+        # themselves across a voxel edge/corner produce legitimate
+        # non-manifold edges with multiplicity 2, so exact-once is too
+        # strong there)
         e = np.concatenate([f[:, [0, 1]], f[:, [1, 2]], f[:, [2, 0]]])
         keys = e[:, 0].astype(np.int64) * len(v) + e[:, 1]
         rkeys = e[:, 1].astype(np.int64) * len(v) + e[:, 0]
         uk, counts = np.unique(keys, return_counts=True)
-        assert counts.max() == 1, f"label {lab}: repeated directed edge"
-        assert np.array_equal(uk, np.unique(rkeys)), \
-            f"label {lab}: unpaired directed edge (not watertight)"
-        E = len(keys) // 2
-        F = len(f)
-        assert 2 * E == 3 * F, f"label {lab}: 2E != 3F"
-        chi = len(v) - E + F
-        assert chi % 2 == 0 and chi <= 2, \
-            f"label {lab}: Euler characteristic {chi}"
+        ruk, rcounts = np.unique(rkeys, return_counts=True)
+        assert np.array_equal(uk, ruk) and np.array_equal(counts, rcounts), \
+            f"label {lab}: unbalanced directed edges (not watertight)"
+        if counts.max() == 1:  # manifold label: full checks apply
+            E = len(keys) // 2
+            F = len(f)
+            assert 2 * E == 3 * F, f"label {lab}: 2E != 3F"
+            chi = len(v) - E + F
+            assert chi % 2 == 0 and chi <= 2, \
+                f"label {lab}: Euler characteristic {chi}"
+            manifold += 1
         checked += 1
         if checked >= 300:
             break
     assert checked >= 100
+    assert manifold >= 50
 
 
 def test_device_dust_parity(eng):
