@@ -453,9 +453,12 @@ def test_topology_properties_at_512(eng):
             E = len(keys) // 2
             F = len(f)
             assert 2 * E == 3 * F, f"label {lab}: 2E != 3F"
+            # chi = sum over closed components of (2 - 2*genus):
+            # always EVEN; can exceed 2 when background carving splits
+            # a label into several components
             chi = len(v) - E + F
-            assert chi % 2 == 0 and chi <= 2, \
-                f"label {lab}: Euler characteristic {chi}"
+            assert chi % 2 == 0, \
+                f"label {lab}: odd Euler characteristic {chi}"
             manifold += 1
         checked += 1
         if checked >= 300:
