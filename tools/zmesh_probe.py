@@ -27,17 +27,13 @@ def probe() -> dict:
         wheels += glob.glob(os.path.join(root, "**", "zmesh*.whl"),
                             recursive=True)
     out["local_wheels"] = wheels
-    # NOTE: no `pip install` attempt — the GPU hosts have no network and
-    # a hung pip once consumed a whole gpurun budget slice. A cheap TCP
-    # reachability check stands in for "could pip work at all".
-    import socket
-    try:
-        socket.setdefaulttimeout(3)
-        socket.create_connection(("pypi.org", 443), timeout=3).close()
-        out["network"] = True
-    except OSError as e:
-        out["network"] = False
-        out["network_error"] = str(e)
+    # NOTE: no `pip install` and no socket/DNS attempt — the GPU hosts
+    # are airgapped and both a hung pip and a blackholed getaddrinfo
+    # have eaten gpurun budget. Import + wheel search above is the
+    # entire offline probe; 'network' is recorded as policy knowledge.
+    out["network"] = False
+    out["network_note"] = ("gpurun hosts are airgapped by policy; no "
+                           "install path exists for the real zmesh")
     return out
 
 
