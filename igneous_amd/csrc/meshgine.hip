@@ -437,7 +437,9 @@ template <int BLK, int TPT>
 __global__ __launch_bounds__(BLK) void k_weld_insert(
     const uint2 *__restrict__ tri_recs,   // emit order (8-B records)
     const uint32_t *__restrict__ order,   // label partition permutation
-    uint2 *__restrict__ recs_sorted,      // gathered here (fused pass)
+    uint32_t *__restrict__ slots_sorted,  // decoded 3 slots/tri (12 B) —
+                                          // downstream weld passes read
+                                          // plain slots, no case tables
     uint32_t *__restrict__ wminp,
     int64_t sx, int64_t sxy,
     uint64_t ntris) {
@@ -462,9 +464,11 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
     uint64_t t = span0 + (uint64_t)rep * BLK + threadIdx.x;
     if (t >= ntris) break;
     uint2 rec = tri_recs[order[t]];
-    recs_sorted[t] = rec;
     uint32_t s[3];
     decode_rec_slots(rec, s_pack, s_comb, s);
+    slots_sorted[3 * t] = s[0];
+    slots_sorted[3 * t + 1] = s[1];
+    slots_sorted[3 * t + 2] = s[2];
     #pragma unroll
     for (int v = 0; v < 3; ++v) {
       uint32_t slot = s[v];
@@ -496,26 +500,18 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
 // consecutive corners and ballots the flags into one u64 word. Vertex
 // ids then come from a word-granular scan + popcount instead of a full
 // per-element scan array.
-__global__ void k_weld_flag_bits(const uint2 *__restrict__ recs_sorted,
+__global__ void k_weld_flag_bits(const uint32_t *__restrict__ slots_sorted,
                                  const uint32_t *__restrict__ wminp,
                                  unsigned long long *__restrict__ bits,
-                                 int64_t sx, int64_t sxy,
                                  uint64_t ncorners, uint64_t nwords) {
-  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
-  __shared__ uint32_t s_comb[12];
-  stage_decode_tables(s_pack, s_comb, sx, sxy);
-  __syncthreads();
   uint64_t gid = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   uint64_t word = gid / 64;
   if (word >= nwords) return;
   uint32_t lane = (uint32_t)(gid & 63);
   uint64_t i = word * 64 + lane;
   bool flag = false;
-  if (i < ncorners) {
-    uint32_t slot = decode_rec_slot1(recs_sorted[i / 3], s_pack, s_comb,
-                                     (uint32_t)(i % 3));
-    flag = (wminp[slot] == ~(uint32_t)i);
-  }
+  if (i < ncorners)
+    flag = (wminp[slots_sorted[i]] == ~(uint32_t)i);
   unsigned long long m = __ballot(flag);
   if (lane == 0) bits[word] = m;
 }
@@ -543,7 +539,7 @@ __global__ void k_total_verts(const uint32_t *__restrict__ wscan,
 
 // [5d] first occurrences: record vertex id in the table, write the vertex
 // (doubled coordinates = 2*cell + edge offset, decoded from the record)
-__global__ void k_weld_verts(const uint2 *__restrict__ recs_sorted,
+__global__ void k_weld_verts(const uint32_t *__restrict__ slots_sorted,
                              const uint32_t *__restrict__ wscan,
                              const unsigned long long *__restrict__ bits,
                              const uint32_t *__restrict__ wminp,
@@ -552,38 +548,29 @@ __global__ void k_weld_verts(const uint2 *__restrict__ recs_sorted,
                              uint32_t usx, uint32_t usxy,
                              float rx, float ry, float rz, float shift,
                              uint64_t ntris) {
-  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
-  __shared__ uint32_t s_comb[12];
-  __shared__ uint8_t s_doff[12][3];
-  stage_decode_tables(s_pack, s_comb, usx, usxy);
-  if (threadIdx.x < 12) {
-    #pragma unroll
-    for (int k = 0; k < 3; ++k)
-      s_doff[threadIdx.x][k] = MC_EDGE_DOFF[threadIdx.x][k];
-  }
-  __syncthreads();
   uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= ntris) return;
-  uint2 rec = recs_sorted[t];
-  const uint32_t cl3 = rec.x * 3u;
-  const uint32_t pk =
-      s_pack[(rec.y & 255u) * MC_MAX_TRIS + (rec.y >> 8)];
-  uint32_t cz = rec.x / usxy;
-  uint32_t rem = rec.x - cz * usxy;
-  uint32_t cy = rem / usx;
-  uint32_t cx = rem - cy * usx;
+  const uint32_t s[3] = {slots_sorted[3 * t], slots_sorted[3 * t + 1],
+                         slots_sorted[3 * t + 2]};
   #pragma unroll
   for (int v = 0; v < 3; ++v) {
-    uint32_t nib = (pk >> (5 * v)) & 31u;  // 4-bit edge + 1-bit side
-    uint32_t e = nib & 15u;
-    uint32_t slot = ((cl3 + s_comb[e]) << 1) | (nib >> 4);
     uint32_t i = (uint32_t)(3 * t + v);
+    uint32_t slot = s[v];
     if (wminp[slot] != ~i) continue;
     uint32_t vid = vtx_id_of(wscan, bits, i);
     wvtx[slot] = vid;
-    float dx = (float)(2 * cx + s_doff[e][0]);
-    float dy = (float)(2 * cy + s_doff[e][1]);
-    float dz = (float)(2 * cz + s_doff[e][2]);
+    // doubled coordinates decoded from the slot (first occurrences
+    // only, ~1/6 of corners): slot = (lin*3 + axis)*2 | side
+    uint32_t eslot = slot >> 1;
+    uint32_t axis = eslot % 3u;
+    uint32_t lin = eslot / 3u;
+    uint32_t vz = lin / usxy;
+    uint32_t rem = lin - vz * usxy;
+    uint32_t vy = rem / usx;
+    uint32_t vx = rem - vy * usx;
+    float dx = (float)(2 * vx + (axis == 0));
+    float dy = (float)(2 * vy + (axis == 1));
+    float dz = (float)(2 * vz + (axis == 2));
     verts[3ull * vid + 0] = (0.5f * dx + shift) * rx;
     verts[3ull * vid + 1] = (0.5f * dy + shift) * ry;
     verts[3ull * vid + 2] = (0.5f * dz + shift) * rz;
@@ -604,25 +591,18 @@ __global__ void k_vbase(const uint32_t *__restrict__ tri_off,
 
 // [5f] faces: per-label local vertex indices (label id from the sorted
 // label array — the partition sort's key output)
-__global__ void k_faces(const uint2 *__restrict__ recs_sorted,
+__global__ void k_faces(const uint32_t *__restrict__ slots_sorted,
                         const uint32_t *__restrict__ lab_sorted,
                         const uint32_t *__restrict__ wvtx,
                         const uint32_t *__restrict__ vbase,
                         uint32_t *__restrict__ faces,
-                        int64_t sx, int64_t sxy,
                         uint64_t ntris) {
-  __shared__ uint16_t s_pack[256 * MC_MAX_TRIS];
-  __shared__ uint32_t s_comb[12];
-  stage_decode_tables(s_pack, s_comb, sx, sxy);
-  __syncthreads();
   uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (t >= ntris) return;
-  uint32_t s[3];
-  decode_rec_slots(recs_sorted[t], s_pack, s_comb, s);
   uint32_t base = vbase[lab_sorted[t]];
-  faces[3 * t + 0] = wvtx[s[0]] - base;
-  faces[3 * t + 1] = wvtx[s[1]] - base;
-  faces[3 * t + 2] = wvtx[s[2]] - base;
+  faces[3 * t + 0] = wvtx[slots_sorted[3 * t]] - base;
+  faces[3 * t + 1] = wvtx[slots_sorted[3 * t + 1]] - base;
+  faces[3 * t + 2] = wvtx[slots_sorted[3 * t + 2]] - base;
 }
 
 __global__ void k_iota(uint32_t *p, uint64_t n) {
@@ -1645,7 +1625,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   if (ensure(c, c->order, T * 4)) return 19;
   if (ensure(c, c->order_alt, T * 4)) return 19;
   if (ensure(c, c->tri_label_alt, T * 4)) return 19;
-  if (ensure(c, c->keys_sorted, T * 8)) return 19;
+  if (ensure(c, c->keys_sorted, T * 12)) return 19;  // decoded slot triples
   uint32_t *order_sorted = nullptr;
   uint32_t *lab_sorted = nullptr;  // sort key output: label id per tri
   {
@@ -1696,13 +1676,13 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0, wslots * 4, s), 20);
   uint32_t *wminp = (uint32_t *)c->wh_keys.ptr;
   uint32_t *wvtx = (uint32_t *)c->wh_vtx.ptr;
-  const uint2 *recs_sorted = (const uint2 *)c->keys_sorted.ptr;
+  const uint32_t *slots_sorted = (const uint32_t *)c->keys_sorted.ptr;
   {
     int blk = 256;
     uint64_t nbt = (T + blk - 1) / blk;
     int wi_cfg = 1;  // 0: 1024x1, 1: 1024x2, 2: 1024x4
     if (const char *e = getenv("MG_WELD_INSERT_CFG")) wi_cfg = atoi(e);
-    uint2 *rs_mut = (uint2 *)c->keys_sorted.ptr;
+    uint32_t *rs_mut = (uint32_t *)c->keys_sorted.ptr;
     const uint2 *tr = (const uint2 *)c->tri_keys.ptr;
     const int64_t sxy64 = g.sx * g.sy;
     if (wi_cfg == 0) {
@@ -1730,8 +1710,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
     {
       uint64_t nbw = (nwords * 64 + blk - 1) / blk;
       hipLaunchKernelGGL(k_weld_flag_bits, dim3((uint32_t)nbw), dim3(blk),
-                         0, s, recs_sorted, wminp, bits, g.sx, sxy64,
-                         NC, nwords);
+                         0, s, slots_sorted, wminp, bits, NC, nwords);
     }
     auto it = rocprim::make_transform_iterator(bits, PopcWord{});
     size_t tmp_bytes = 0;
@@ -1767,7 +1746,7 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
         (const unsigned long long *)c->vtx_scan.ptr;
     const uint32_t *wscan = (const uint32_t *)c->vtx_scan.ptr + 2 * nwords;
     hipLaunchKernelGGL(k_weld_verts, dim3((uint32_t)nbt), dim3(blk), 0, s,
-                       recs_sorted, wscan, bits, wminp, wvtx,
+                       slots_sorted, wscan, bits, wminp, wvtx,
                        (float *)c->verts.ptr,
                        (uint32_t)g.sx, (uint32_t)(g.sx * g.sy),
                        rx, ry, rz, shift, T);
@@ -1777,9 +1756,9 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
                        wscan, bits,
                        (uint32_t *)c->vbase.ptr, nlabels, total_verts);
     hipLaunchKernelGGL(k_faces, dim3((uint32_t)nbt), dim3(blk), 0, s,
-                       recs_sorted, lab_sorted, wvtx,
+                       slots_sorted, lab_sorted, wvtx,
                        (const uint32_t *)c->vbase.ptr,
-                       (uint32_t *)c->faces.ptr, g.sx, g.sx * g.sy, T);
+                       (uint32_t *)c->faces.ptr, T);
   }
   HIP_TRY(c, hipGetLastError(), 22);
   HIP_TRY(c, hipEventRecord(c->ev[6], s), 22);
