@@ -981,9 +981,10 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     // MG_SIMP_SMALL=1 enables for experiments.
     const char *sm = getenv("MG_SIMP_SMALL");
     const uint32_t small_cap = (sm && sm[0] == '1') ? 512u : 0u;
-    auto launch_band = [&](decltype(&k_simplify_label<false, 256, 2048>) fn,
-                           hipStream_t st, int bsz,
-                           uint32_t nv_lo, uint32_t nv_hi) {
+    auto launch_band = [&](auto fn, hipStream_t st, int bsz,
+                           uint32_t nv_lo, uint32_t nv_hi,
+                           uint32_t nt_lo = 0u,
+                           uint32_t nt_hi = 0xFFFFFFFFu) {
       hipLaunchKernelGGL(fn, dim3((uint32_t)L), dim3(bsz), 0, st,
                          faces_g, (uint32_t *)c->simp_faces_alt.ptr,
                          (const uint32_t *)c->tri_off.ptr,
@@ -1003,7 +1004,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                              : nullptr,
                          d_rh,
                          max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs,
-                         nv_lo, nv_hi, d_sched);
+                         nv_lo, nv_hi, nt_lo, nt_hi, d_sched);
     };
     if (small_cap)
       launch_band(k_simplify_label<false, 64, 512>, s, 64, 0u, small_cap);
@@ -1016,11 +1017,23 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     const bool bands = !(bandsenv && bandsenv[0] == '0') && !use_cl;
     if (bands) {
       // record BEFORE enqueuing band A: stream2 must wait only for the
-      // shared setup, so bands B/C run CONCURRENT with A (recording
-      // after A serialized them — measured 51 ms vs 38 ms overlapped)
+      // shared setup, so the other bands run CONCURRENT with A
+      // (recording after A serialized them — 51 ms vs 38 ms overlapped)
       HIP_TRY(c, hipEventRecord(c->ev[9], s), 40);
       HIP_TRY(c, hipStreamWaitEvent(c->stream2, c->ev[9], 0), 40);
-      launch_band(ksl, s, bs, small_cap, 2048u);
+      // WAVEMODE band (MG_SIMP_WAVE=0 disables): one wave per label
+      // for the dominant small-label mass (nt<=8192, nv<=2048) — 20 KB
+      // LDS/label -> ~7 resident labels/CU vs 3, barrier-free loop
+      const char *wenv = getenv("MG_SIMP_WAVE");
+      const bool wave = !(wenv && wenv[0] == '0');
+      if (wave) {
+        launch_band(k_simplify_label<false, 64, 2048, true>, s, 64,
+                    small_cap, 2048u, 0u, 8192u);
+        launch_band(ksl, c->stream2, bs, small_cap, 2048u,
+                    8192u, 0xFFFFFFFFu);
+      } else {
+        launch_band(ksl, s, bs, small_cap, 2048u);
+      }
       launch_band(k_simplify_label<false, 256, 4096>, c->stream2, 256,
                   2048u, 4096u);
       launch_band(ksl, c->stream2, bs, 4096u, 0xFFFFFFFFu);

@@ -555,24 +555,24 @@ __device__ uint32_t blk_exscan(const uint32_t *src, uint32_t *dst,
 // the (cheap, LDS) rm gathers instead of paying a 12 B/face global
 // write + re-read. Identical output order to the old rewrite + compact
 // pair.
-template <int BS>
+template <int BS, int MAXN, typename RMT>
 __device__ uint32_t blk_rewrite_compact(const uint32_t *__restrict__ fa,
                                         uint32_t *__restrict__ fb,
-                                        const uint32_t *__restrict__ rm,
+                                        const RMT *__restrict__ rm,
                                         uint32_t v0,
                                         uint32_t n, uint32_t *s_sums) {
   const uint32_t tid = threadIdx.x;
   const uint32_t chunk = (n + BS - 1) / BS;
   const uint32_t lo = tid * chunk;
   const uint32_t hi = lo + chunk < n ? lo + chunk : n;
-  // chunk <= big_cap/BS; bitmask register array sized for big_cap=65536
-  constexpr uint32_t MAXW = (65536u / BS + 63) / 64 + 1;
+  // bitmask register array sized for the band's face-count cap MAXN
+  constexpr uint32_t MAXW = ((uint32_t)MAXN / BS + 63) / 64 + 1;
   unsigned long long bm[MAXW] = {};
   uint32_t sum = 0;
   for (uint32_t i = lo; i < hi; ++i) {
-    uint32_t i0 = rm[fa[3*i] - v0];
-    uint32_t i1 = rm[fa[3*i+1] - v0];
-    uint32_t i2 = rm[fa[3*i+2] - v0];
+    uint32_t i0 = (uint32_t)rm[fa[3*i] - v0];
+    uint32_t i1 = (uint32_t)rm[fa[3*i+1] - v0];
+    uint32_t i2 = (uint32_t)rm[fa[3*i+2] - v0];
     if (i0 != i1 && i1 != i2 && i0 != i2) {
       bm[(i - lo) >> 6] |= 1ull << ((i - lo) & 63);
       ++sum;
@@ -582,9 +582,9 @@ __device__ uint32_t blk_rewrite_compact(const uint32_t *__restrict__ fa,
   uint32_t run = blk_prefix<BS>(sum, s_sums, &total);
   for (uint32_t i = lo; i < hi; ++i)
     if (bm[(i - lo) >> 6] & (1ull << ((i - lo) & 63))) {
-      fb[3*run] = v0 + rm[fa[3*i] - v0];
-      fb[3*run+1] = v0 + rm[fa[3*i+1] - v0];
-      fb[3*run+2] = v0 + rm[fa[3*i+2] - v0];
+      fb[3*run] = v0 + (uint32_t)rm[fa[3*i] - v0];
+      fb[3*run+1] = v0 + (uint32_t)rm[fa[3*i+1] - v0];
+      fb[3*run+2] = v0 + (uint32_t)rm[fa[3*i+2] - v0];
       ++run;
     }
   __syncthreads();  // s_sums reusable after this
@@ -601,7 +601,13 @@ __device__ uint32_t blk_rewrite_compact(const uint32_t *__restrict__ fa,
 #ifndef SIMP_D16_TIER
 #define SIMP_D16_TIER 0
 #endif
-template <bool CLLDS, int BS, int CAPVT>
+// WAVEMODE (BS=64, one wave per label): the whole round loop runs
+// wave-synchronously — every __syncthreads() in a 64-thread block
+// lowers to a waitcnt, the pick table ALIASES the (phase-disjoint)
+// degree/cursor array, and remap is u16 — 20 KB LDS per label instead
+// of 33, so ~7 labels are resident per CU instead of 3. Owns the
+// (nt_lo, nt_hi] x (nv_lo, nv_hi] band its launch names.
+template <bool CLLDS, int BS, int CAPVT, bool WAVEMODE = false>
 __global__ __launch_bounds__(BS)
 __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     uint32_t *__restrict__ faces_g,       // slices at 3*tri_off[b]
@@ -625,6 +631,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     uint32_t *__restrict__ roundhist,  // 160 u32 or null (see host)
     float max_cost, uint32_t nlabels, uint32_t big_cap,
     uint32_t subs, uint32_t nv_lo, uint32_t nv_hi,
+    uint32_t nt_lo, uint32_t nt_hi,
     const uint32_t *__restrict__ sched /*block->label, largest first*/) {
   if (blockIdx.x >= nlabels) return;
   // biggest-label-first dispatch: per-label serial time scales with nt0,
@@ -638,9 +645,10 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
   if (nt0 > big_cap) return;  // global-rounds path handles big labels
   const uint32_t v0 = vbase[b];
   const uint32_t nv = vbase[b + 1] - v0;
-  // size-class dispatch: each launch variant owns an (nv_lo, nv_hi]
-  // vertex-count band (single-wave small-LDS variant for tiny labels)
+  // size-class dispatch: each launch variant owns an (nv_lo, nv_hi] x
+  // (nt_lo, nt_hi] band
   if (nv <= nv_lo || nv > nv_hi) return;
+  if (nt0 <= nt_lo || nt0 > nt_hi) return;
   if (!active[b]) {
     // already at/below target: final faces = original faces; park them
     for (uint32_t i = threadIdx.x; i < 3 * nt0; i += BS)
@@ -662,7 +670,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
   // atomicAdd/atomicMin streams serialize on hot L2 lines — LDS atomics
   // are bank-parallel. (~25 KB -> ~6 blocks/CU.)
   constexpr uint32_t CAPV = CAPVT;
-  __shared__ uint32_t s_deg[CAPV];
+  __shared__ uint32_t s_deg[WAVEMODE ? 1 : CAPV];
   __shared__ unsigned long long s_pick[CAPV];
   // optional LDS-resident CSR payload (face ids, u16) for rounds whose
   // face count fits; costs 24 KB LDS -> fewer blocks/CU, so env-gated
@@ -670,13 +678,20 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
   constexpr uint32_t CAPF = CLLDS ? 4096u : 1u;
   __shared__ uint16_t s_cl[3 * CAPF];
   // remap holds label-LOCAL canonical ids; LDS-resident when the label
-  // fits (rewrite's 3 gathers/face are the hot readers)
-  __shared__ uint32_t s_remap[CAPV];
+  // fits (rewrite's 3 gathers/face are the hot readers). WAVEMODE packs
+  // it to u16 (label-local ids < CAPV).
+  using RMT = typename std::conditional<WAVEMODE, uint16_t,
+                                        uint32_t>::type;
+  __shared__ RMT s_remap[CAPV];
   const bool lds_mode = (nv <= CAPV);
-  uint32_t *dg = lds_mode ? s_deg : (deg + v0);
+  // WAVEMODE: the degree/cursor array lives in the PICK bytes — deg is
+  // dead before the first pick write of each group (the reset pass
+  // below runs after the last cursor read, behind a barrier)
+  uint32_t *dg = WAVEMODE ? (uint32_t *)s_pick
+                          : (lds_mode ? s_deg : (deg + v0));
   unsigned long long *pick_l =
       lds_mode ? s_pick : (pick + v0);
-  uint32_t *rm = lds_mode ? s_remap : (remap + v0);
+  RMT *rm = lds_mode ? s_remap : (RMT *)(remap + v0);
   // ping-pong face buffers: rewrite reads fa, compaction scatters into
   // fb, then the buffers swap — no copy-back pass. Parking at the end
   // reads whichever buffer is current.
@@ -862,9 +877,16 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
         qo[1] = make_float4(q[4], q[5], q[6], q[7]);
         qo[2] = make_float4(q[8], q[9], 0.0f, 0.0f);
       }
-      pick_l[v] = ~0ull;  // fused pick reset (same-thread slot)
+      if (!WAVEMODE)
+        pick_l[v] = ~0ull;  // fused pick reset (same-thread slot)
     }
     __syncthreads();
+    if (WAVEMODE) {
+      // dg (aliasing the pick bytes) had its last read above; now the
+      // pick table takes the space back
+      for (uint32_t v = tid; v < nv; v += BS) pick_l[v] = ~0ull;
+      __syncthreads();
+    }
     PHASE_MARK(3)  // sort + quadric accumulate
     // sub-rounds: reuse merged quadrics (Q[u] += Q[w] on collapse) for
     // up to `subs` pick/collapse/compact passes per recompute (oracle
@@ -955,7 +977,8 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     __syncthreads();
     if (s_collapses == 0) break;
     // [8] fused rewrite + stable compact (oracle step 4)
-    uint32_t kept = blk_rewrite_compact<BS>(fa, fb, rm, v0, nt, s_sums);
+    uint32_t kept = blk_rewrite_compact<BS, WAVEMODE ? 8192 : 65536, RMT>(
+        fa, fb, rm, v0, nt, s_sums);
     { uint32_t *t = fa; fa = fb; fb = t; }  // compacted faces now in fa
     if (tid == 0) s_nt = kept;
     __syncthreads();
