@@ -1021,11 +1021,13 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       // (recording after A serialized them — 51 ms vs 38 ms overlapped)
       HIP_TRY(c, hipEventRecord(c->ev[9], s), 40);
       HIP_TRY(c, hipStreamWaitEvent(c->stream2, c->ev[9], 0), 40);
-      // WAVEMODE band (MG_SIMP_WAVE=0 disables): one wave per label
-      // for the dominant small-label mass (nt<=8192, nv<=2048) — 20 KB
-      // LDS/label -> ~7 resident labels/CU vs 3, barrier-free loop
+      // WAVEMODE band (MG_SIMP_WAVE=1 enables; default OFF — measured
+      // 19% WORSE on 512^3/50k: 7 labels/CU resident vs 3, but one
+      // 64-lane wave pays 4x the per-pass iterations and its in-flight
+      // load window can't cover the longer chains): one wave per label
+      // for the small-label mass (nt<=8192, nv<=2048), 20 KB LDS/label
       const char *wenv = getenv("MG_SIMP_WAVE");
-      const bool wave = !(wenv && wenv[0] == '0');
+      const bool wave = (wenv && wenv[0] == '1');
       if (wave) {
         launch_band(k_simplify_label<false, 64, 2048, true>, s, 64,
                     small_cap, 2048u, 0u, 8192u);
