@@ -9,4 +9,6 @@ from .mesh import (
 from .multires import (
     MultiResShardedMeshMergeTask,
     MultiResUnshardedMeshMergeTask,
+    MultiResShardedFromUnshardedMeshMergeTask,
 )
+from .spatial_index import SpatialIndexTask

@@ -50,6 +50,7 @@ from ..volume import PrecomputedVolume
 __all__ = [
     "MultiResShardedMeshMergeTask",
     "MultiResUnshardedMeshMergeTask",
+    "MultiResShardedFromUnshardedMeshMergeTask",
     "process_mesh",
     "generate_lods",
     "set_simplifier",
@@ -449,3 +450,45 @@ def MultiResShardedMeshMergeTask(
         content_type='application/octet-stream',
         cache_control='no-cache',
     )
+
+
+def MultiResShardedFromUnshardedMeshMergeTask(
+        src: str,
+        dest: str,
+        shard_no: str,
+        cache_control: bool = False,
+        draco_compression_level: int = 1,
+        mesh_dir: Optional[str] = None,
+        num_lod: int = 1,
+        progress: bool = False,
+        min_chunk_size=(512, 512, 512)):
+    """(multires.py:262-306): convert an existing UNSHARDED legacy mesh
+    layer into sharded multires — gather each shard label's legacy
+    fragment files from src, shard-merge into dest."""
+    svol, smesh_dir, _ = _mesh_meta(src, mesh_dir)
+    dvol, dmesh_dir, dmesh_info = _mesh_meta(dest, mesh_dir or smesh_dir)
+    spec = sharding.ShardingSpecification.from_dict(dmesh_info["sharding"])
+
+    cf_dest = CloudFiles(f"{dest.rstrip('/')}/{dmesh_dir}")
+    labels = cf_dest.get_json(shard_no + ".labels") or []
+
+    cf_src = CloudFiles(f"{src.rstrip('/')}/{smesh_dir}")
+    files_per_label = get_mesh_filenames_subset(src, smesh_dir, "")
+    meshes = {}
+    for label in labels:
+        fnames = sorted(files_per_label.get(int(label), []))
+        frags = [cf_src.get(fn) for fn in fnames]
+        frags = [Mesh.from_precomputed(f) for f in frags if f]
+        if not frags:
+            continue
+        meshes[int(label)] = meshops.consolidate(
+            Mesh.concatenate(*frags, id=int(label)))
+
+    fname, shard = create_mesh_shard(
+        dvol, dmesh_info, meshes, num_lod, draco_compression_level,
+        shard_no, min_chunk_size)
+    if shard is None:
+        return
+    cf_dest.put(fname, shard, compress=False,
+                content_type='application/octet-stream',
+                cache_control='no-cache')

@@ -218,3 +218,67 @@ def test_meshtask_draco_encoding(tmp_layer_path, oracle_mesher):
     assert verts[:, 0].max() <= 4 * 63 + 4.1
     assert verts[:, 2].max() <= 40 * 63 + 40.1
     assert len(faces) > 0
+
+
+def test_spatial_index_task(tmp_layer_path):
+    """Standalone SpatialIndexTask (reference tasks/spatial_index.py):
+    rebuild the .spatial files without MeshTask, precision-formatted."""
+    from igneous_amd.tasks import SpatialIndexTask
+    data = np.zeros((64, 64, 64), dtype=np.uint32)
+    data[2:20, 2:20, 2:20] = 9
+    data[30:60, 30:60, 30:60] = 12
+    PrecomputedVolume.from_numpy(
+        data, tmp_layer_path, resolution=(4, 4, 40),
+        chunk_size=(64, 64, 64), mesh_dir="mesh")
+    SpatialIndexTask(tmp_layer_path, shape=(64, 64, 64), offset=(0, 0, 0),
+                     subdir="mesh", precision=0, mip=0)
+    cf = CloudFiles(tmp_layer_path)
+    names = [n for n in cf.list("mesh/") if n.endswith(".spatial")]
+    assert len(names) == 1
+    content = cf.get_json(names[0])
+    assert set(content.keys()) == {"9", "12"}
+    # bbox of label 9 in nm: voxels [2,20) * resolution
+    assert content["9"] == [8, 8, 80, 80, 80, 800]
+
+
+def test_sharded_from_unsharded_merge(tmp_layer_path, oracle_mesher,
+                                      oracle_simplifier, tmp_path):
+    """MultiResShardedFromUnshardedMeshMergeTask: legacy unsharded
+    fragments in src -> multires shard files in dest."""
+    from igneous_amd.formats import sharding as _sh
+    from igneous_amd.tasks import MultiResShardedFromUnshardedMeshMergeTask
+    from igneous_amd.task_creation import configure_multires_info
+    _make_two_chunk_layer(tmp_layer_path)
+    for t in create_meshing_tasks(
+            tmp_layer_path, mip=0, shape=(64, 64, 64), sharded=False,
+            spatial_index=False, simplification=False):
+        t.execute()
+
+    dest = f"file://{tmp_path}/dest"
+    data = np.zeros((4, 4, 4), dtype=np.uint64)
+    PrecomputedVolume.from_numpy(
+        data, dest, resolution=(4, 4, 40), chunk_size=(4, 4, 4),
+        mesh_dir="mesh")
+    mesh_info = configure_multires_info(dest, 16, "mesh")
+    spec = _sh.ShardingSpecification(
+        preshift_bits=0, minishard_bits=2, shard_bits=1,
+        hash="murmurhash3_x86_128", minishard_index_encoding="gzip",
+        data_encoding="raw")
+    mesh_info["sharding"] = spec.to_dict()
+    cf_dest = CloudFiles(dest)
+    cf_dest.put_json("mesh/info", mesh_info)
+    shard_labels = _sh.assign_labels_to_shards(
+        np.array([5, 77], dtype=np.uint64), 0, 1, 2)
+    for shardno, labels in shard_labels.items():
+        cf_dest.put_json(f"mesh/{shardno}.labels", labels)
+        MultiResShardedFromUnshardedMeshMergeTask(
+            tmp_layer_path, dest, shardno, num_lod=0,
+            min_chunk_size=(16, 16, 16))
+    shard_files = [n for n in cf_dest.list("mesh/")
+                   if n.endswith(".shard")]
+    assert shard_files
+    reader = _sh.ShardReader(spec,
+                             lambda n: cf_dest.get(f"mesh/{n}"))
+    for label in (5, 77):
+        man = reader.get(label)
+        assert man is not None and len(man) > 0
