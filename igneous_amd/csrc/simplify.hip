@@ -599,7 +599,7 @@ __device__ uint32_t blk_rewrite_compact(const uint32_t *__restrict__ fa,
 // — degree>8 vertices are rare enough that the insertion-sort fallback
 // plus the extra wave slot wins (A/B'd on the 512^3/50k config).
 #ifndef SIMP_D16_TIER
-#define SIMP_D16_TIER 0
+#define SIMP_D16_TIER 1
 #endif
 // WAVEMODE (BS=64, one wave per label): the whole round loop runs
 // wave-synchronously — every __syncthreads() in a 64-thread block
@@ -877,14 +877,19 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
         qo[1] = make_float4(q[4], q[5], q[6], q[7]);
         qo[2] = make_float4(q[8], q[9], 0.0f, 0.0f);
       }
-      if (!WAVEMODE)
+      if (!WAVEMODE) {
         pick_l[v] = ~0ull;  // fused pick reset (same-thread slot)
+        rm[v] = (RMT)v;     // fused remap identity (consumed in collapse)
+      }
     }
     __syncthreads();
     if (WAVEMODE) {
       // dg (aliasing the pick bytes) had its last read above; now the
       // pick table takes the space back
-      for (uint32_t v = tid; v < nv; v += BS) pick_l[v] = ~0ull;
+      for (uint32_t v = tid; v < nv; v += BS) {
+        pick_l[v] = ~0ull;
+        rm[v] = (RMT)v;
+      }
       __syncthreads();
     }
     PHASE_MARK(3)  // sort + quadric accumulate
@@ -896,7 +901,10 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     if (nt <= tgt) break;
     ++n_subs;
     if (sub > 0) {
-      for (uint32_t v = tid; v < nv; v += BS) pick_l[v] = ~0ull;
+      for (uint32_t v = tid; v < nv; v += BS) {
+        pick_l[v] = ~0ull;
+        rm[v] = (RMT)v;
+      }
       __syncthreads();
     }
     // [6] picks (oracle step 2). Q rows are 12 floats (48 B, 16-B
@@ -945,9 +953,9 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       }
     }
     PHASE_MARK(4)  // edge picks
-    // [7] matched-pair collapse (oracle step 3)
+    // [7] matched-pair collapse (oracle step 3); rm was pre-set to the
+    // identity in the same pass that reset the pick table
     if (tid == 0) s_collapses = 0;
-    for (uint32_t v = tid; v < nv; v += BS) rm[v] = v;
     __syncthreads();
     for (uint32_t v = tid; v < nv; v += BS) {
       uint32_t u = v0 + v;
