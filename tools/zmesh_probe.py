@@ -22,10 +22,13 @@ def probe() -> dict:
     except ImportError as e:
         out["importable"] = False
         out["import_error"] = str(e)
+    # bounded search: shallow patterns only (a recursive glob over the
+    # box's filesystem once stalled a suite run)
     wheels = []
-    for root in ("/opt", "/usr/share", "/root", "/tmp", "/var/cache"):
-        wheels += glob.glob(os.path.join(root, "**", "zmesh*.whl"),
-                            recursive=True)
+    for pat in ("/opt/*.whl", "/opt/wheels/*.whl", "/opt/wheelhouse/*.whl",
+                "/root/*.whl", "/tmp/*.whl", "/var/cache/pip/*.whl"):
+        wheels += [w for w in glob.glob(pat)
+                   if "zmesh" in os.path.basename(w)]
     out["local_wheels"] = wheels
     # NOTE: no `pip install` and no socket/DNS attempt — the GPU hosts
     # are airgapped and both a hung pip and a blackholed getaddrinfo
