@@ -68,7 +68,9 @@ def voronoi_labels(shape, K: int, seed: int, dtype=np.uint64,
         out[:, :, z0:z1] = lab.reshape(
             (shape[0], shape[1], nz), order="F")
     if cache_path is not None:
-        tmp = cache_path + ".tmp.npy"
+        # pid-unique tmp: concurrent ranks generating the same chunk
+        # (driver 8-GPU scale runs) must not tear each other's writes
+        tmp = f"{cache_path}.tmp{os.getpid()}.npy"
         np.save(tmp, out)
         os.replace(tmp, cache_path)
     return out
