@@ -156,7 +156,7 @@ def run_chunk512(eng, data, steps, warmup, simplify, barrier_sync, max_rank):
 
 
 def run_chunks256(local_rank, ndev, steps, warmup, streams,
-                  barrier_sync, max_rank):
+                  barrier_sync, max_rank, red=0, name="chunks256"):
     """BASELINE configs[3] shape: independent 256^3 chunks, FULL path
     (H2D + kernels + host extract), overlapped on per-thread HIP
     contexts. PCIe-inclusive by design — reported as its own configs[]
@@ -175,7 +175,7 @@ def run_chunks256(local_rank, ndev, steps, warmup, streams,
         e = engine_mod.Engine.get(local_rank % ndev)
         # results discarded before the thread's next call: zero-copy
         e.mesh_chunk(chunk, resolution=RESOLUTION,
-                     reduction_factor=0, max_error=40.0, copy=False)
+                     reduction_factor=red, max_error=40.0, copy=False)
         return e
 
     def step():
@@ -195,14 +195,14 @@ def run_chunks256(local_rank, ndev, steps, warmup, streams,
     agg = {"n_labels": 0, "total_tris": 0}
     for chunk in chunks:
         e = engine_mod.Engine.get(local_rank % ndev)
-        e.mesh_chunk(chunk, resolution=RESOLUTION, reduction_factor=0,
+        e.mesh_chunk(chunk, resolution=RESOLUTION, reduction_factor=red,
                      max_error=40.0, copy=False)
         st = e.stats()
         agg["n_labels"] += int(st["n_labels"])
         agg["total_tris"] += int(st["total_tris"])
     pool.shutdown()
     return {
-        "name": "chunks256",
+        "name": name,
         "value": None,
         "elapsed": elapsed,
         "step_vox": nchunks * 256 ** 3,
@@ -210,12 +210,15 @@ def run_chunks256(local_rank, ndev, steps, warmup, streams,
             "workload": (f"{nchunks} x 256^3 u64 Voronoi chunks "
                          f"({seeds_per_chunk} labels each, BASELINE "
                          f"configs[3] shape), FULL path: H2D + kernels + "
-                         f"host extract, {streams} streams/GPU"),
+                         f"host extract, {streams} streams/GPU"
+                         + (f", simplification_factor={red} (the "
+                            f"reference's production default)"
+                            if red else "")),
             "chunk": [256, 256, 256],
             "n_chunks": nchunks,
             "labels": seeds_per_chunk * nchunks,
             "resolution_nm": list(RESOLUTION),
-            "simplification": 0,
+            "simplification": red,
             "input_residency": "host (PCIe-inclusive production path)",
             "n_labels_meshed": agg["n_labels"],
             "total_tris": agg["total_tris"],
@@ -296,7 +299,7 @@ def main():
     elif args.mode == "chunks256":
         which = ["chunks256"]
     else:
-        which = ["mc", "simplify", "chunks256"]
+        which = ["mc", "simplify", "chunks256", "chunks256_prod"]
 
     results = []
     for name in which:
@@ -306,11 +309,19 @@ def main():
         elif name == "simplify":
             results.append(run_chunk512(eng, data, args.steps, args.warmup,
                                         True, barrier_sync, max_rank))
-        else:
-            # runs LAST: flips the engine module to per-thread contexts
+        elif name == "chunks256":
+            # chunks256 benches run LAST: they flip the engine module to
+            # per-thread contexts
             results.append(run_chunks256(local_rank, ndev, args.steps,
                                          args.warmup, args.streams,
                                          barrier_sync, max_rank))
+        else:
+            # the reference's production DEFAULT: fan-out WITH
+            # simplification_factor=100 (task_creation/mesh.py:216)
+            results.append(run_chunks256(local_rank, ndev, args.steps,
+                                         args.warmup, args.streams,
+                                         barrier_sync, max_rank,
+                                         red=100, name="chunks256_prod"))
 
     for r in results:
         total_vox = r["step_vox"] * args.steps * world

@@ -487,3 +487,32 @@ def test_device_dust_parity(eng):
     want = oracle.mesh_chunk(host, resolution=res)
     _assert_meshsets_equal(got, want, "device dust")
     assert not (set(got.keys()) & dust)
+
+
+def test_full_parity_448_default_task_shape(eng):
+    """The reference's DEFAULT task shape is 448^3
+    (task_creation/mesh.py:161 shape=(448,448,448)): full bit-exact
+    parity at that production shape, dims not a multiple of the wave
+    width times segment count."""
+    import oracle
+    from igneous_amd.synth import voronoi_labels
+    data = voronoi_labels((448, 448, 448), 30000, 505, dtype=np.uint64)
+    res = (16.0, 16.0, 40.0)
+    got = eng.mesh_chunk(data, resolution=res)
+    want = oracle.mesh_chunk(data, resolution=res)
+    _assert_meshsets_equal(got, want, "448^3/30k")
+
+
+def test_simplify_parity_384(eng):
+    """Config-5 semantics at scale: 384^3/20k labels through the FULL
+    simplification path (reduction_factor=100, max_error=40), every
+    label bit-exact vs the oracle (oracle side ~60 s single-core)."""
+    import oracle
+    from igneous_amd.synth import voronoi_labels
+    data = voronoi_labels((384, 384, 384), 20000, 404, dtype=np.uint64)
+    res = (16.0, 16.0, 40.0)
+    got = eng.mesh_chunk(data, resolution=res, reduction_factor=100,
+                         max_error=40.0)
+    want = oracle.mesh_chunk(data, resolution=res, reduction_factor=100,
+                             max_error=40.0)
+    _assert_meshsets_equal(got, want, "384^3/20k simplified")
