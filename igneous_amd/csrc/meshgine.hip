@@ -655,7 +655,7 @@ struct mg_ctx {
       simp_Q, simp_pick, simp_remap, simp_flab, simp_flab_alt,
       simp_faces_alt, simp_verts_alt, simp_vbase_alt, simp_meta, simp_ref,
       simp_keep, simp_keep_scan, simp_park, simp_first, simp_troff_final,
-      simp_deg, simp_adj, simp_rh, simp_sched;
+      simp_deg, simp_adj, simp_rh, simp_sched, simp_accept;
   uint64_t lh_slots = 1ull << 20;
   HostBuf h_verts, h_faces;  // pinned output staging, reused across calls
   hipEvent_t ev[16] = {};
@@ -767,7 +767,7 @@ void mg_destroy(mg_ctx *c) {
                   &c->simp_meta, &c->simp_ref, &c->simp_keep,
                   &c->simp_keep_scan, &c->simp_park, &c->simp_first,
                   &c->simp_troff_final, &c->simp_deg, &c->simp_adj,
-                  &c->simp_rh, &c->simp_sched}) {
+                  &c->simp_rh, &c->simp_sched, &c->simp_accept}) {
     if (b->ptr) (void)hipFree(b->ptr);
   }
   if (c->h_verts.ptr) (void)hipHostFree(c->h_verts.ptr);
@@ -888,6 +888,14 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   if (ensure(c, c->simp_troff_final, (L + 1) * 4)) return 40;
   if (ensure(c, c->simp_deg, V * 4)) return 40;
   if (ensure(c, c->simp_adj, V * 4)) return 40;
+  if (ensure(c, c->simp_accept, V * 8)) return 40;
+  // proposal-acceptance second matching wave (contract knob shared
+  // with the oracle; MG_SIMP_PROPOSE=0 disables on both sides)
+  uint32_t propose = 1;
+  {
+    const char *e = getenv("MG_SIMP_PROPOSE");
+    if (e && e[0] == '0') propose = 0;
+  }
 
   uint32_t *faces_g = (uint32_t *)c->faces.ptr;
   float *verts = (float *)c->verts.ptr;
@@ -1004,7 +1012,9 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                              : nullptr,
                          d_rh,
                          max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs,
-                         nv_lo, nv_hi, nt_lo, nt_hi, d_sched);
+                         nv_lo, nv_hi, nt_lo, nt_hi,
+                         (unsigned long long *)c->simp_accept.ptr,
+                         propose, d_sched);
     };
     if (small_cap)
       launch_band(k_simplify_label<false, 64, 512>, s, 64, 0u, small_cap);
@@ -1184,6 +1194,8 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     for (uint32_t sub = 0; sub < subs_global; ++sub) {
     uint64_t nbs = (T + blk - 1) / blk;  // T shrinks between subs
     HIP_TRY(c, hipMemsetAsync(c->simp_pick.ptr, 0xFF, V * 8, s), 43);
+    if (propose)
+      HIP_TRY(c, hipMemsetAsync(c->simp_accept.ptr, 0xFF, V * 8, s), 43);
     hipLaunchKernelGGL(k_edge_pick, dim3((uint32_t)nbs), dim3(blk), 0, s,
                        faces_g, active, flab,
                        (const uint32_t *)c->vbase.ptr, verts,
@@ -1210,9 +1222,19 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       hipLaunchKernelGGL(k_iota, dim3((uint32_t)nbv), dim3(blk), 0, s,
                          (uint32_t *)c->simp_remap.ptr, V);
       hipLaunchKernelGGL(k_collapse, dim3((uint32_t)nbv), dim3(blk), 0, s,
-                         (const unsigned long long *)c->simp_pick.ptr, verts,
+                         (unsigned long long *)c->simp_pick.ptr, verts,
                          (uint32_t *)c->simp_remap.ptr,
                          (float *)c->simp_Q.ptr, V);
+      if (propose) {
+        hipLaunchKernelGGL(k_propose, dim3((uint32_t)nbv), dim3(blk), 0, s,
+                           (const unsigned long long *)c->simp_pick.ptr,
+                           (unsigned long long *)c->simp_accept.ptr, V);
+        hipLaunchKernelGGL(k_accept, dim3((uint32_t)nbv), dim3(blk), 0, s,
+                           (const unsigned long long *)c->simp_pick.ptr,
+                           (const unsigned long long *)c->simp_accept.ptr,
+                           verts, (uint32_t *)c->simp_remap.ptr,
+                           (float *)c->simp_Q.ptr, V);
+      }
     }
     hipLaunchKernelGGL(k_remap_faces, dim3((uint32_t)nbs), dim3(blk), 0, s,
                        faces_g, (const uint32_t *)c->simp_remap.ptr,

@@ -235,8 +235,10 @@ __global__ void k_edge_pick(const uint32_t *__restrict__ faces_g,
   }
 }
 
-// [S5] matched pairs collapse to the midpoint; u (smaller id) survives
-__global__ void k_collapse(const unsigned long long *__restrict__ pick,
+// [S5] matched pairs collapse; u (smaller id) survives. Matched
+// vertices leave the pick graph (blocked marking for the proposal
+// wave; the plain-store races are decision-invariant).
+__global__ void k_collapse(unsigned long long *__restrict__ pick,
                            float *__restrict__ verts,
                            uint32_t *__restrict__ remap,
                            float *__restrict__ Q,
@@ -249,6 +251,58 @@ __global__ void k_collapse(const unsigned long long *__restrict__ pick,
   if (w <= u) return;
   unsigned long long pw = pick[w];
   if (pw == ~0ull || (uint32_t)pw != u) return;
+  {
+    float mx = 0.5f*(verts[3*u]+verts[3*w]);
+    float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+    float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+    float S[10];
+    #pragma unroll
+    for (int k = 0; k < 10; ++k)
+      S[k] = Q[12*u + k] + Q[12*(uint64_t)w + k];
+    float px, py, pz;
+    (void)sq_place(S, mx, my, mz, &px, &py, &pz);
+    verts[3*u] = px; verts[3*u+1] = py; verts[3*u+2] = pz;
+  }
+  #pragma unroll
+  for (int k = 0; k < 10; ++k) Q[12*u + k] += Q[12*(uint64_t)w + k];
+  remap[w] = (uint32_t)u;
+  pick[u] = ~0ull;
+  pick[w] = ~0ull;
+}
+
+// [S5b] proposal wave (oracle step 3b), global-rounds path. Vertex ids
+// in pick are global; the proposer id in the acceptance key preserves
+// the oracle's local tie-break order (ids within one label are
+// monotone in the global numbering).
+__global__ void k_propose(const unsigned long long *__restrict__ pick,
+                          unsigned long long *__restrict__ accept,
+                          uint64_t nverts) {
+  uint64_t v = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (v >= nverts) return;
+  unsigned long long pu = pick[v];
+  if (pu == ~0ull) return;
+  uint32_t w = (uint32_t)pu;
+  if (w <= v) return;                  // propose-up only
+  if (pick[w] == ~0ull) return;        // blocked target
+  unsigned long long key =
+      (pu & 0xFFFFFFFF00000000ull) | (unsigned long long)(v + 1);
+  atomicMin(&accept[w], key);
+}
+
+__global__ void k_accept(const unsigned long long *__restrict__ pick,
+                         const unsigned long long *__restrict__ accept,
+                         float *__restrict__ verts,
+                         uint32_t *__restrict__ remap,
+                         float *__restrict__ Q,
+                         uint64_t nverts) {
+  uint64_t w = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (w >= nverts) return;
+  unsigned long long aw = accept[w];
+  if (aw == ~0ull) return;
+  unsigned long long pw = pick[w];
+  if (pw == ~0ull) return;             // matched or pickless
+  if ((uint32_t)pw > w) return;        // proposers never accept
+  uint64_t u = (uint64_t)((uint32_t)(aw & 0xFFFFFFFFull)) - 1;
   {
     float mx = 0.5f*(verts[3*u]+verts[3*w]);
     float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
@@ -632,6 +686,8 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     float max_cost, uint32_t nlabels, uint32_t big_cap,
     uint32_t subs, uint32_t nv_lo, uint32_t nv_hi,
     uint32_t nt_lo, uint32_t nt_hi,
+    unsigned long long *__restrict__ accept_g,  // per-vertex (wave 2)
+    uint32_t propose,
     const uint32_t *__restrict__ sched /*block->label, largest first*/) {
   if (blockIdx.x >= nlabels) return;
   // biggest-label-first dispatch: per-label serial time scales with nt0,
@@ -692,6 +748,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
   unsigned long long *pick_l =
       lds_mode ? s_pick : (pick + v0);
   RMT *rm = lds_mode ? s_remap : (RMT *)(remap + v0);
+  unsigned long long *accept_l = accept_g + v0;
   // ping-pong face buffers: rewrite reads fa, compaction scatters into
   // fb, then the buffers swap — no copy-back pass. Parking at the end
   // reads whichever buffer is current.
@@ -880,6 +937,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       if (!WAVEMODE) {
         pick_l[v] = ~0ull;  // fused pick reset (same-thread slot)
         rm[v] = (RMT)v;     // fused remap identity (consumed in collapse)
+        if (propose) accept_l[v] = ~0ull;
       }
     }
     __syncthreads();
@@ -889,6 +947,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       for (uint32_t v = tid; v < nv; v += BS) {
         pick_l[v] = ~0ull;
         rm[v] = (RMT)v;
+        if (propose) accept_l[v] = ~0ull;
       }
       __syncthreads();
     }
@@ -904,6 +963,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       for (uint32_t v = tid; v < nv; v += BS) {
         pick_l[v] = ~0ull;
         rm[v] = (RMT)v;
+        if (propose) accept_l[v] = ~0ull;
       }
       __syncthreads();
     }
@@ -980,9 +1040,58 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       #pragma unroll
       for (int k = 0; k < 10; ++k) Q[12ull*u + k] += Q[12ull*w + k];
       rm[w - v0] = u - v0;
+      // blocked marking: matched vertices leave the pick graph (the
+      // races on these plain stores are benign: any reader's decision
+      // is identical for the old and the cleared value — DESIGN §2)
+      pick_l[v] = ~0ull;
+      pick_l[w - v0] = ~0ull;
       atomicAdd(&s_collapses, 1u);
     }
     __syncthreads();
+    // [7b] proposal-acceptance second wave (oracle step 3b): mutual
+    // picks converge on cost-minima stars, so unmatched vertices whose
+    // pick points UP propose to that peer; an unmatched non-proposing
+    // peer accepts its minimum (costbits, local proposer+1) proposal.
+    if (propose) {
+      for (uint32_t v = tid; v < nv; v += BS) {
+        unsigned long long pu = pick_l[v];
+        if (pu == ~0ull) continue;
+        uint32_t w = (uint32_t)pu;
+        if (w <= v0 + v) continue;            // propose-up only
+        if (pick_l[w - v0] == ~0ull) continue;  // blocked target
+        unsigned long long key =
+            (pu & 0xFFFFFFFF00000000ull) | (unsigned long long)(v + 1);
+        atomicMin(&accept_l[w - v0], key);
+      }
+      __syncthreads();
+      for (uint32_t v = tid; v < nv; v += BS) {
+        unsigned long long aw = accept_l[v];
+        if (aw == ~0ull) continue;
+        unsigned long long pw = pick_l[v];
+        if (pw == ~0ull) continue;            // matched or pickless
+        uint32_t w = v0 + v;
+        if ((uint32_t)pw > w) continue;       // proposers never accept
+        uint32_t ul = (uint32_t)(aw & 0xFFFFFFFFull) - 1;
+        uint32_t u = v0 + ul;
+        {
+          float mx = 0.5f*(verts[3ull*u]+verts[3ull*w]);
+          float my = 0.5f*(verts[3ull*u+1]+verts[3ull*w+1]);
+          float mz = 0.5f*(verts[3ull*u+2]+verts[3ull*w+2]);
+          float S[10];
+          #pragma unroll
+          for (int k = 0; k < 10; ++k)
+            S[k] = Q[12ull*u + k] + Q[12ull*w + k];
+          float px, py, pz;
+          (void)sq_place(S, mx, my, mz, &px, &py, &pz);
+          verts[3ull*u] = px; verts[3ull*u+1] = py; verts[3ull*u+2] = pz;
+        }
+        #pragma unroll
+        for (int k = 0; k < 10; ++k) Q[12ull*u + k] += Q[12ull*w + k];
+        rm[v] = (RMT)ul;
+        atomicAdd(&s_collapses, 1u);
+      }
+      __syncthreads();
+    }
     if (s_collapses == 0) break;
     // [8] fused rewrite + stable compact (oracle step 4)
     uint32_t kept = blk_rewrite_compact<BS, WAVEMODE ? 8192 : 65536, RMT>(

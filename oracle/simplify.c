@@ -169,7 +169,7 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
         if (enc_w < pick[w]) pick[w] = enc_w;
       }
     }
-    /* 3. matched pairs collapse to midpoint (u<v keeps u) */
+    /* 3. matched pairs collapse (u<v keeps u) */
     for (uint32_t v = 0; v < nv; v++) remap[v] = v;
     uint32_t collapses = 0;
     for (uint32_t u = 0; u < nv; u++) {
@@ -190,7 +190,63 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
       }
       for (int k = 0; k < 10; k++) Q[u].q[k] += Q[w].q[k];
       remap[w] = u;
+      /* blocked marking: matched vertices leave the pick graph. A live
+       * pick entry therefore means "unmatched with an eligible edge";
+       * note a proposal target ALWAYS has a live pick if unmatched,
+       * because the edge (u,w) was offered to both endpoints. */
+      pick[u] = UINT64_MAX; pick[w] = UINT64_MAX;
       collapses++;
+    }
+    /* 3b. PROPOSAL-ACCEPTANCE second wave (contract, r02): mutual picks
+     * converge on local cost minima ("stars": many vertices pick one),
+     * so the mutual wave alone matches ~10-20%/round. Unmatched
+     * vertices whose pick points UP (peer id > self) PROPOSE to that
+     * peer; an unmatched NON-proposing peer accepts its minimum
+     * (costbits, proposer+1) proposal. Deterministic: acceptance is a
+     * min-reduction; proposers never accept; each proposer targets one
+     * vertex, so the extra pairs are disjoint among themselves and
+     * with the mutual wave. Placement/cost identical to wave 1.
+     * MG_SIMP_PROPOSE=0 disables (engine reads the same knob). */
+    {
+      static int propose_on = -1;
+      if (propose_on < 0) {
+        const char *e = getenv("MG_SIMP_PROPOSE");
+        propose_on = !(e && e[0] == '0');
+      }
+      if (propose_on) {
+        uint64_t *accept = (uint64_t*)malloc(nv * sizeof(uint64_t));
+        for (uint32_t v = 0; v < nv; v++) accept[v] = UINT64_MAX;
+        for (uint32_t u = 0; u < nv; u++) {
+          if (pick[u] == UINT64_MAX) continue;   /* matched or pickless */
+          uint32_t w = (uint32_t)pick[u];
+          if (w <= u) continue;          /* propose-up only */
+          if (pick[w] == UINT64_MAX) continue;   /* blocked target */
+          uint64_t key = (pick[u] & 0xFFFFFFFF00000000ull)
+                       | (uint64_t)(u + 1);
+          if (key < accept[w]) accept[w] = key;
+        }
+        for (uint32_t w = 0; w < nv; w++) {
+          if (accept[w] == UINT64_MAX) continue;
+          /* acceptors: unmatched (live pick) non-proposers */
+          if (pick[w] == UINT64_MAX) continue;
+          if ((uint32_t)pick[w] > w) continue;
+          uint32_t u = (uint32_t)(accept[w] & 0xFFFFFFFFull) - 1;
+          {
+            float mx = 0.5f*(verts[3*u]+verts[3*w]);
+            float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
+            float mz = 0.5f*(verts[3*u+2]+verts[3*w+2]);
+            quad10 S;
+            for (int k = 0; k < 10; k++) S.q[k] = Q[u].q[k] + Q[w].q[k];
+            float px, py, pz;
+            (void)quad_place(&S, mx, my, mz, &px, &py, &pz);
+            verts[3*u] = px; verts[3*u+1] = py; verts[3*u+2] = pz;
+          }
+          for (int k = 0; k < 10; k++) Q[u].q[k] += Q[w].q[k];
+          remap[w] = u;
+          collapses++;
+        }
+        free(accept);
+      }
     }
     if (!collapses) break;
     /* 4. rewrite faces, drop degenerates */
