@@ -888,7 +888,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
   if (ensure(c, c->simp_troff_final, (L + 1) * 4)) return 40;
   if (ensure(c, c->simp_deg, V * 4)) return 40;
   if (ensure(c, c->simp_adj, V * 4)) return 40;
-  if (ensure(c, c->simp_accept, V * 8)) return 40;
+  if (ensure(c, c->simp_accept, V * 4)) return 40;
   // proposal-acceptance second matching wave (contract knob shared
   // with the oracle; MG_SIMP_PROPOSE=0 disables on both sides)
   uint32_t propose = 1;
@@ -1013,7 +1013,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                          d_rh,
                          max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs,
                          nv_lo, nv_hi, nt_lo, nt_hi,
-                         (unsigned long long *)c->simp_accept.ptr,
+                         (uint32_t *)c->simp_accept.ptr,
                          propose, d_sched);
     };
     if (small_cap)
@@ -1195,7 +1195,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     uint64_t nbs = (T + blk - 1) / blk;  // T shrinks between subs
     HIP_TRY(c, hipMemsetAsync(c->simp_pick.ptr, 0xFF, V * 8, s), 43);
     if (propose)
-      HIP_TRY(c, hipMemsetAsync(c->simp_accept.ptr, 0xFF, V * 8, s), 43);
+      HIP_TRY(c, hipMemsetAsync(c->simp_accept.ptr, 0xFF, V * 4, s), 43);
     hipLaunchKernelGGL(k_edge_pick, dim3((uint32_t)nbs), dim3(blk), 0, s,
                        faces_g, active, flab,
                        (const uint32_t *)c->vbase.ptr, verts,
@@ -1228,10 +1228,10 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
       if (propose) {
         hipLaunchKernelGGL(k_propose, dim3((uint32_t)nbv), dim3(blk), 0, s,
                            (const unsigned long long *)c->simp_pick.ptr,
-                           (unsigned long long *)c->simp_accept.ptr, V);
+                           (uint32_t *)c->simp_accept.ptr, V);
         hipLaunchKernelGGL(k_accept, dim3((uint32_t)nbv), dim3(blk), 0, s,
                            (const unsigned long long *)c->simp_pick.ptr,
-                           (const unsigned long long *)c->simp_accept.ptr,
+                           (const uint32_t *)c->simp_accept.ptr,
                            verts, (uint32_t *)c->simp_remap.ptr,
                            (float *)c->simp_Q.ptr, V);
       }

@@ -275,7 +275,7 @@ __global__ void k_collapse(unsigned long long *__restrict__ pick,
 // the oracle's local tie-break order (ids within one label are
 // monotone in the global numbering).
 __global__ void k_propose(const unsigned long long *__restrict__ pick,
-                          unsigned long long *__restrict__ accept,
+                          uint32_t *__restrict__ accept,
                           uint64_t nverts) {
   uint64_t v = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (v >= nverts) return;
@@ -284,25 +284,23 @@ __global__ void k_propose(const unsigned long long *__restrict__ pick,
   uint32_t w = (uint32_t)pu;
   if (w <= v) return;                  // propose-up only
   if (pick[w] == ~0ull) return;        // blocked target
-  unsigned long long key =
-      (pu & 0xFFFFFFFF00000000ull) | (unsigned long long)(v + 1);
-  atomicMin(&accept[w], key);
+  atomicMin(&accept[w], (uint32_t)(v + 1));
 }
 
 __global__ void k_accept(const unsigned long long *__restrict__ pick,
-                         const unsigned long long *__restrict__ accept,
+                         const uint32_t *__restrict__ accept,
                          float *__restrict__ verts,
                          uint32_t *__restrict__ remap,
                          float *__restrict__ Q,
                          uint64_t nverts) {
   uint64_t w = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (w >= nverts) return;
-  unsigned long long aw = accept[w];
-  if (aw == ~0ull) return;
+  uint32_t aw = accept[w];
+  if (aw == 0xFFFFFFFFu) return;
   unsigned long long pw = pick[w];
   if (pw == ~0ull) return;             // matched or pickless
   if ((uint32_t)pw > w) return;        // proposers never accept
-  uint64_t u = (uint64_t)((uint32_t)(aw & 0xFFFFFFFFull)) - 1;
+  uint64_t u = (uint64_t)aw - 1;
   {
     float mx = 0.5f*(verts[3*u]+verts[3*w]);
     float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
@@ -686,7 +684,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     float max_cost, uint32_t nlabels, uint32_t big_cap,
     uint32_t subs, uint32_t nv_lo, uint32_t nv_hi,
     uint32_t nt_lo, uint32_t nt_hi,
-    unsigned long long *__restrict__ accept_g,  // per-vertex (wave 2)
+    uint32_t *__restrict__ accept_g,  // per-vertex (wave 2)
     uint32_t propose,
     const uint32_t *__restrict__ sched /*block->label, largest first*/) {
   if (blockIdx.x >= nlabels) return;
@@ -748,7 +746,11 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
   unsigned long long *pick_l =
       lds_mode ? s_pick : (pick + v0);
   RMT *rm = lds_mode ? s_remap : (RMT *)(remap + v0);
-  unsigned long long *accept_l = accept_g + v0;
+  // accept table (proposal wave): u32 id-only keys — lives in the
+  // degree array's bytes (dead during the collapse phases) when the
+  // label is LDS-resident; global fallback otherwise
+  uint32_t *accept_l = (!WAVEMODE && lds_mode)
+                           ? s_deg : (accept_g + v0);
   // ping-pong face buffers: rewrite reads fa, compaction scatters into
   // fb, then the buffers swap — no copy-back pass. Parking at the end
   // reads whichever buffer is current.
@@ -937,7 +939,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       if (!WAVEMODE) {
         pick_l[v] = ~0ull;  // fused pick reset (same-thread slot)
         rm[v] = (RMT)v;     // fused remap identity (consumed in collapse)
-        if (propose) accept_l[v] = ~0ull;
+        if (propose) accept_l[v] = 0xFFFFFFFFu;
       }
     }
     __syncthreads();
@@ -947,7 +949,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       for (uint32_t v = tid; v < nv; v += BS) {
         pick_l[v] = ~0ull;
         rm[v] = (RMT)v;
-        if (propose) accept_l[v] = ~0ull;
+        if (propose) accept_l[v] = 0xFFFFFFFFu;
       }
       __syncthreads();
     }
@@ -963,7 +965,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       for (uint32_t v = tid; v < nv; v += BS) {
         pick_l[v] = ~0ull;
         rm[v] = (RMT)v;
-        if (propose) accept_l[v] = ~0ull;
+        if (propose) accept_l[v] = 0xFFFFFFFFu;
       }
       __syncthreads();
     }
@@ -1059,19 +1061,17 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
         uint32_t w = (uint32_t)pu;
         if (w <= v0 + v) continue;            // propose-up only
         if (pick_l[w - v0] == ~0ull) continue;  // blocked target
-        unsigned long long key =
-            (pu & 0xFFFFFFFF00000000ull) | (unsigned long long)(v + 1);
-        atomicMin(&accept_l[w - v0], key);
+        atomicMin(&accept_l[w - v0], v + 1);
       }
       __syncthreads();
       for (uint32_t v = tid; v < nv; v += BS) {
-        unsigned long long aw = accept_l[v];
-        if (aw == ~0ull) continue;
+        uint32_t aw = accept_l[v];
+        if (aw == 0xFFFFFFFFu) continue;
         unsigned long long pw = pick_l[v];
         if (pw == ~0ull) continue;            // matched or pickless
         uint32_t w = v0 + v;
         if ((uint32_t)pw > w) continue;       // proposers never accept
-        uint32_t ul = (uint32_t)(aw & 0xFFFFFFFFull) - 1;
+        uint32_t ul = aw - 1;
         uint32_t u = v0 + ul;
         {
           float mx = 0.5f*(verts[3ull*u]+verts[3ull*w]);

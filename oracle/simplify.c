@@ -214,23 +214,25 @@ void omc_simplify_mesh(float *verts, uint32_t *nverts_io,
         propose_on = !(e && e[0] == '0');
       }
       if (propose_on) {
-        uint64_t *accept = (uint64_t*)malloc(nv * sizeof(uint64_t));
-        for (uint32_t v = 0; v < nv; v++) accept[v] = UINT64_MAX;
+        /* id-only key: the acceptor takes its MINIMUM-ID proposer
+         * (every proposal already satisfies the cost bound; a 32-bit
+         * key lets the engine keep the accept table in otherwise-dead
+         * LDS bytes) */
+        uint32_t *accept = (uint32_t*)malloc(nv * sizeof(uint32_t));
+        for (uint32_t v = 0; v < nv; v++) accept[v] = UINT32_MAX;
         for (uint32_t u = 0; u < nv; u++) {
           if (pick[u] == UINT64_MAX) continue;   /* matched or pickless */
           uint32_t w = (uint32_t)pick[u];
           if (w <= u) continue;          /* propose-up only */
           if (pick[w] == UINT64_MAX) continue;   /* blocked target */
-          uint64_t key = (pick[u] & 0xFFFFFFFF00000000ull)
-                       | (uint64_t)(u + 1);
-          if (key < accept[w]) accept[w] = key;
+          if (u + 1 < accept[w]) accept[w] = u + 1;
         }
         for (uint32_t w = 0; w < nv; w++) {
-          if (accept[w] == UINT64_MAX) continue;
+          if (accept[w] == UINT32_MAX) continue;
           /* acceptors: unmatched (live pick) non-proposers */
           if (pick[w] == UINT64_MAX) continue;
           if ((uint32_t)pick[w] > w) continue;
-          uint32_t u = (uint32_t)(accept[w] & 0xFFFFFFFFull) - 1;
+          uint32_t u = accept[w] - 1;
           {
             float mx = 0.5f*(verts[3*u]+verts[3*w]);
             float my = 0.5f*(verts[3*u+1]+verts[3*w+1]);
