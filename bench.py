@@ -6,15 +6,17 @@ GPU, weak scaling (each rank meshes its own chunk per step).
 
     python bench.py --gpus N --steps K --warmup W
 
-With no mode flags, THREE benches run back-to-back and all appear in the
+With no mode flags, FOUR benches run back-to-back and all appear in the
 single JSON line's "configs" array (the top-level value/ms_per_step stay
 the headline configs[2] device rate):
-  - mc          BASELINE configs[2]: 512^3 u64 device-resident MC-only
-  - simplify    BASELINE configs[4] ("config 5"): same chunk with
-                simplification_factor=100, max_error=40
-  - chunks256   BASELINE configs[3] shape: 8 x 256^3 chunks through the
-                FULL path (H2D + kernels + host extract) on --streams
-                HIP streams/contexts per GPU
+  - mc             BASELINE configs[2]: 512^3 u64 device-resident MC-only
+  - simplify       BASELINE configs[4] ("config 5"): same chunk with
+                   simplification_factor=100, max_error=40
+  - chunks256      BASELINE configs[3] shape: 8 x 256^3 chunks through
+                   the FULL path (H2D + kernels + host extract) on
+                   --streams HIP streams/contexts per GPU
+  - chunks256_prod the same fan-out with the reference's production
+                   default simplification_factor=100
 --mode/--simplify restrict the run to one bench (targeted profiling).
 
 For N>1 the driver launches this under torch.distributed.run with one rank
