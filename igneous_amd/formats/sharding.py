@@ -361,7 +361,8 @@ class ShardReader:
         i = int(hit[0])
         if self.spec.data_encoding == "gzip":
             raw = blob[idx_end + starts[i]: idx_end + starts[i] + sizes[i]]
-            return (gzip.decompress(raw), 0, None)
+            out = gzip.decompress(raw)
+            return (out, 0, len(out))
         return (blob, idx_end + int(starts[i]), int(sizes[i]))
 
     def list_labels_in_shard(self, shard_name: str):
