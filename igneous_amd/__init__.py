@@ -6,7 +6,7 @@ equivalent), LocalTaskQueue/RegisteredTask (in-process queue), the mesh
 tasks, and create_meshing_tasks. Compute runs as hand-written HIP/CDNA4
 kernels behind the C ABI in include/meshgine.h.
 """
-from .mesher import Mesher
+from .mesher import Mesher, simplify_fqmr, chunk_mesh
 from .meshes import Mesh
 from .queue import LocalTaskQueue, RegisteredTask
 from .tasks import (

@@ -108,3 +108,33 @@ class Mesher:
             self._data, resolution=self.voxel_res,
             reduction_factor=reduction_factor, max_error=max_error,
             voxel_centered=voxel_centered)
+
+
+# ---------------------------------------------------------------------------
+# zmesh module-level function compat (the reference imports these from
+# zmesh in the multires merge: multires.py:321 zmesh.Mesh, :342
+# zmesh.simplify_fqmr, :550/:574 zmesh.chunk_mesh).
+
+def simplify_fqmr(mesh, target_count: int, aggressiveness: float = 7.0,
+                  preserve_border: bool = False,
+                  return_iterations: bool = False,
+                  max_iterations: int = 0, K: int = 3, alpha: float = None,
+                  update_rate: int = None):
+    """zmesh.simplify_fqmr-shaped wrapper over the GPU quadric
+    simplifier (mg_simplify_mesh). The pyfqmr aggressiveness/K error
+    schedule is NOT replicated: this repo's
+    deterministic contract (reduction_factor = ntris//target, unbounded
+    max_error) drives the reduction instead; the schedule arguments are
+    accepted for signature compatibility."""
+    from . import engine
+    out = engine.simplify_mesh(mesh, target_count)
+    if return_iterations:
+        return out, 1
+    return out
+
+
+def chunk_mesh(mesh, scale, offset):
+    """zmesh.chunk_mesh equivalent: grid-partition with triangle
+    clipping at cell boundaries (igneous_amd.meshops)."""
+    from . import meshops
+    return meshops.chunk_mesh(mesh, scale, offset)
