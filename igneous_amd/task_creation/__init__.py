@@ -4,5 +4,6 @@ from .mesh import (
     create_sharded_multires_mesh_tasks,
     create_unsharded_multires_mesh_tasks,
     configure_multires_info,
+    create_spatial_index_mesh_tasks,
 )
 from .common import FinelyDividedTaskIterator, num_tasks

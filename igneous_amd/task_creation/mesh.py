@@ -19,7 +19,8 @@ from .common import FinelyDividedTaskIterator, operator_contact
 __all__ = ["create_meshing_tasks", "create_mesh_manifest_tasks",
            "create_sharded_multires_mesh_tasks",
            "create_unsharded_multires_mesh_tasks",
-           "configure_multires_info"]
+           "configure_multires_info",
+           "create_spatial_index_mesh_tasks"]
 
 
 def create_meshing_tasks(
@@ -304,3 +305,60 @@ def create_sharded_multires_mesh_tasks(
                 min_chunk_size=min_chunk_size)
         for shard_no in shard_labels.keys()
     ]
+
+
+def create_spatial_index_mesh_tasks(
+        cloudpath: str,
+        shape=(448, 448, 448),
+        mip: int = 0,
+        fill_missing: bool = False,
+        compress='gzip',
+        mesh_dir: Optional[str] = None):
+    """Mirror of task_creation/mesh.py:363-435: (re)build the mesh
+    spatial index over a grid of SpatialIndexTasks."""
+    import copy as _c
+    from functools import partial
+    from ..tasks.spatial_index import SpatialIndexTask
+
+    shape = Vec(*shape)
+    vol = PrecomputedVolume(cloudpath, mip=mip)
+    if mesh_dir is None and not vol.info.get("mesh", None):
+        mesh_dir = f"mesh_mip_{mip}_err_40"
+    elif mesh_dir is None:
+        mesh_dir = vol.info["mesh"]
+    if "mesh" not in vol.info:
+        vol.info["mesh"] = mesh_dir
+        vol.commit_info()
+
+    cf = CloudFiles(cloudpath)
+    info_filename = f"{mesh_dir}/info"
+    mesh_info = cf.get_json(info_filename) or {}
+    new_mesh_info = _c.deepcopy(mesh_info)
+    new_mesh_info['@type'] = new_mesh_info.get(
+        '@type', 'neuroglancer_legacy_mesh')
+    new_mesh_info['mip'] = new_mesh_info.get("mip", int(mip))
+    new_mesh_info['chunk_size'] = shape.tolist()
+    new_mesh_info['spatial_index'] = {
+        'resolution': vol.resolution.tolist(),
+        'chunk_size': (shape * vol.resolution).tolist(),
+    }
+    if new_mesh_info != mesh_info:
+        cf.put_json(info_filename, new_mesh_info)
+
+    precision = (new_mesh_info['spatial_index'].get('precision')
+                 if isinstance(new_mesh_info.get('spatial_index'), dict)
+                 else None)
+
+    class SpatialIndexMeshTaskIterator(FinelyDividedTaskIterator):
+        def task(self, tshape, offset):
+            return partial(SpatialIndexTask,
+                           cloudpath=cloudpath,
+                           shape=tuple(int(x) for x in tshape),
+                           offset=tuple(int(x) for x in offset),
+                           subdir=mesh_dir,
+                           precision=precision,
+                           mip=int(mip),
+                           fill_missing=bool(fill_missing),
+                           compress=compress)
+
+    return SpatialIndexMeshTaskIterator(vol.mip_bounds(mip), shape)

@@ -282,3 +282,22 @@ def test_sharded_from_unsharded_merge(tmp_layer_path, oracle_mesher,
     for label in (5, 77):
         man = reader.get(label)
         assert man is not None and len(man) > 0
+
+
+def test_create_spatial_index_mesh_tasks(tmp_layer_path):
+    from igneous_amd.task_creation import create_spatial_index_mesh_tasks
+    data = np.zeros((64, 64, 64), dtype=np.uint32)
+    data[5:40, 5:40, 5:40] = 3
+    PrecomputedVolume.from_numpy(
+        data, tmp_layer_path, resolution=(4, 4, 40),
+        chunk_size=(64, 64, 64), mesh_dir="mesh")
+    tasks = list(create_spatial_index_mesh_tasks(
+        tmp_layer_path, shape=(64, 64, 64)))
+    assert len(tasks) == 1
+    for t in tasks:
+        t()
+    cf = CloudFiles(tmp_layer_path)
+    names = [n for n in cf.list("mesh/") if n.endswith(".spatial")]
+    assert len(names) == 1
+    assert cf.get_json("mesh/info")["spatial_index"]["resolution"] == [4, 4, 40]
+    assert set(cf.get_json(names[0]).keys()) == {"3"}
