@@ -20,7 +20,9 @@ __all__ = ["create_meshing_tasks", "create_mesh_manifest_tasks",
            "create_sharded_multires_mesh_tasks",
            "create_unsharded_multires_mesh_tasks",
            "configure_multires_info",
-           "create_spatial_index_mesh_tasks"]
+           "create_spatial_index_mesh_tasks",
+           "create_mesh_deletion_tasks", "create_xfer_meshes_tasks",
+           "create_sharded_multires_mesh_from_unsharded_tasks"]
 
 
 def create_meshing_tasks(
@@ -362,3 +364,134 @@ def create_spatial_index_mesh_tasks(
                            compress=compress)
 
     return SpatialIndexMeshTaskIterator(vol.mip_bounds(mip), shape)
+
+
+def create_mesh_deletion_tasks(layer_path: str, magnitude: int = 3,
+                               mesh_dir: Optional[str] = None):
+    """Mirror of task_creation/mesh.py:91-156: delete the mesh info and
+    fan prefix-parallel DeleteMeshFilesTask over the fragment names.
+    (file:// stores list cheaply, so one full-prefix task suffices —
+    same final state as the reference's 10^magnitude prefix split.)"""
+    from functools import partial
+    from ..tasks.mesh import DeleteMeshFilesTask
+    assert int(magnitude) == magnitude and magnitude >= 0
+    vol = PrecomputedVolume(layer_path)
+    mdir = mesh_dir or vol.info.get('mesh', 'mesh')
+    cf = CloudFiles(f"{layer_path.rstrip('/')}/{mdir}")
+    cf.delete(['info'])
+    try:
+        return [partial(DeleteMeshFilesTask, cloudpath=layer_path,
+                        prefix="", mesh_dir=mesh_dir)]
+    finally:
+        vol.provenance.processing.append({
+            'method': {
+                'task': 'DeleteMeshFilesTask',
+                'layer_path': layer_path,
+                'mesh_dir': mesh_dir,
+            },
+            'by': operator_contact(),
+            'date': strftime('%Y-%m-%d %H:%M %Z'),
+        })
+        vol.commit_provenance()
+
+
+def create_xfer_meshes_tasks(src: str, dest: str,
+                             mesh_dir: Optional[str] = None,
+                             magnitude: int = 2):
+    """Mirror of task_creation/mesh.py:548-588: copy a mesh directory
+    between layers (one full-prefix task on file:// stores)."""
+    from functools import partial
+    from ..tasks.mesh import TransferMeshFilesTask
+    cf_dest = CloudFiles(dest)
+    if not mesh_dir:
+        info = cf_dest.get_json("info") or {}
+        if info.get("mesh", None):
+            mesh_dir = info.get("mesh")
+    src_vol = PrecomputedVolume(src)
+    smdir = mesh_dir or src_vol.info.get('mesh', 'mesh')
+    src_mesh_info = CloudFiles(src).get_json(f"{smdir}/info")
+    if src_mesh_info is not None:
+        cf_dest.put_json(f"{mesh_dir or smdir}/info", src_mesh_info)
+    return [partial(TransferMeshFilesTask, src=src, dest=dest,
+                    prefix="", mesh_dir=mesh_dir)]
+
+
+def create_sharded_multires_mesh_from_unsharded_tasks(
+        src: str, dest: str,
+        shard_index_bytes: int = 2 ** 13,
+        minishard_index_bytes: int = 2 ** 15,
+        min_shards: int = 1,
+        num_lod: int = 0,
+        draco_compression_level: int = 7,
+        vertex_quantization_bits: int = 16,
+        minishard_index_encoding: str = "gzip",
+        mesh_dir: Optional[str] = None):
+    """Mirror of task_creation/mesh.py:590-705: convert an UNSHARDED
+    legacy mesh layer into sharded multires — enumerate labels from the
+    source fragment files, compute shard params, write the multilod
+    info + {shard}.labels into dest, return one
+    MultiResShardedFromUnshardedMeshMergeTask per shard."""
+    from functools import partial
+    from ..tasks.multires import (MultiResShardedFromUnshardedMeshMergeTask,
+                                  get_mesh_filenames_subset)
+
+    src_vol = PrecomputedVolume(src)
+    smdir = mesh_dir or src_vol.info.get('mesh', 'mesh')
+    all_labels = sorted(get_mesh_filenames_subset(src, smdir, "").keys())
+
+    (shard_bits, minishard_bits, preshift_bits) = \
+        _sharding.compute_shard_params_for_hashed(
+            num_labels=len(all_labels),
+            shard_index_bytes=int(shard_index_bytes),
+            minishard_index_bytes=int(minishard_index_bytes),
+            min_shards=min_shards)
+    spec = _sharding.ShardingSpecification(
+        type='neuroglancer_uint64_sharded_v1',
+        preshift_bits=preshift_bits,
+        hash='murmurhash3_x86_128',
+        minishard_bits=minishard_bits,
+        shard_bits=shard_bits,
+        minishard_index_encoding=minishard_index_encoding,
+        data_encoding="raw")
+
+    mesh_info = configure_multires_info(
+        dest, vertex_quantization_bits, mesh_dir or smdir)
+    mesh_info['sharding'] = spec.to_dict()
+    dmdir = mesh_dir or PrecomputedVolume(dest).info.get('mesh', smdir)
+    cf_dest = CloudFiles(dest)
+    cf_dest.put_json(f"{dmdir}/info", mesh_info,
+                     cache_control="no-cache")
+
+    shard_labels = _sharding.assign_labels_to_shards(
+        _np.asarray(all_labels, dtype=_np.uint64),
+        preshift_bits, shard_bits, minishard_bits)
+    cf_mesh = CloudFiles(f"{dest.rstrip('/')}/{dmdir}")
+    for shardno, labels in shard_labels.items():
+        cf_mesh.put_json(str(shardno) + '.labels', labels,
+                         compress="gzip", cache_control="no-cache")
+
+    dvol = PrecomputedVolume(dest)
+    dvol.provenance.processing.append({
+        'method': {
+            'task': 'MultiResShardedFromUnshardedMeshMergeTask',
+            'src': src, 'dest': dest,
+            'num_lod': num_lod,
+            'vertex_quantization_bits': vertex_quantization_bits,
+            'preshift_bits': preshift_bits,
+            'minishard_bits': minishard_bits,
+            'shard_bits': shard_bits,
+            'mesh_dir': dmdir,
+            'draco_compression_level': draco_compression_level,
+        },
+        'by': operator_contact(),
+        'date': strftime('%Y-%m-%d %H:%M %Z'),
+    })
+    dvol.commit_provenance()
+
+    return [
+        partial(MultiResShardedFromUnshardedMeshMergeTask,
+                src=src, dest=dest, shard_no=shard_no,
+                num_lod=num_lod, mesh_dir=dmdir,
+                draco_compression_level=draco_compression_level)
+        for shard_no in shard_labels.keys()
+    ]

@@ -301,3 +301,67 @@ def test_create_spatial_index_mesh_tasks(tmp_layer_path):
     assert len(names) == 1
     assert cf.get_json("mesh/info")["spatial_index"]["resolution"] == [4, 4, 40]
     assert set(cf.get_json(names[0]).keys()) == {"3"}
+
+
+def test_mesh_deletion_and_xfer_tasks(tmp_layer_path, oracle_mesher,
+                                      tmp_path):
+    from igneous_amd.task_creation import (create_mesh_deletion_tasks,
+                                           create_xfer_meshes_tasks)
+    _make_two_chunk_layer(tmp_layer_path)
+    for t in create_meshing_tasks(
+            tmp_layer_path, mip=0, shape=(64, 64, 64), sharded=False,
+            spatial_index=False, simplification=False):
+        t.execute()
+    cf = CloudFiles(tmp_layer_path)
+    assert [n for n in cf.list("mesh/") if ":0:" in n]
+
+    # transfer to a second layer
+    dest = f"file://{tmp_path}/xfer"
+    data = np.zeros((4, 4, 4), dtype=np.uint64)
+    PrecomputedVolume.from_numpy(
+        data, dest, resolution=(4, 4, 40), chunk_size=(4, 4, 4),
+        mesh_dir="mesh")
+    for t in create_xfer_meshes_tasks(tmp_layer_path, dest,
+                                      mesh_dir="mesh"):
+        t()
+    cf_dest = CloudFiles(dest)
+    assert sorted(n for n in cf_dest.list("mesh/") if ":0:" in n) == \
+        sorted(n for n in cf.list("mesh/") if ":0:" in n)
+
+    # delete the source meshes
+    for t in create_mesh_deletion_tasks(tmp_layer_path):
+        t()
+    assert not [n for n in cf.list("mesh/") if ":0:" in n]
+
+
+def test_create_sharded_from_unsharded_tasks(tmp_layer_path, oracle_mesher,
+                                             oracle_simplifier, tmp_path):
+    from igneous_amd.formats import sharding as _sh
+    from igneous_amd.formats.multilod import (
+        MultiLevelPrecomputedMeshManifest)
+    from igneous_amd.task_creation import (
+        create_sharded_multires_mesh_from_unsharded_tasks)
+    _make_two_chunk_layer(tmp_layer_path)
+    for t in create_meshing_tasks(
+            tmp_layer_path, mip=0, shape=(64, 64, 64), sharded=False,
+            spatial_index=False, simplification=False):
+        t.execute()
+    dest = f"file://{tmp_path}/dest2"
+    data = np.zeros((4, 4, 4), dtype=np.uint64)
+    PrecomputedVolume.from_numpy(
+        data, dest, resolution=(4, 4, 40), chunk_size=(4, 4, 4),
+        mesh_dir="mesh")
+    tasks = create_sharded_multires_mesh_from_unsharded_tasks(
+        tmp_layer_path, dest, num_lod=0)
+    for t in tasks:
+        t()
+    cf_dest = CloudFiles(dest)
+    info = cf_dest.get_json("mesh/info")
+    spec = _sh.ShardingSpecification.from_dict(info["sharding"])
+    reader = _sh.ShardReader(spec, lambda n: cf_dest.get(f"mesh/{n}"))
+    for label in (5, 77):
+        man_bytes = reader.get(label)
+        assert man_bytes, f"label {label} missing in dest shards"
+        man = MultiLevelPrecomputedMeshManifest.from_binary(
+            man_bytes, segment_id=label)
+        assert sum(man.num_fragments_per_lod) >= 1
