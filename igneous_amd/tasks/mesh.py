@@ -176,7 +176,12 @@ class MeshTask(RegisteredTask):
         mesher_probe = _get_mesher()
         device_dust = 0
         if (opts['dust_threshold'] and not opts['dust_global']
+                and opts['remap_table'] is None
                 and getattr(mesher_probe, 'handles_dust', False)):
+            # remap_table must force the HOST order: the reference dusts
+            # BEFORE remapping (mesh.py:193-198), so labels merged by
+            # the remap pool their voxel counts — device dust runs after
+            # the host remap and would see the merged labels
             device_dust = int(opts['dust_threshold'])
         else:
             data = self._remove_dust(data, opts['dust_threshold'],

@@ -275,3 +275,18 @@ def test_chunk_mesh_interior_only():
     chunks = meshops.chunk_mesh(box, scale=(64, 64, 64), offset=(0, 0, 0))
     assert list(chunks.keys()) == [(0, 0, 0)]
     assert abs(_area(chunks[(0, 0, 0)]) - _area(box)) < 1e-9
+
+
+def test_draco_varint_index_width():
+    """Index-width tiers: u8 (<256 points), u16 (<65536) are fuzz
+    covered; pin the varint tier (>=2^16 points) explicitly."""
+    from igneous_amd.formats import draco as draco_fmt
+    nv = 70000
+    rng = np.random.default_rng(9)
+    v = rng.integers(0, 2 ** 20, size=(nv, 3)).astype(np.uint32)
+    f = np.stack([np.arange(nv - 2), np.arange(1, nv - 1),
+                  np.arange(2, nv)], axis=1).astype(np.uint32)[:500]
+    blob = draco_fmt.encode(v, f)
+    v2, f2 = draco_fmt.decode(blob)
+    assert np.array_equal(v, v2)
+    assert np.array_equal(f, f2)
