@@ -52,3 +52,24 @@ class Mesh:
             fs.append(m.faces + offs)
             offs += m.vertices.shape[0]
         return cls(np.concatenate(vs), np.concatenate(fs), id=id)
+
+
+# cloud-volume Mesh API conveniences used by the reference merge code
+# (multires.py:239 mesh.consolidate(), :551 merge_close_vertices)
+def _consolidate(self) -> "Mesh":
+    from . import meshops
+    return meshops.consolidate(self)
+
+
+def _merge_close_vertices(self, radius: float = 1e-5) -> "Mesh":
+    from . import meshops
+    return meshops.merge_close_vertices(self, radius=radius)
+
+
+def _empty(self) -> bool:
+    return self.vertices.shape[0] == 0 or self.faces.shape[0] == 0
+
+
+Mesh.consolidate = _consolidate
+Mesh.merge_close_vertices = _merge_close_vertices
+Mesh.empty = _empty
