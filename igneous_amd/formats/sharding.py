@@ -36,7 +36,7 @@ spec requires.
 from __future__ import annotations
 
 import gzip
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, Optional
 
 import numpy as np

@@ -40,7 +40,7 @@ from ..formats.mapbuffer import MapBuffer
 from ..formats.multilod import (MultiLevelPrecomputedMeshManifest,
                                 to_stored_model_space)
 from ..formats import sharding
-from ..lib import Bbox, Vec
+from ..lib import Bbox
 from ..meshes import Mesh
 from .. import meshops
 from ..spatial_index import SpatialIndex

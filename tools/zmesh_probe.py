@@ -6,7 +6,6 @@ lease. Writes gpurun_out/zmesh_probe.json when run under gpurun."""
 import glob
 import json
 import os
-import subprocess
 import sys
 import time
 
