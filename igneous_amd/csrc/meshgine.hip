@@ -896,13 +896,6 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
     const char *e = getenv("MG_SIMP_PROPOSE");
     if (e && e[0] == '0') propose = 0;
   }
-  // selective pick recomputation (pure memoization, results identical;
-  // MG_SIMP_MEMO=0 disables for A/B)
-  uint32_t memo = 1;
-  {
-    const char *e = getenv("MG_SIMP_MEMO");
-    if (e && e[0] == '0') memo = 0;
-  }
 
   uint32_t *faces_g = (uint32_t *)c->faces.ptr;
   float *verts = (float *)c->verts.ptr;
@@ -1021,7 +1014,7 @@ static int run_simplify(mg_ctx *c, uint32_t nlabels,
                          max_cost, (uint32_t)L, SIMP_BIG_CAP, simp_subs,
                          nv_lo, nv_hi, nt_lo, nt_hi,
                          (uint32_t *)c->simp_accept.ptr,
-                         propose, memo, d_sched);
+                         propose, d_sched);
     };
     if (small_cap)
       launch_band(k_simplify_label<false, 64, 512>, s, 64, 0u, small_cap);

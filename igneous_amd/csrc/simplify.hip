@@ -612,9 +612,7 @@ __device__ uint32_t blk_rewrite_compact(const uint32_t *__restrict__ fa,
                                         uint32_t *__restrict__ fb,
                                         const RMT *__restrict__ rm,
                                         uint32_t v0,
-                                        uint32_t n, uint32_t *s_sums,
-                                        const unsigned long long *s_win,
-                                        unsigned long long *s_dirty) {
+                                        uint32_t n, uint32_t *s_sums) {
   const uint32_t tid = threadIdx.x;
   const uint32_t chunk = (n + BS - 1) / BS;
   const uint32_t lo = tid * chunk;
@@ -627,21 +625,6 @@ __device__ uint32_t blk_rewrite_compact(const uint32_t *__restrict__ fa,
     uint32_t i0 = (uint32_t)rm[fa[3*i] - v0];
     uint32_t i1 = (uint32_t)rm[fa[3*i+1] - v0];
     uint32_t i2 = (uint32_t)rm[fa[3*i+2] - v0];
-    if (s_win) {
-      // pick memoization: a face's edge costs go stale iff it touches
-      // a collapse winner (a remapped corner's post id IS a winner, so
-      // this one test covers both adjacency cases) — mark its corners
-      // dirty for the next sub-round's selective re-pick
-      bool touches =
-          ((s_win[i0 >> 6] >> (i0 & 63)) & 1ull) |
-          ((s_win[i1 >> 6] >> (i1 & 63)) & 1ull) |
-          ((s_win[i2 >> 6] >> (i2 & 63)) & 1ull);
-      if (touches) {
-        atomicOr(&s_dirty[i0 >> 6], 1ull << (i0 & 63));
-        atomicOr(&s_dirty[i1 >> 6], 1ull << (i1 & 63));
-        atomicOr(&s_dirty[i2 >> 6], 1ull << (i2 & 63));
-      }
-    }
     if (i0 != i1 && i1 != i2 && i0 != i2) {
       bm[(i - lo) >> 6] |= 1ull << ((i - lo) & 63);
       ++sum;
@@ -702,7 +685,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     uint32_t subs, uint32_t nv_lo, uint32_t nv_hi,
     uint32_t nt_lo, uint32_t nt_hi,
     uint32_t *__restrict__ accept_g,  // per-vertex (wave 2)
-    uint32_t propose, uint32_t memo,
+    uint32_t propose,
     const uint32_t *__restrict__ sched /*block->label, largest first*/) {
   if (blockIdx.x >= nlabels) return;
   // biggest-label-first dispatch: per-label serial time scales with nt0,
@@ -754,11 +737,6 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
   using RMT = typename std::conditional<WAVEMODE, uint16_t,
                                         uint32_t>::type;
   __shared__ RMT s_remap[CAPV];
-  // pick memoization bitsets (MG_SIMP_MEMO): winners of this sub's
-  // collapses, and vertices whose pick goes stale for the next sub
-  constexpr uint32_t NBW = (CAPV + 63) / 64;
-  __shared__ unsigned long long s_win[NBW];
-  __shared__ unsigned long long s_dirty[NBW];
   const bool lds_mode = (nv <= CAPV);
   // WAVEMODE: the degree/cursor array lives in the PICK bytes — deg is
   // dead before the first pick write of each group (the reset pass
@@ -768,7 +746,6 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
   unsigned long long *pick_l =
       lds_mode ? s_pick : (pick + v0);
   RMT *rm = lds_mode ? s_remap : (RMT *)(remap + v0);
-  const bool memo_on = memo && lds_mode;
   // accept table (proposal wave): u32 id-only keys — lives in the
   // degree array's bytes (dead during the collapse phases) when the
   // label is LDS-resident; global fallback otherwise
@@ -986,11 +963,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     ++n_subs;
     if (sub > 0) {
       for (uint32_t v = tid; v < nv; v += BS) {
-        // memoized picks: only STALE vertices (1-ring of a collapse
-        // winner, marked in the previous sub's rewrite) re-enter the
-        // pick min-reduction; clean entries stay valid verbatim
-        if (!memo_on || ((s_dirty[v >> 6] >> (v & 63)) & 1ull))
-          pick_l[v] = ~0ull;
+        pick_l[v] = ~0ull;
         rm[v] = (RMT)v;
         if (propose) accept_l[v] = 0xFFFFFFFFu;
       }
@@ -1004,14 +977,6 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     // like the global k_edge_pick.
     for (uint32_t f = tid; f < nt; f += BS) {
       uint32_t fc[3] = {fa[3*f], fa[3*f+1], fa[3*f+2]};
-      if (memo_on && sub > 0) {
-        uint32_t l0 = fc[0] - v0, l1 = fc[1] - v0, l2 = fc[2] - v0;
-        bool any_dirty =
-            ((s_dirty[l0 >> 6] >> (l0 & 63)) & 1ull) |
-            ((s_dirty[l1 >> 6] >> (l1 & 63)) & 1ull) |
-            ((s_dirty[l2 >> 6] >> (l2 & 63)) & 1ull);
-        if (!any_dirty) continue;  // all edge costs memoized
-      }
       float cq[3][10], cp[3][3];
       #pragma unroll
       for (int ci = 0; ci < 3; ++ci) {
@@ -1053,14 +1018,6 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     // [7] matched-pair collapse (oracle step 3); rm was pre-set to the
     // identity in the same pass that reset the pick table
     if (tid == 0) s_collapses = 0;
-    if (memo_on) {
-      // dirty was consumed by this sub's pick; both bitsets restart
-      // (win set in collapse below, dirty re-marked in rewrite)
-      for (uint32_t i = tid; i < NBW; i += BS) {
-        s_win[i] = 0;
-        s_dirty[i] = 0;
-      }
-    }
     __syncthreads();
     for (uint32_t v = tid; v < nv; v += BS) {
       uint32_t u = v0 + v;
@@ -1090,7 +1047,6 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
       // is identical for the old and the cleared value — DESIGN §2)
       pick_l[v] = ~0ull;
       pick_l[w - v0] = ~0ull;
-      if (memo_on) atomicOr(&s_win[v >> 6], 1ull << (v & 63));
       atomicAdd(&s_collapses, 1u);
     }
     __syncthreads();
@@ -1132,7 +1088,6 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
         #pragma unroll
         for (int k = 0; k < 10; ++k) Q[12ull*u + k] += Q[12ull*w + k];
         rm[v] = (RMT)ul;
-        if (memo_on) atomicOr(&s_win[ul >> 6], 1ull << (ul & 63));
         atomicAdd(&s_collapses, 1u);
       }
       __syncthreads();
@@ -1140,7 +1095,7 @@ __attribute__((amdgpu_waves_per_eu(4))) void k_simplify_label(
     if (s_collapses == 0) break;
     // [8] fused rewrite + stable compact (oracle step 4)
     uint32_t kept = blk_rewrite_compact<BS, WAVEMODE ? 8192 : 65536, RMT>(
-        fa, fb, rm, v0, nt, s_sums, memo_on ? s_win : nullptr, s_dirty);
+        fa, fb, rm, v0, nt, s_sums);
     { uint32_t *t = fa; fa = fb; fb = t; }  // compacted faces now in fa
     if (tid == 0) s_nt = kept;
     __syncthreads();
