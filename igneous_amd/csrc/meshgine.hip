@@ -463,7 +463,7 @@ __global__ __launch_bounds__(BLK) void k_weld_insert(
   for (int rep = 0; rep < TPT; ++rep) {
     uint64_t t = span0 + (uint64_t)rep * BLK + threadIdx.x;
     if (t >= ntris) break;
-    uint2 rec = tri_recs[order[t]];
+    uint2 rec = order ? tri_recs[order[t]] : tri_recs[t];
     uint32_t s[3];
     decode_rec_slots(rec, s_pack, s_comb, s);
     slots_sorted[3 * t] = s[0];
@@ -1658,36 +1658,57 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   HIP_TRY(c, hipGetLastError(), 18);
   HIP_TRY(c, hipEventRecord(c->ev[4], s), 18);
 
-  // [4] stable partition by label id
-  if (ensure(c, c->order, T * 4)) return 19;
-  if (ensure(c, c->order_alt, T * 4)) return 19;
+  // [4] stable partition by label id. Default (MG_SORT_RECS=0 reverts):
+  // the 8-B records ride the radix sort as its 64-bit VALUES, so the
+  // weld insert reads them back SEQUENTIALLY instead of paying a
+  // random 8-B gather through the permutation.
+  const char *sre = getenv("MG_SORT_RECS");
+  const bool sort_recs = !(sre && sre[0] == '0');
+  if (ensure(c, c->keys_sorted, T * 20)) return 19;  // [sort-alt 8B] + [slot triples 12B]
   if (ensure(c, c->tri_label_alt, T * 4)) return 19;
-  if (ensure(c, c->keys_sorted, T * 12)) return 19;  // decoded slot triples
   uint32_t *order_sorted = nullptr;
   uint32_t *lab_sorted = nullptr;  // sort key output: label id per tri
+  const uint2 *rec_src = (const uint2 *)c->tri_keys.ptr;
+  uint32_t *slots_out = (uint32_t *)c->keys_sorted.ptr + 2 * T;
   {
     int blk = 256;
     uint64_t nb = (T + blk - 1) / blk;
-    hipLaunchKernelGGL(k_iota, dim3((uint32_t)nb), dim3(blk), 0, s,
-                       (uint32_t *)c->order.ptr, T);
     unsigned begin_bit = 0;
     unsigned end_bit = 1;
     while ((1u << end_bit) < nlabels) ++end_bit;
     if (nlabels == 1) end_bit = 1;
     rocprim::double_buffer<uint32_t> d_keys(
         (uint32_t *)c->tri_label.ptr, (uint32_t *)c->tri_label_alt.ptr);
-    rocprim::double_buffer<uint32_t> d_vals(
-        (uint32_t *)c->order.ptr, (uint32_t *)c->order_alt.ptr);
-    size_t tmp_bytes = 0;
-    hipError_t e = rocprim::radix_sort_pairs(
-        nullptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
-    if (e != hipSuccess) { SET_ERR(c, "radix_sort size query failed"); return 19; }
-    if (ensure(c, c->sort_tmp, tmp_bytes)) return 19;
-    e = rocprim::radix_sort_pairs(
-        c->sort_tmp.ptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
-    if (e != hipSuccess) { SET_ERR(c, "radix_sort failed"); return 19; }
-    // (the record gather is fused into k_weld_insert below)
-    order_sorted = d_vals.current();
+    if (sort_recs) {
+      rocprim::double_buffer<uint64_t> d_vals(
+          (uint64_t *)c->tri_keys.ptr, (uint64_t *)c->keys_sorted.ptr);
+      size_t tmp_bytes = 0;
+      hipError_t e = rocprim::radix_sort_pairs(
+          nullptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
+      if (e != hipSuccess) { SET_ERR(c, "radix_sort size query failed"); return 19; }
+      if (ensure(c, c->sort_tmp, tmp_bytes)) return 19;
+      e = rocprim::radix_sort_pairs(
+          c->sort_tmp.ptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
+      if (e != hipSuccess) { SET_ERR(c, "radix_sort failed"); return 19; }
+      rec_src = (const uint2 *)d_vals.current();
+      order_sorted = nullptr;  // records already label-partitioned
+    } else {
+      if (ensure(c, c->order, T * 4)) return 19;
+      if (ensure(c, c->order_alt, T * 4)) return 19;
+      hipLaunchKernelGGL(k_iota, dim3((uint32_t)nb), dim3(blk), 0, s,
+                         (uint32_t *)c->order.ptr, T);
+      rocprim::double_buffer<uint32_t> d_vals(
+          (uint32_t *)c->order.ptr, (uint32_t *)c->order_alt.ptr);
+      size_t tmp_bytes = 0;
+      hipError_t e = rocprim::radix_sort_pairs(
+          nullptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
+      if (e != hipSuccess) { SET_ERR(c, "radix_sort size query failed"); return 19; }
+      if (ensure(c, c->sort_tmp, tmp_bytes)) return 19;
+      e = rocprim::radix_sort_pairs(
+          c->sort_tmp.ptr, tmp_bytes, d_keys, d_vals, T, begin_bit, end_bit, s);
+      if (e != hipSuccess) { SET_ERR(c, "radix_sort failed"); return 19; }
+      order_sorted = d_vals.current();
+    }
     lab_sorted = d_keys.current();
     // label ranges
     if (ensure(c, c->tri_off, ((uint64_t)nlabels + 1) * 4)) return 19;
@@ -1713,14 +1734,14 @@ static int mesh_chunk_impl(mg_ctx *c, const void *labels_host,
   HIP_TRY(c, hipMemsetAsync(c->wh_keys.ptr, 0, wslots * 4, s), 20);
   uint32_t *wminp = (uint32_t *)c->wh_keys.ptr;
   uint32_t *wvtx = (uint32_t *)c->wh_vtx.ptr;
-  const uint32_t *slots_sorted = (const uint32_t *)c->keys_sorted.ptr;
+  const uint32_t *slots_sorted = slots_out;
   {
     int blk = 256;
     uint64_t nbt = (T + blk - 1) / blk;
     int wi_cfg = 1;  // 0: 1024x1, 1: 1024x2, 2: 1024x4
     if (const char *e = getenv("MG_WELD_INSERT_CFG")) wi_cfg = atoi(e);
-    uint32_t *rs_mut = (uint32_t *)c->keys_sorted.ptr;
-    const uint2 *tr = (const uint2 *)c->tri_keys.ptr;
+    uint32_t *rs_mut = slots_out;
+    const uint2 *tr = rec_src;
     const int64_t sxy64 = g.sx * g.sy;
     if (wi_cfg == 0) {
       uint64_t nb2 = (T + 1023) / 1024;
