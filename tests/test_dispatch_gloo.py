@@ -88,6 +88,14 @@ def test_two_rank_gloo_dispatch(tmp_path):
         "mesh/9:0:32-64_32-64_0-32",
     ])
 
+    # regression (advisor r01): every rank drains the task iterator, but
+    # on_finish side effects (the provenance append) must run exactly
+    # once — rank 0 only
+    prov = json.loads(cf.get("provenance"))
+    mesh_entries = [p for p in prov["processing"]
+                    if p["method"]["task"] == "MeshTask"]
+    assert len(mesh_entries) == 1,         f"provenance written {len(mesh_entries)}x across ranks"
+
 
 def test_shard_tasks_disjoint_complete():
     from igneous_amd.dispatch import shard_tasks
