@@ -37,7 +37,9 @@ layer = os.environ["MESHGINE_LAYER"]
 
 tasks = create_meshing_tasks(layer, mip=0, shape=(32, 32, 32),
                              simplification=False, spatial_index=False)
-ids = [i for i, _ in enumerate(tasks) if i % world == rank]
+# NOTE: enumerate the iterator ONCE — like the reference's, every full
+# iteration runs on_finish (one provenance append on rank 0)
+ids = [i for i in range(len(tasks)) if i % world == rank]
 n = execute_tasks(tasks)
 assert n == len(ids), (n, ids)
 dist.barrier()
